@@ -1,0 +1,47 @@
+"""Per-round model checkpointing.
+
+Keeps the reference's templated checkpoint naming contract
+(utils_run_task.py:327-397): the aggregated global model of round r is
+persisted under ``model_update_style`` with ``{task_id}`` and
+``{current_round}`` substituted — default
+``{task_id}_{current_round}_result_model.safetensors`` (the reference's
+default suffix is .mnn, a phone-side format; payload here is
+safetensors).  Round r>0 resumes by loading round r-1's artifact.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, Optional
+
+import torch
+
+DEFAULT_STYLE = "{task_id}_{current_round}_result_model.safetensors"
+
+
+def checkpoint_name(task_id: str, current_round: int,
+                    model_update_style: str = "") -> str:
+    style = model_update_style or DEFAULT_STYLE
+    return style.format(task_id=task_id, current_round=current_round)
+
+
+def save_checkpoint(directory: str, task_id: str, current_round: int,
+                    state: Dict[str, torch.Tensor],
+                    model_update_style: str = "") -> str:
+    from safetensors.torch import save_file
+    os.makedirs(directory, exist_ok=True)
+    path = os.path.join(directory, checkpoint_name(
+        task_id, current_round, model_update_style))
+    save_file({k: v.detach().cpu().contiguous() for k, v in state.items()}, path)
+    return path
+
+
+def load_checkpoint(directory: str, task_id: str, current_round: int,
+                    model_update_style: str = "",
+                    device: str = "cpu") -> Optional[Dict[str, torch.Tensor]]:
+    from safetensors.torch import load_file
+    path = os.path.join(directory, checkpoint_name(
+        task_id, current_round, model_update_style))
+    if not os.path.exists(path):
+        return None
+    return load_file(path, device=device)

@@ -1,0 +1,62 @@
+"""Engine job specification.
+
+The runnable form of a task's logical simulation: what the reference
+serialises into the Ray job entrypoint JSON (taskMgr/task_runner.py:69-75
+submitting run_task.py --task '<json>').  Built either directly (bench,
+tests) or from a TaskConfig by task/runner.py.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, Optional
+
+
+@dataclass
+class EngineJob:
+    task_id: str = "job"
+    model_name: str = "mlp"
+    model_kwargs: Dict[str, Any] = field(default_factory=dict)
+
+    # population & per-round work
+    clients: int = 10               # population size (this rank's shard)
+    cohort_size: int = 0            # clients trained per round (0 = all)
+    rounds: int = 1
+    local_steps: int = 2            # E: local SGD steps per client per round
+    batch_size: int = 8             # per-client local batch
+    lr: float = 0.05
+    prox_mu: float = 0.0            # FedProx proximal coefficient (0 = FedAvg)
+
+    # execution
+    dtype: str = "float32"          # compute dtype of client replicas
+    device: str = "cpu"
+    chunk_clients: int = 0          # cohort chunk co-resident on the GPU (0 = auto)
+    seed: int = 1234
+
+    # data
+    num_classes: int = 10
+    dirichlet_alpha: float = 0.1
+    shard_size: int = 64
+    vocab_size: int = 0             # >0 switches to LM data
+    seq_len: int = 0
+
+    # behaviour simulation (deviceflow): arrival/offline/drop shaping
+    behavior_strategy: str = ""     # gradient-house strategy JSON ("" = none)
+
+    # bookkeeping / checkpointing
+    checkpoint_dir: str = ""
+    model_update_style: str = ""    # e.g. "{task_id}_{current_round}_result_model.safetensors"
+    save_every_round: bool = False
+
+    # failure-tolerance accounting (reference total_simulation semantics)
+    data_name: str = "data_0"
+    device_tier: str = "high"
+    dynamic_num: int = 0            # tolerated failures per round
+
+    def resolved_cohort(self) -> int:
+        return self.cohort_size if self.cohort_size > 0 else self.clients
+
+    def torch_dtype(self):
+        import torch
+        return {"float32": torch.float32, "bfloat16": torch.bfloat16,
+                "float16": torch.float16}[self.dtype]

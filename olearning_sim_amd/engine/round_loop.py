@@ -1,0 +1,233 @@
+"""The simulation round loop — the hot path.
+
+MI355X-native equivalent of the reference's RayRunner
+(ols_core/taskMgr/run_task.py:212-322): for each operator-flow round,
+select the cohort, train every cohort client locally (client-batched on
+the GPU instead of one subprocess per phone), aggregate the weighted
+deltas (RCCL all-reduce across GPUs), update the global model, record
+per-round success/failed counts with the reference's tolerance
+semantics, and checkpoint under the reference's templated name.
+
+Behaviour simulation (the deviceflow gradient house): an offline mask
+removes clients before training (they count as failed machine-times, the
+arrival/spike curves), a drop mask zeroes a trained client's aggregation
+weight (message dropped in flight — the device itself still succeeded,
+matching deviceflow semantics where drops happen after NotifyComplete).
+"""
+
+from __future__ import annotations
+
+import json
+import time
+from typing import Any, Callable, Dict, List, Optional
+
+import torch
+
+from ..models import build_model
+from ..ops import fused
+from ..parallel import dist as pdist
+from .checkpoint import save_checkpoint
+from .client_manager import FlatParams, chunk_ids
+from .data import SyntheticFederatedData
+from .job import EngineJob
+from .local_train import LocalTrainer
+
+# signature: sampler(round_idx, cohort_size) -> (offline_mask[C], drop_mask[C]) bool tensors
+BehaviorFn = Callable[[int, int], tuple]
+
+
+class LogicalEngine:
+    def __init__(self, job: EngineJob,
+                 dist_ctx: Optional[pdist.DistContext] = None,
+                 behavior: Optional[BehaviorFn] = None,
+                 result_sink: Optional[Callable[[Dict[str, Any]], None]] = None):
+        self.job = job
+        self.ctx = dist_ctx or pdist.DistContext(device=job.device)
+        self.device = torch.device(self.ctx.device if dist_ctx else job.device)
+        self.dtype = job.torch_dtype()
+        self.result_sink = result_sink
+        self.stop_requested = False
+
+        self.model = build_model(job.model_name, **job.model_kwargs)
+        gen = torch.Generator().manual_seed(job.seed)
+        global_params = self.model.init_global(
+            device="cpu", dtype=torch.float32, generator=gen)
+        global_params = {k: v.to(self.device) for k, v in global_params.items()}
+        self.master = FlatParams(global_params)
+        if self.ctx.enabled:
+            # all ranks seed identically, but broadcast pins exact equality
+            pdist.broadcast_flat(self.master.flat)
+
+        self.data = SyntheticFederatedData(
+            clients=job.clients, num_classes=job.num_classes,
+            input_shape=self.model.input_shape,
+            dirichlet_alpha=job.dirichlet_alpha, shard_size=job.shard_size,
+            seed=job.seed + 7919 * self.ctx.rank, device=str(self.device),
+            vocab_size=job.vocab_size, seq_len=job.seq_len)
+        self.trainer = LocalTrainer(
+            self.model, self.master, self.data, lr=job.lr,
+            prox_mu=job.prox_mu, local_steps=job.local_steps,
+            batch_size=job.batch_size, dtype=self.dtype,
+            report_loss=result_sink is not None)
+        if behavior is None and job.behavior_strategy:
+            from ..deviceflow.sampler import BehaviorSampler
+            behavior = BehaviorSampler(job.behavior_strategy,
+                                       seed=job.seed + self.ctx.rank,
+                                       device=str(self.device))
+        self.behavior = behavior
+
+        self._delta = self.master.zeros_like_flat()
+        self._scalar = torch.zeros(3, dtype=torch.float64, device=self.device)
+        # totals across rounds (reference logical_result accounting)
+        self.success_total = 0
+        self.failed_total = 0
+
+    # ------------------------------------------------------------------
+    def _chunk_size(self, cohort: int) -> int:
+        if self.job.chunk_clients > 0:
+            return min(self.job.chunk_clients, cohort)
+        if self.device.type != "cuda":
+            return min(cohort, 64)
+        # auto: bound replica+grad+activation memory to a fraction of HBM.
+        p = self.master.numel()
+        bytes_per_client = p * self.dtype.itemsize * 2  # weights + grad
+        # activations per client (rough: 40x input bytes per local batch)
+        act = 1
+        for s in self.model.input_shape:
+            act *= s
+        bytes_per_client += self.job.batch_size * act * self.dtype.itemsize * 40
+        free = torch.cuda.get_device_properties(self.device).total_memory
+        budget = int(free * 0.55)
+        return max(1, min(cohort, budget // max(1, bytes_per_client)))
+
+    def select_cohort(self, round_idx: int) -> torch.Tensor:
+        """Deterministic rotating window over this rank's population
+        (the reference enumerates machine-times exhaustively per round;
+        a cohort < population rotates so every client participates)."""
+        job = self.job
+        cohort = min(job.resolved_cohort(), job.clients)
+        start = (round_idx * cohort) % job.clients
+        ids = (start + torch.arange(cohort)) % job.clients
+        return ids.to(torch.int64)
+
+    # ------------------------------------------------------------------
+    def run_round(self, round_idx: int) -> Dict[str, Any]:
+        job = self.job
+        ids = self.select_cohort(round_idx)
+        cohort = int(ids.numel())
+
+        if self.behavior is not None:
+            offline, dropped = self.behavior(round_idx, cohort)
+        else:
+            offline = torch.zeros(cohort, dtype=torch.bool)
+            dropped = torch.zeros(cohort, dtype=torch.bool)
+        offline = offline.cpu()
+        dropped = dropped.cpu()
+        active_ids = ids[~offline]
+        active_drop = dropped[~offline]
+
+        self.trainer.begin_round()
+        self._delta.zero_()
+        weights_all = (~active_drop).float()
+        losses: List[float] = []
+        trained = 0
+        chunk = self._chunk_size(max(1, int(active_ids.numel())))
+        for lo in range(0, int(active_ids.numel()), chunk):
+            cid = active_ids[lo:lo + chunk].to(self.device)
+            w = weights_all[lo:lo + chunk].to(self.device)
+            stats = self.trainer.train_chunk(cid, w, round_idx, self._delta)
+            trained += stats["clients"]
+            if stats["loss"]:
+                losses.append(stats["loss"])
+
+        local_weight = float(weights_all.sum())
+        success_local = int(active_ids.numel())
+        failed_local = int(offline.sum())
+
+        # cross-GPU aggregation: one RCCL all-reduce of the flat delta +
+        # one small all-reduce carrying (total_weight, success, failed)
+        if self.ctx.enabled:
+            self._scalar[0] = local_weight
+            self._scalar[1] = success_local
+            self._scalar[2] = failed_local
+            work = pdist.all_reduce_flat(self._delta, async_op=True)
+            pdist.all_reduce_flat(self._scalar)
+            if work is not None:
+                work.wait()
+            total_weight = float(self._scalar[0])
+            success = int(self._scalar[1])
+            failed = int(self._scalar[2])
+        else:
+            total_weight, success, failed = local_weight, success_local, failed_local
+
+        if total_weight > 0:
+            fused.apply_aggregate([self.master.flat], [self._delta], total_weight)
+
+        self.success_total += success
+        self.failed_total += failed
+
+        record = {
+            "round": round_idx,
+            "success": success,
+            "failed": failed,
+            "trained": trained,
+            "loss": sum(losses) / len(losses) if losses else None,
+            "round_failed": failed > job.dynamic_num,
+        }
+        if job.save_every_round and job.checkpoint_dir and self.ctx.rank == 0:
+            record["checkpoint"] = save_checkpoint(
+                job.checkpoint_dir, job.task_id, round_idx,
+                self.master.state_dict(), job.model_update_style)
+        return record
+
+    # ------------------------------------------------------------------
+    def run(self) -> Dict[str, Any]:
+        job = self.job
+        records: List[Dict[str, Any]] = []
+        t0 = time.time()
+        for r in range(job.rounds):
+            if self.stop_requested:
+                break
+            rec = self.run_round(r)
+            records.append(rec)
+            if self.result_sink is not None:
+                self.result_sink(self._round_result(rec))
+            if rec["round_failed"]:
+                break
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+        elapsed = time.time() - t0
+        rounds_done = len(records)
+        world = max(1, self.ctx.world_size)
+        return {
+            "task_id": job.task_id,
+            "rounds": rounds_done,
+            "elapsed_s": elapsed,
+            "rounds_per_s": rounds_done / elapsed if elapsed > 0 else 0.0,
+            "clients_per_s": (self.success_total / elapsed) if elapsed > 0 else 0.0,
+            "success_total": self.success_total,
+            "failed_total": self.failed_total,
+            "final_loss": records[-1]["loss"] if records else None,
+            "world_size": world,
+            "records": records,
+        }
+
+    def _round_result(self, rec: Dict[str, Any]) -> Dict[str, Any]:
+        """Shape a round record like the reference's logical_result row
+        (run_task.py analyze_results:149-210)."""
+        job = self.job
+        return {
+            "task_id": job.task_id,
+            "logical_round": rec["round"] + 1,
+            "logical_operator": "train",
+            "logical_result": {
+                "logical_result": [{
+                    "name": job.data_name,
+                    "simulation_target": {
+                        "devices": [job.device_tier],
+                        "success_num": [rec["success"]],
+                        "failed_num": [rec["failed"]],
+                    }}]},
+            "round_failed": rec["round_failed"],
+            "loss": rec["loss"],
+        }

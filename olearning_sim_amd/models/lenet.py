@@ -1,0 +1,52 @@
+"""LeNet-5 for the CIFAR-10 1k-client config (BASELINE config 2)."""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from .base import (ClientBatchedModel, Params, binit, kaiming,
+                   blinear, bconv2d)
+
+
+class LeNet5(ClientBatchedModel):
+    name = "lenet"
+    num_classes = 10
+    input_shape = (3, 32, 32)
+
+    def __init__(self, in_ch: int = 3, num_classes: int = 10):
+        self.in_ch = in_ch
+        self.num_classes = num_classes
+        self.input_shape = (in_ch, 32, 32)
+
+    def init_global(self, device="cpu", dtype=torch.float32,
+                    generator: Optional[torch.Generator] = None) -> Params:
+        ic, k = self.in_ch, self.num_classes
+        g = generator
+        return {
+            "conv1.w": kaiming((6, ic, 5, 5), ic * 25, device, dtype, g),
+            "conv1.b": binit((6,), ic * 25, device, dtype, g),
+            "conv2.w": kaiming((16, 6, 5, 5), 6 * 25, device, dtype, g),
+            "conv2.b": binit((16,), 6 * 25, device, dtype, g),
+            "fc1.w": binit((120, 400), 400, device, dtype, g),
+            "fc1.b": binit((120,), 400, device, dtype, g),
+            "fc2.w": binit((84, 120), 120, device, dtype, g),
+            "fc2.b": binit((84,), 120, device, dtype, g),
+            "fc3.w": binit((k, 84), 84, device, dtype, g),
+            "fc3.b": binit((k,), 84, device, dtype, g),
+        }
+
+    def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        # x: [C, B, 3, 32, 32] -> channel-grouped [B, C*3, 32, 32]
+        C, B = x.shape[0], x.shape[1]
+        xg = x.permute(1, 0, 2, 3, 4).reshape(B, C * self.in_ch, 32, 32)
+        h = F.relu(bconv2d(xg, params["conv1.w"], C, params["conv1.b"]))   # 28
+        h = F.max_pool2d(h, 2)                                             # 14
+        h = F.relu(bconv2d(h, params["conv2.w"], C, params["conv2.b"]))    # 10
+        h = F.max_pool2d(h, 2)                                             # 5
+        h = h.reshape(B, C, 16 * 5 * 5).permute(1, 0, 2)                   # [C,B,400]
+        h = F.relu(blinear(h, params["fc1.w"], params["fc1.b"]))
+        h = F.relu(blinear(h, params["fc2.w"], params["fc2.b"]))
+        return blinear(h, params["fc3.w"], params["fc3.b"])

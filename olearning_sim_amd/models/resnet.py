@@ -1,0 +1,101 @@
+"""ResNet-18 (CIFAR variant, GroupNorm) — the flagship FedAvg CNN.
+
+BASELINE config 3: FedAvg ResNet-18 on CIFAR-100 non-IID, 10k clients
+across 8 GPUs.  BatchNorm is replaced with GroupNorm(8) as is standard
+for federated training (per-client running stats are meaningless when
+every client re-starts from the global model each round); all clients'
+convs run as one grouped conv per layer in the channel-grouped layout
+[B, C*ch, H, W] (models/base.py).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from .base import (ClientBatchedModel, Params, binit, kaiming,
+                   blinear, bconv2d, bgroupnorm)
+
+_STAGES = (64, 128, 256, 512)
+_GN_GROUPS = 8
+
+
+class ResNet18(ClientBatchedModel):
+    name = "resnet18"
+    num_classes = 100
+    input_shape = (3, 32, 32)
+
+    def __init__(self, num_classes: int = 100, in_ch: int = 3,
+                 width_mult: float = 1.0):
+        self.num_classes = num_classes
+        self.in_ch = in_ch
+        self.widths = [int(w * width_mult) for w in _STAGES]
+        self.input_shape = (in_ch, 32, 32)
+
+    # ------------------------------------------------------------------
+    def init_global(self, device="cpu", dtype=torch.float32,
+                    generator: Optional[torch.Generator] = None) -> Params:
+        g = generator
+        p: Params = {}
+
+        def conv(name, oc, ic, k):
+            p[f"{name}.w"] = kaiming((oc, ic, k, k), ic * k * k, device, dtype, g)
+
+        def gn(name, ch):
+            p[f"{name}.g"] = torch.ones(ch, device=device, dtype=dtype)
+            p[f"{name}.b"] = torch.zeros(ch, device=device, dtype=dtype)
+
+        w = self.widths
+        conv("stem", w[0], self.in_ch, 3)
+        gn("stem.gn", w[0])
+        in_c = w[0]
+        for s, out_c in enumerate(w):
+            for blk in range(2):
+                pre = f"s{s}.b{blk}"
+                ic = in_c if blk == 0 else out_c
+                conv(f"{pre}.c1", out_c, ic, 3)
+                gn(f"{pre}.gn1", out_c)
+                conv(f"{pre}.c2", out_c, out_c, 3)
+                gn(f"{pre}.gn2", out_c)
+                if blk == 0 and (ic != out_c or s > 0):
+                    conv(f"{pre}.down", out_c, ic, 1)
+                    gn(f"{pre}.gndown", out_c)
+            in_c = out_c
+        p["fc.w"] = binit((self.num_classes, w[3]), w[3], device, dtype, g)
+        p["fc.b"] = binit((self.num_classes,), w[3], device, dtype, g)
+        return p
+
+    # ------------------------------------------------------------------
+    def _block(self, params: Params, x: torch.Tensor, C: int, pre: str,
+               stride: int, has_down: bool) -> torch.Tensor:
+        h = bconv2d(x, params[f"{pre}.c1.w"], C, stride=stride, padding=1)
+        h = F.relu(bgroupnorm(h, C, _GN_GROUPS,
+                              params[f"{pre}.gn1.g"], params[f"{pre}.gn1.b"]))
+        h = bconv2d(h, params[f"{pre}.c2.w"], C, stride=1, padding=1)
+        h = bgroupnorm(h, C, _GN_GROUPS,
+                       params[f"{pre}.gn2.g"], params[f"{pre}.gn2.b"])
+        if has_down:
+            sc = bconv2d(x, params[f"{pre}.down.w"], C, stride=stride)
+            sc = bgroupnorm(sc, C, _GN_GROUPS,
+                            params[f"{pre}.gndown.g"], params[f"{pre}.gndown.b"])
+        else:
+            sc = x
+        return F.relu(h + sc)
+
+    def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        # x: [C, B, 3, 32, 32]
+        C, B = x.shape[0], x.shape[1]
+        h = x.permute(1, 0, 2, 3, 4).reshape(B, C * self.in_ch, 32, 32)
+        h = bconv2d(h, params["stem.w"], C, stride=1, padding=1)
+        h = F.relu(bgroupnorm(h, C, _GN_GROUPS,
+                              params["stem.gn.g"], params["stem.gn.b"]))
+        for s in range(4):
+            stride = 1 if s == 0 else 2
+            has_down = s > 0
+            h = self._block(params, h, C, f"s{s}.b0", stride, has_down)
+            h = self._block(params, h, C, f"s{s}.b1", 1, False)
+        # h: [B, C*512, 4, 4] -> global average pool -> [C, B, 512]
+        h = h.mean(dim=(2, 3)).reshape(B, C, self.widths[3]).permute(1, 0, 2)
+        return blinear(h, params["fc.w"], params["fc.b"])

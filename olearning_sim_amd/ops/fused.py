@@ -1,0 +1,202 @@
+"""Python surface of the fused HIP ops.
+
+Replaces the per-device subprocess training loop of the reference
+(ols_core/taskMgr/utils/utils_run_task.py:481-514) with client-batched
+device kernels.  Three bandwidth-bound fused ops carry the optimizer/
+aggregation tail of every local-train step:
+
+- fused_sgd_update:      w_c -= lr * (g_c + mu * (w_c - w_global))
+                         (mu=0 -> FedAvg local SGD; mu>0 -> FedProx)
+- weighted_delta_accum:  delta += sum_c alpha_c * (w_c - w_global)
+- apply_aggregate:       w_global += delta / total_weight  (fp32 master)
+- cross_entropy_fwd_bwd: fused softmax CE loss + dlogits in one pass
+
+On ROCm these run as hand-written gfx950 kernels (csrc/*.hip, bf16x8 per
+lane); on CPU they fall back to eager torch so the control-plane tests
+run here.  A CUDA tensor with no extension raises (no silent fallback).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Sequence
+
+import torch
+
+_HIP_OPS = None
+_HIP_OPS_TRIED = False
+
+_SO_NAME = "_hip_ops.so"
+
+
+def _so_path() -> str:
+    return os.path.join(os.path.dirname(os.path.abspath(__file__)), _SO_NAME)
+
+
+def load_hip_ops(required: bool = False):
+    """Load the in-tree extension if built. Caches the result."""
+    global _HIP_OPS, _HIP_OPS_TRIED
+    if _HIP_OPS is not None:
+        return _HIP_OPS
+    if _HIP_OPS_TRIED and not required:
+        return None
+    _HIP_OPS_TRIED = True
+    path = _so_path()
+    if os.path.exists(path):
+        torch.ops.load_library(path)
+        _HIP_OPS = torch.ops.olsim_hip
+        return _HIP_OPS
+    if required:
+        raise RuntimeError(
+            f"olearning_sim_amd HIP extension missing: {path} not built. "
+            f"Run `python -m olearning_sim_amd.ops.build` (hipcc, gfx950).")
+    return None
+
+
+def hip_ops_available() -> bool:
+    return load_hip_ops() is not None
+
+
+def _gpu_ops(t: torch.Tensor):
+    """Return the HIP op namespace for a GPU tensor; raise if unavailable."""
+    if t.is_cuda:
+        return load_hip_ops(required=True)
+    return None
+
+
+# ---------------------------------------------------------------------------
+def fused_sgd_update(params: Sequence[torch.Tensor],
+                     grads: Sequence[torch.Tensor],
+                     lr: float,
+                     mu: float = 0.0,
+                     global_params: Optional[Sequence[torch.Tensor]] = None) -> None:
+    """In-place SGD / FedProx step over per-client parameter tensors.
+
+    params[i]: [C, ...] client replicas; grads[i] same shape;
+    global_params[i]: [...] (no client dim) — required when mu > 0.
+    """
+    if not params:
+        return
+    ops = _gpu_ops(params[0])
+    if ops is not None:
+        gl: List[torch.Tensor] = (
+            [g.reshape(-1) for g in global_params] if (global_params is not None and mu != 0.0)
+            else [torch.empty(0, dtype=params[0].dtype, device=params[0].device)] * len(params))
+        ops.fused_sgd_update(list(params), list(grads), gl, float(lr), float(mu))
+        return
+    with torch.no_grad():
+        for i, (w, g) in enumerate(zip(params, grads)):
+            upd = g
+            if mu != 0.0 and global_params is not None:
+                upd = g + mu * (w - global_params[i].to(w.dtype))
+            w.add_(upd, alpha=-lr)
+
+
+def weighted_delta_accum(delta: Sequence[torch.Tensor],
+                         client_params: Sequence[torch.Tensor],
+                         global_params: Sequence[torch.Tensor],
+                         weights: torch.Tensor) -> None:
+    """delta[i] (fp32, shape [...]) += sum_c weights[c] * (client_params[i][c] - global[i])."""
+    if not delta:
+        return
+    ops = _gpu_ops(client_params[0])
+    if ops is not None:
+        ops.weighted_delta_accum(list(delta), list(client_params),
+                                 list(global_params), weights.float())
+        return
+    with torch.no_grad():
+        for dl, cw, gw in zip(delta, client_params, global_params):
+            diff = cw.float() - gw.float().unsqueeze(0)
+            shape = [-1] + [1] * (cw.dim() - 1)
+            dl.add_((diff * weights.view(shape)).sum(dim=0))
+
+
+def apply_aggregate(global_params: Sequence[torch.Tensor],
+                    delta: Sequence[torch.Tensor],
+                    total_weight: float) -> None:
+    """w_global += delta / total_weight, in place (fp32 master weights)."""
+    if total_weight == 0:
+        return
+    with torch.no_grad():
+        inv = 1.0 / float(total_weight)
+        torch._foreach_add_([g for g in global_params],
+                            [d for d in delta], alpha=inv)
+
+
+# -- flat (single-buffer) forms used by the engine hot path ----------------
+#
+# Replica layout (engine/client_manager.replicate_flat): the 1-D buffer of
+# length C*P is the concatenation over parameter blocks of [C, n_b]; block
+# b covers global-master elements offsets[b]:offsets[b+1].
+
+def fused_sgd_update_flat(buf: torch.Tensor, grad: torch.Tensor,
+                          global_flat: torch.Tensor, clients: int,
+                          lr: float, mu: float,
+                          offsets: Optional[torch.Tensor] = None) -> None:
+    """buf -= lr * (grad + mu * (buf - replicated(global_flat))), in place."""
+    if buf.is_cuda:
+        ops = load_hip_ops(required=True)
+        ops.fused_sgd_update_flat(
+            buf, grad,
+            global_flat if mu != 0.0 else torch.empty(0, dtype=buf.dtype, device=buf.device),
+            offsets if (mu != 0.0 and offsets is not None) else
+            torch.empty(0, dtype=torch.int64, device=buf.device),
+            int(clients), float(lr), float(mu))
+        return
+    with torch.no_grad():
+        if mu == 0.0 or offsets is None:
+            buf.add_(grad, alpha=-lr)
+            return
+        offs = offsets.tolist()
+        for b in range(len(offs) - 1):
+            g0, g1 = offs[b], offs[b + 1]
+            n = g1 - g0
+            blk = buf[clients * g0:clients * g1].view(clients, n)
+            gblk = grad[clients * g0:clients * g1].view(clients, n)
+            prox = blk - global_flat[g0:g1].unsqueeze(0)
+            blk.add_(gblk + mu * prox, alpha=-lr)
+
+
+def weighted_delta_accum_flat(delta: torch.Tensor, buf: torch.Tensor,
+                              global_flat: torch.Tensor,
+                              weights: torch.Tensor, clients: int,
+                              offsets: Optional[torch.Tensor] = None) -> None:
+    """delta[j] += sum_c weights[c] * (buf[c,j] - global_flat[j]) per block."""
+    if buf.is_cuda:
+        ops = load_hip_ops(required=True)
+        ops.weighted_delta_accum_flat(delta, buf, global_flat,
+                                      weights.float(), offsets, int(clients))
+        return
+    with torch.no_grad():
+        offs = offsets.tolist()
+        w = weights.float().view(clients, 1)
+        for b in range(len(offs) - 1):
+            g0, g1 = offs[b], offs[b + 1]
+            n = g1 - g0
+            blk = buf[clients * g0:clients * g1].view(clients, n).float()
+            diff = blk - global_flat[g0:g1].float().unsqueeze(0)
+            delta[g0:g1].add_((diff * w).sum(dim=0))
+
+
+class _CrossEntropyFn(torch.autograd.Function):
+    """Fused CE over [N, K] logits (HIP kernel fwd computes loss AND
+    dlogits in one pass; backward is a scale)."""
+
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, labels: torch.Tensor):
+        ops = load_hip_ops(required=True)
+        loss, dlogits = ops.cross_entropy_fwd_bwd(logits, labels)
+        ctx.save_for_backward(dlogits)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (dlogits,) = ctx.saved_tensors
+        return dlogits * grad_out, None
+
+
+def cross_entropy_fwd_bwd(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """Mean cross-entropy over all rows. logits [N, K] (bf16/f32), labels [N] int64."""
+    if logits.is_cuda:
+        return _CrossEntropyFn.apply(logits, labels)
+    return torch.nn.functional.cross_entropy(logits.float(), labels)

@@ -1,0 +1,154 @@
+"""Deviceflow behaviour-model math, checked against hand-computed values
+of the reference algorithm (ols_core/deviceflow/non_grpc/strategy.py)."""
+
+import json
+import math
+import random
+
+import pytest
+
+from olearning_sim_amd.deviceflow.strategy import Strategy
+
+
+def flow(spec):
+    return json.dumps({"flow_dispatch": dict({"use_strategy": True}, **spec)})
+
+
+def test_real_time_detection_and_analysis():
+    s = json.dumps({"real_time_dispatch": {
+        "use_strategy": True, "dispatch_batch_sizes": [10, 20],
+        "drop_simulation": {"drop_probability": 0.25}}})
+    assert Strategy.check_real_time_dispatch(s)
+    sizes, p = Strategy.real_time_strategy_analysis(s)
+    assert sizes == [10, 20] and p == 0.25
+    assert not Strategy.check_real_time_dispatch(json.dumps({}))
+
+
+def test_specific_timing_relative():
+    s = flow({"total_dispatch_amount": 60,
+              "specific_timing": {"use": True, "time_type": "relative",
+                                  "timings": [0, 5, 10],
+                                  "amounts": [10, 20, 30]}})
+    timing, amounts, drops = Strategy.flow_strategy_analysis(s, "t_op_0")
+    assert timing == [0, 5, 10]
+    assert amounts == [10, 20, 30]
+    assert drops == [[], [], []]
+
+
+def test_specific_timing_mismatched_lengths_rejected():
+    s = flow({"total_dispatch_amount": 10,
+              "specific_timing": {"use": True, "timings": [0, 1],
+                                  "amounts": [10]}})
+    assert Strategy.flow_strategy_analysis(s, "t_op_0") == ([], [], [])
+
+
+def test_both_timing_and_interval_rejected():
+    s = flow({"total_dispatch_amount": 10,
+              "specific_timing": {"use": True, "timings": [0], "amounts": [10]},
+              "specific_interval": {"use": True}})
+    assert Strategy.flow_strategy_analysis(s, "t_op_0") == ([], [], [])
+
+
+def test_interval_constant_rate_splits_evenly():
+    # f(t)=1 over a 10 s interval: 10 slots, equal area -> 10 msgs/slot
+    s = flow({"total_dispatch_amount": 100,
+              "specific_interval": {
+                  "use": True, "time_type": "relative",
+                  "intervals": [[0, 10]],
+                  "dispatch_rules": {"domains": [[0.0, 10.0]],
+                                     "functions": ["1"]}}})
+    timing, amounts, drops = Strategy.flow_strategy_analysis(s, "t_op_0")
+    assert len(amounts) == 10
+    assert sum(amounts) == 100
+    assert amounts == [10] * 10
+    assert timing == [0] + [1] * 9
+
+
+def test_interval_sin_spike_total_preserved():
+    # the reference README example: math.sin(t)+1 over [0, 6.28]
+    s = flow({"total_dispatch_amount": 500,
+              "specific_interval": {
+                  "use": True,
+                  "intervals": [[0, 10]],
+                  "dispatch_rules": {"domains": [[0.0, 6.28]],
+                                     "functions": ["math.sin(t)+1"]}}})
+    timing, amounts, drops = Strategy.flow_strategy_analysis(s, "t_op_0")
+    assert sum(amounts) == 500
+    assert len(amounts) == 10
+    # rate peaks near t=pi/2 (slot 2-3 of 10) and dips near 3*pi/2
+    assert max(amounts) == amounts[2]
+    assert min(amounts) == amounts[7]
+
+
+def test_two_intervals_area_apportioning():
+    # f=2 on 5 s vs f=1 on 5 s: amounts split 2:1
+    s = flow({"total_dispatch_amount": 150,
+              "specific_interval": {
+                  "use": True,
+                  "intervals": [[0, 5], [10, 15]],
+                  "dispatch_rules": {"domains": [[0.0, 5.0], [0.0, 5.0]],
+                                     "functions": ["2", "1"]}}})
+    timing, amounts, drops = Strategy.flow_strategy_analysis(s, "t_op_0")
+    assert sum(amounts) == 150
+    assert sum(amounts[:5]) == 100
+    assert sum(amounts[5:]) == 50
+    # gap between interval 0 (slots 0..4) and interval 1 (slots 10..14)
+    assert timing[5] == 6  # 10 - 4
+
+
+def test_negative_rate_only_positive_area_counts():
+    # f(t) = -1 everywhere -> zero positive area -> rejected
+    s = flow({"total_dispatch_amount": 10,
+              "specific_interval": {
+                  "use": True, "intervals": [[0, 5]],
+                  "dispatch_rules": {"domains": [[0.0, 5.0]],
+                                     "functions": ["-1"]}}})
+    assert Strategy.flow_strategy_analysis(s, "t_op_0") == ([], [], [])
+
+
+def test_drop_probability_bounds():
+    rng = random.Random(0)
+    drops = Strategy.generate_drop_list(
+        [10, 10, 10], {"drop_probability": [0, 1, 0.5]}, rng)
+    assert drops[0] == []
+    assert drops[1] == list(range(10))
+    assert all(0 <= i < 10 for i in drops[2])
+
+
+def test_drop_amounts_exact():
+    rng = random.Random(1)
+    drops = Strategy.generate_drop_list(
+        [10, 10, 10], {"drop_amounts": [0, 4, 15]}, rng)
+    assert drops[0] == []
+    assert len(drops[1]) == 4 and drops[1] == sorted(drops[1])
+    assert drops[2] == list(range(10))   # over-drop clamps to everything
+
+
+def test_interval_drop_probability_expansion():
+    s = flow({"total_dispatch_amount": 100,
+              "specific_interval": {
+                  "use": True, "intervals": [[0, 10]],
+                  "dispatch_rules": {"domains": [[0.0, 10.0]],
+                                     "functions": ["1"]},
+                  "drop_simulation": {"drop_probability": [1.0]}}})
+    timing, amounts, drops = Strategy.flow_strategy_analysis(
+        s, "t_op_0", rng=random.Random(3))
+    assert all(d == list(range(a)) for d, a in zip(drops, amounts))
+
+
+def test_round_indexed_absolute_interval_uses_flow_round():
+    base = {"total_dispatch_amount": 50,
+            "specific_interval": {
+                "use": True, "time_type": "absolute",
+                "intervals": [
+                    [["2099-01-01 00:00:00", "2099-01-01 00:00:05"]],
+                    [["2099-01-01 01:00:00", "2099-01-01 01:00:05"]]],
+                "dispatch_rules": {"domains": [[0.0, 5.0]],
+                                   "functions": ["1"]}}}
+    from datetime import datetime
+    now = datetime(2098, 12, 31, 23, 59, 0)
+    t0, a0, _ = Strategy.flow_strategy_analysis(flow(base), "t_op_0", now=now)
+    t1, a1, _ = Strategy.flow_strategy_analysis(flow(base), "t_op_1", now=now)
+    assert sum(a0) == 50 and sum(a1) == 50
+    # round 1 starts one hour later than round 0
+    assert t1[0] - t0[0] == pytest.approx(3600, abs=1)
