@@ -1,0 +1,170 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: FedAvg CNN federated-simulation throughput.
+
+One step = one full FL round: cohort selection -> per-client local
+training (client-batched on GPU, E local SGD steps each) -> weighted
+delta aggregation (HIP kernel) -> RCCL all-reduce across GPUs -> global
+model update.  Weak scaling: each GPU simulates a fixed shard of
+clients (default 1250 -> 10k clients at 8 GPUs, BASELINE config 3).
+
+value = whole-job virtual clients trained per second (aggregate over
+all ranks); rounds/sec is value / clients_per_round.
+
+Run:  python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches it under torch.distributed.run with one
+rank per GPU (RANK/LOCAL_RANK/WORLD_SIZE read from the env).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from olearning_sim_amd.engine import EngineJob, LogicalEngine  # noqa: E402
+from olearning_sim_amd.parallel import dist as pdist  # noqa: E402
+
+PRESETS = {
+    # BASELINE config 2: LeNet CIFAR-10, 1k clients, bf16, 1 GPU
+    "lenet-1k": dict(model_name="lenet", model_kwargs={"num_classes": 10},
+                     clients_per_gpu=1000, num_classes=10, local_steps=2,
+                     batch_size=16, lr=0.05, dtype="bfloat16"),
+    # BASELINE config 3 (flagship): ResNet-18 CIFAR-100 non-IID, 10k @ 8 GPUs
+    "resnet-10k": dict(model_name="resnet18", model_kwargs={"num_classes": 100},
+                       clients_per_gpu=1250, num_classes=100, local_steps=2,
+                       batch_size=16, lr=0.05, dtype="bfloat16",
+                       dirichlet_alpha=0.1),
+    # BASELINE config 4: FedProx + deviceflow churn, 50k @ 8 GPUs
+    "fedprox-churn-50k": dict(
+        model_name="lenet", model_kwargs={"num_classes": 10},
+        clients_per_gpu=6250, num_classes=10, local_steps=2, batch_size=16,
+        lr=0.05, prox_mu=0.01, dtype="bfloat16",
+        behavior_strategy=json.dumps({
+            "offline_simulation": {"offline_probability": 0.05,
+                                   "spike_period": 5,
+                                   "spike_offline_fraction": 0.5},
+            "real_time_dispatch": {"use_strategy": True,
+                                   "drop_simulation": {"drop_probability": 0.02}},
+        })),
+    # BASELINE config 5: federated BERT-base next-word, 1k clients
+    "bert-base": dict(model_name="bert-base", model_kwargs={},
+                      clients_per_gpu=125, num_classes=0, local_steps=1,
+                      batch_size=4, lr=0.02, dtype="bfloat16",
+                      vocab_size=30522, seq_len=128),
+    # BASELINE config 1: CPU plumbing path
+    "mlp-cpu": dict(model_name="mlp", model_kwargs={}, clients_per_gpu=10,
+                    num_classes=10, local_steps=2, batch_size=8, lr=0.1,
+                    dtype="float32", force_cpu=True),
+}
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--preset", default="resnet-10k", choices=sorted(PRESETS))
+    ap.add_argument("--clients-per-gpu", type=int, default=0)
+    ap.add_argument("--chunk", type=int, default=0)
+    ap.add_argument("--local-steps", type=int, default=0)
+    ap.add_argument("--batch", type=int, default=0)
+    args = ap.parse_args()
+
+    preset = dict(PRESETS[args.preset])
+    force_cpu = preset.pop("force_cpu", False)
+    clients_per_gpu = args.clients_per_gpu or preset.pop("clients_per_gpu")
+    preset.pop("clients_per_gpu", None)
+    if args.local_steps:
+        preset["local_steps"] = args.local_steps
+    if args.batch:
+        preset["batch_size"] = args.batch
+
+    use_gpu = torch.cuda.is_available() and not force_cpu
+    ctx = pdist.init_distributed()
+    world = max(1, ctx.world_size)
+
+    job = EngineJob(
+        task_id=f"bench_{args.preset}",
+        clients=clients_per_gpu, cohort_size=0, rounds=args.steps,
+        chunk_clients=args.chunk, seed=1234,
+        device=ctx.device if use_gpu else "cpu",
+        dynamic_num=10 ** 9,  # never early-abort on simulated churn
+        **preset)
+    if not use_gpu:
+        job.device = "cpu"
+        job.dtype = "float32"
+        ctx = pdist.DistContext(rank=ctx.rank, world_size=ctx.world_size,
+                                local_rank=ctx.local_rank, device="cpu",
+                                backend=ctx.backend)
+
+    eng = LogicalEngine(job, dist_ctx=ctx if ctx.enabled else None)
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize(eng.device)
+
+    for r in range(args.warmup):
+        eng.run_round(r)
+    pdist.barrier()
+    sync()
+
+    t0 = time.perf_counter()
+    for r in range(args.warmup, args.warmup + args.steps):
+        eng.run_round(r)
+    pdist.barrier()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=eng.device if ctx.enabled and use_gpu else "cpu")
+    if ctx.enabled:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t[0])
+
+    clients_per_round = clients_per_gpu * world
+    value = clients_per_round * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if ctx.rank == 0:
+        out = {
+            "metric": "virtual clients/sec (simulated FL rounds, FedAvg CNN)",
+            "value": round(value, 2),
+            "unit": "clients/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": job.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": job.model_name,
+                "preset": args.preset,
+                "clients_total": clients_per_round,
+                "clients_per_gpu": clients_per_gpu,
+                "local_steps": job.local_steps,
+                "local_batch": job.batch_size,
+                "num_classes": job.num_classes,
+                "dirichlet_alpha": job.dirichlet_alpha,
+                "rounds_per_s": round(args.steps / elapsed, 4),
+                "parallelism": f"dp{world}",
+                "global_batch": clients_per_round * job.batch_size,
+                "seq_len": job.seq_len or None,
+            },
+        }
+        print(json.dumps(out))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

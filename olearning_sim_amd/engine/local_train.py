@@ -54,7 +54,8 @@ class LocalTrainer:
         return self.master.views_of(self._cast_flat)
 
     def train_chunk(self, client_ids: torch.Tensor, weights: torch.Tensor,
-                    round_idx: int, delta_flat: torch.Tensor) -> Dict[str, float]:
+                    round_idx: int, delta_flat: torch.Tensor,
+                    wsum: Optional[float] = None) -> Dict[str, float]:
         """Train one chunk of clients; accumulate weighted deltas.
 
         client_ids: [C] int64; weights: [C] fp32 aggregation weights
@@ -85,5 +86,6 @@ class LocalTrainer:
         with torch.no_grad():
             fused.weighted_delta_accum_flat(
                 delta_flat, buf.detach(), self._cast_flat, weights, C,
-                offsets=self.master.offsets)
+                offsets=self.master.offsets,
+                wsum=wsum if wsum is not None else float(weights.sum()))
         return {"loss": last_loss, "clients": C}

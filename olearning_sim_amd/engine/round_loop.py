@@ -88,16 +88,14 @@ class LogicalEngine:
             return min(self.job.chunk_clients, cohort)
         if self.device.type != "cuda":
             return min(cohort, 64)
-        # auto: bound replica+grad+activation memory to a fraction of HBM.
+        # auto: bound replica+grad+activation memory to a fraction of HBM
         p = self.master.numel()
         bytes_per_client = p * self.dtype.itemsize * 2  # weights + grad
-        # activations per client (rough: 40x input bytes per local batch)
-        act = 1
-        for s in self.model.input_shape:
-            act *= s
-        bytes_per_client += self.job.batch_size * act * self.dtype.itemsize * 40
+        bytes_per_client += (self.job.batch_size
+                             * self.model.act_elems_per_sample
+                             * self.dtype.itemsize)
         free = torch.cuda.get_device_properties(self.device).total_memory
-        budget = int(free * 0.55)
+        budget = int(free * 0.5)
         return max(1, min(cohort, budget // max(1, bytes_per_client)))
 
     def select_cohort(self, round_idx: int) -> torch.Tensor:
@@ -134,8 +132,10 @@ class LogicalEngine:
         chunk = self._chunk_size(max(1, int(active_ids.numel())))
         for lo in range(0, int(active_ids.numel()), chunk):
             cid = active_ids[lo:lo + chunk].to(self.device)
-            w = weights_all[lo:lo + chunk].to(self.device)
-            stats = self.trainer.train_chunk(cid, w, round_idx, self._delta)
+            w_cpu = weights_all[lo:lo + chunk]
+            w = w_cpu.to(self.device)
+            stats = self.trainer.train_chunk(cid, w, round_idx, self._delta,
+                                             wsum=float(w_cpu.sum()))
             trained += stats["clients"]
             if stats["loss"]:
                 losses.append(stats["loss"])

@@ -35,6 +35,9 @@ class ClientBatchedModel:
     num_classes: int = 10
     input_shape = (1,)          # per-sample shape (no batch dims)
     sequence_model: bool = False
+    # rough activation elements held live per sample during fwd+bwd
+    # (sizes the engine's auto chunking; override per model)
+    act_elems_per_sample: int = 4096
 
     # -- parameter management -------------------------------------------
     def param_shapes(self) -> Dict[str, tuple]:

@@ -14,6 +14,7 @@ from .base import (ClientBatchedModel, Params, binit, kaiming,
 class LeNet5(ClientBatchedModel):
     name = "lenet"
     num_classes = 10
+    act_elems_per_sample = 40_000
     input_shape = (3, 32, 32)
 
     def __init__(self, in_ch: int = 3, num_classes: int = 10):

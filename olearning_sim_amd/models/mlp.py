@@ -13,6 +13,7 @@ from .base import ClientBatchedModel, Params, binit, blinear
 class MLP(ClientBatchedModel):
     name = "mlp"
     num_classes = 10
+    act_elems_per_sample = 2048
     input_shape = (784,)
 
     def __init__(self, in_features: int = 784, hidden: int = 200,

@@ -25,6 +25,7 @@ _GN_GROUPS = 8
 class ResNet18(ClientBatchedModel):
     name = "resnet18"
     num_classes = 100
+    act_elems_per_sample = 1_600_000
     input_shape = (3, 32, 32)
 
     def __init__(self, num_classes: int = 100, in_ch: int = 3,

@@ -1,0 +1,71 @@
+// Weighted client-delta reduction: the FedAvg aggregation the reference
+// hands to an external service via its Pulsar outbound
+// (ols_core/deviceflow/non_grpc/dispatcher.py:47-63) runs here as one
+// on-GPU reduction over the co-resident clients:
+//
+//   delta[j] += sum_c w[c] * (buf[c,j] - master[j])
+//            =  sum_c w[c] * buf[blk_base + c*n + q]  -  W * master[j]
+//
+// (W = sum_c w[c], hoisted).  Parallel over master elements j; for a
+// fixed c, consecutive threads read consecutive addresses, so each step
+// of the c-loop is a fully coalesced stream; client weights are staged
+// in LDS once per workgroup.  fp32 accumulation regardless of the
+// replica dtype.
+
+#include "common.h"
+
+#define MAX_LDS_W 4096
+
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_delta_accum(
+    float* __restrict__ delta, const T* __restrict__ buf,
+    const T* __restrict__ master, const float* __restrict__ weights,
+    const int64_t* __restrict__ offs, int nblocks, int64_t clients,
+    int64_t pglobal, float wsum) {
+  __shared__ float w_lds[MAX_LDS_W];
+  const bool w_in_lds = clients <= MAX_LDS_W;
+  if (w_in_lds) {
+    for (int c = threadIdx.x; c < clients; c += blockDim.x)
+      w_lds[c] = weights[c];
+    __syncthreads();
+  }
+  const float* w = w_in_lds ? w_lds : weights;
+
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       j < pglobal; j += stride) {
+    int b = find_block(offs, nblocks, j);
+    int64_t n = offs[b + 1] - offs[b];
+    int64_t q = j - offs[b];
+    const T* col = buf + clients * offs[b] + q;
+    float acc = 0.f;
+    for (int64_t c = 0; c < clients; ++c)
+      acc += w[c] * to_f32(col[c * n]);
+    delta[j] += acc - wsum * to_f32(master[j]);
+  }
+}
+
+extern "C" void ols_weighted_delta_accum_flat(
+    float* delta, const void* buf, const void* master, const float* weights,
+    const int64_t* offs, int nblocks, int64_t clients, int64_t pglobal,
+    float wsum, int dtype, hipStream_t stream) {
+  dim3 grid(ols_grid(pglobal, OLS_THREADS));
+  dim3 block(OLS_THREADS);
+  switch (dtype) {
+    case 0:
+      hipLaunchKernelGGL((k_delta_accum<float>), grid, block, 0, stream,
+                         delta, (const float*)buf, (const float*)master,
+                         weights, offs, nblocks, clients, pglobal, wsum);
+      break;
+    case 1:
+      hipLaunchKernelGGL((k_delta_accum<__hip_bfloat16>), grid, block, 0,
+                         stream, delta, (const __hip_bfloat16*)buf,
+                         (const __hip_bfloat16*)master, weights, offs,
+                         nblocks, clients, pglobal, wsum);
+      break;
+    default:
+      hipLaunchKernelGGL((k_delta_accum<__half>), grid, block, 0, stream,
+                         delta, (const __half*)buf, (const __half*)master,
+                         weights, offs, nblocks, clients, pglobal, wsum);
+  }
+}

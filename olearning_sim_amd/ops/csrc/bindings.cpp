@@ -1,0 +1,103 @@
+// Torch bindings for the olearning_sim_amd gfx950 kernels.
+// Registered as torch.ops.olsim_hip.* (loaded from the in-tree .so by
+// olearning_sim_amd/ops/fused.py).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+extern "C" void ols_fused_sgd_update_flat(
+    void* buf, const void* grad, const void* master, const int64_t* offs,
+    int nblocks, int64_t clients, int64_t total, float lr, float mu,
+    int dtype, hipStream_t stream);
+
+extern "C" void ols_weighted_delta_accum_flat(
+    float* delta, const void* buf, const void* master, const float* weights,
+    const int64_t* offs, int nblocks, int64_t clients, int64_t pglobal,
+    float wsum, int dtype, hipStream_t stream);
+
+extern "C" void ols_cross_entropy_fwd_bwd(
+    const void* logits, const int64_t* labels, float* loss, void* dlogits,
+    int64_t nrows, int64_t K, float inv_n, int dtype, hipStream_t stream);
+
+namespace {
+
+int dtype_code(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return 0;
+    case at::kBFloat16: return 1;
+    case at::kHalf: return 2;
+    default:
+      TORCH_CHECK(false, "olsim_hip: unsupported dtype ", t.scalar_type());
+  }
+}
+
+void fused_sgd_update_flat(at::Tensor buf, at::Tensor grad,
+                           at::Tensor global_flat, at::Tensor offsets,
+                           int64_t clients, double lr, double mu) {
+  TORCH_CHECK(buf.is_cuda() && buf.is_contiguous(), "buf must be GPU+contig");
+  TORCH_CHECK(grad.sizes() == buf.sizes() && grad.scalar_type() == buf.scalar_type());
+  const bool prox = mu != 0.0 && global_flat.numel() > 0;
+  if (prox) {
+    TORCH_CHECK(offsets.numel() >= 2 && offsets.scalar_type() == at::kLong);
+    TORCH_CHECK(global_flat.scalar_type() == buf.scalar_type());
+  }
+  auto stream = at::cuda::getCurrentCUDAStream();
+  ols_fused_sgd_update_flat(
+      buf.data_ptr(), grad.data_ptr(),
+      prox ? global_flat.data_ptr() : nullptr,
+      prox ? offsets.data_ptr<int64_t>() : nullptr,
+      prox ? (int)(offsets.numel() - 1) : 0, clients, buf.numel(),
+      (float)lr, (float)mu, dtype_code(buf), stream.stream());
+}
+
+void weighted_delta_accum_flat(at::Tensor delta, at::Tensor buf,
+                               at::Tensor global_flat, at::Tensor weights,
+                               at::Tensor offsets, int64_t clients,
+                               double wsum) {
+  TORCH_CHECK(delta.is_cuda() && delta.scalar_type() == at::kFloat &&
+              delta.is_contiguous());
+  TORCH_CHECK(buf.is_contiguous() && global_flat.is_contiguous());
+  TORCH_CHECK(buf.scalar_type() == global_flat.scalar_type());
+  TORCH_CHECK(weights.scalar_type() == at::kFloat && weights.numel() == clients);
+  TORCH_CHECK(offsets.scalar_type() == at::kLong && offsets.numel() >= 2);
+  TORCH_CHECK(buf.numel() == clients * global_flat.numel());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  ols_weighted_delta_accum_flat(
+      delta.data_ptr<float>(), buf.data_ptr(), global_flat.data_ptr(),
+      weights.data_ptr<float>(), offsets.data_ptr<int64_t>(),
+      (int)(offsets.numel() - 1), clients, global_flat.numel(), (float)wsum,
+      dtype_code(buf), stream.stream());
+}
+
+std::tuple<at::Tensor, at::Tensor> cross_entropy_fwd_bwd(at::Tensor logits,
+                                                         at::Tensor labels) {
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 && logits.is_contiguous());
+  TORCH_CHECK(labels.scalar_type() == at::kLong &&
+              labels.numel() == logits.size(0));
+  int64_t n = logits.size(0), k = logits.size(1);
+  auto loss = at::empty({n}, logits.options().dtype(at::kFloat));
+  auto dlogits = at::empty_like(logits);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  ols_cross_entropy_fwd_bwd(
+      logits.data_ptr(), labels.contiguous().data_ptr<int64_t>(),
+      loss.data_ptr<float>(), dlogits.data_ptr(), n, k, 1.0f / (float)n,
+      dtype_code(logits), stream.stream());
+  return {loss, dlogits};
+}
+
+}  // namespace
+
+TORCH_LIBRARY(olsim_hip, m) {
+  m.def("fused_sgd_update_flat(Tensor(a!) buf, Tensor grad, Tensor global_flat, "
+        "Tensor offsets, int clients, float lr, float mu) -> ()");
+  m.def("weighted_delta_accum_flat(Tensor(a!) delta, Tensor buf, "
+        "Tensor global_flat, Tensor weights, Tensor offsets, int clients, float wsum) -> ()");
+  m.def("cross_entropy_fwd_bwd(Tensor logits, Tensor labels) -> (Tensor, Tensor)");
+}
+
+TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
+  m.impl("fused_sgd_update_flat", &fused_sgd_update_flat);
+  m.impl("weighted_delta_accum_flat", &weighted_delta_accum_flat);
+  m.impl("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd);
+}

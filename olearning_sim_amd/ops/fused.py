@@ -77,12 +77,15 @@ def fused_sgd_update(params: Sequence[torch.Tensor],
     """
     if not params:
         return
-    ops = _gpu_ops(params[0])
-    if ops is not None:
-        gl: List[torch.Tensor] = (
-            [g.reshape(-1) for g in global_params] if (global_params is not None and mu != 0.0)
-            else [torch.empty(0, dtype=params[0].dtype, device=params[0].device)] * len(params))
-        ops.fused_sgd_update(list(params), list(grads), gl, float(lr), float(mu))
+    if params[0].is_cuda:
+        for i, (w, g) in enumerate(zip(params, grads)):
+            n = w[0].numel() if w.dim() > 1 else w.numel()
+            gl = (global_params[i].reshape(-1) if (global_params is not None
+                                                  and mu != 0.0)
+                  else torch.empty(0, dtype=w.dtype, device=w.device))
+            offs = torch.tensor([0, n], dtype=torch.int64, device=w.device)
+            fused_sgd_update_flat(w.reshape(-1), g.reshape(-1), gl,
+                                  w.shape[0], lr, mu, offs)
         return
     with torch.no_grad():
         for i, (w, g) in enumerate(zip(params, grads)):
@@ -99,10 +102,14 @@ def weighted_delta_accum(delta: Sequence[torch.Tensor],
     """delta[i] (fp32, shape [...]) += sum_c weights[c] * (client_params[i][c] - global[i])."""
     if not delta:
         return
-    ops = _gpu_ops(client_params[0])
-    if ops is not None:
-        ops.weighted_delta_accum(list(delta), list(client_params),
-                                 list(global_params), weights.float())
+    if client_params[0].is_cuda:
+        wsum = float(weights.sum())
+        for dl, cw, gw in zip(delta, client_params, global_params):
+            offs = torch.tensor([0, gw.numel()], dtype=torch.int64,
+                                device=cw.device)
+            weighted_delta_accum_flat(dl.reshape(-1), cw.reshape(-1),
+                                      gw.reshape(-1), weights, cw.shape[0],
+                                      offs, wsum)
         return
     with torch.no_grad():
         for dl, cw, gw in zip(delta, client_params, global_params):
@@ -160,12 +167,16 @@ def fused_sgd_update_flat(buf: torch.Tensor, grad: torch.Tensor,
 def weighted_delta_accum_flat(delta: torch.Tensor, buf: torch.Tensor,
                               global_flat: torch.Tensor,
                               weights: torch.Tensor, clients: int,
-                              offsets: Optional[torch.Tensor] = None) -> None:
+                              offsets: Optional[torch.Tensor] = None,
+                              wsum: Optional[float] = None) -> None:
     """delta[j] += sum_c weights[c] * (buf[c,j] - global_flat[j]) per block."""
     if buf.is_cuda:
         ops = load_hip_ops(required=True)
+        if wsum is None:
+            wsum = float(weights.sum())
         ops.weighted_delta_accum_flat(delta, buf, global_flat,
-                                      weights.float(), offsets, int(clients))
+                                      weights.float(), offsets, int(clients),
+                                      float(wsum))
         return
     with torch.no_grad():
         offs = offsets.tolist()
@@ -185,9 +196,10 @@ class _CrossEntropyFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits: torch.Tensor, labels: torch.Tensor):
         ops = load_hip_ops(required=True)
-        loss, dlogits = ops.cross_entropy_fwd_bwd(logits, labels)
+        row_loss, dlogits = ops.cross_entropy_fwd_bwd(
+            logits.contiguous(), labels)
         ctx.save_for_backward(dlogits)
-        return loss
+        return row_loss.mean()
 
     @staticmethod
     def backward(ctx, grad_out):
