@@ -71,27 +71,29 @@ class ResNet18(ClientBatchedModel):
     # ------------------------------------------------------------------
     def _block(self, params: Params, x: torch.Tensor, C: int, pre: str,
                stride: int, has_down: bool) -> torch.Tensor:
+        from ..ops.fused import groupnorm_act
         h = bconv2d(x, params[f"{pre}.c1.w"], C, stride=stride, padding=1)
-        h = F.relu(bgroupnorm(h, C, _GN_GROUPS,
-                              params[f"{pre}.gn1.g"], params[f"{pre}.gn1.b"]))
+        h = groupnorm_act(h, C, _GN_GROUPS, params[f"{pre}.gn1.g"],
+                          params[f"{pre}.gn1.b"], relu=True)
         h = bconv2d(h, params[f"{pre}.c2.w"], C, stride=1, padding=1)
-        h = bgroupnorm(h, C, _GN_GROUPS,
-                       params[f"{pre}.gn2.g"], params[f"{pre}.gn2.b"])
         if has_down:
             sc = bconv2d(x, params[f"{pre}.down.w"], C, stride=stride)
-            sc = bgroupnorm(sc, C, _GN_GROUPS,
-                            params[f"{pre}.gndown.g"], params[f"{pre}.gndown.b"])
+            sc = groupnorm_act(sc, C, _GN_GROUPS, params[f"{pre}.gndown.g"],
+                               params[f"{pre}.gndown.b"])
         else:
             sc = x
-        return F.relu(h + sc)
+        # fused: relu(gn(h) + sc)
+        return groupnorm_act(h, C, _GN_GROUPS, params[f"{pre}.gn2.g"],
+                             params[f"{pre}.gn2.b"], res=sc, relu=True)
 
     def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.fused import groupnorm_act
         # x: [C, B, 3, 32, 32]
         C, B = x.shape[0], x.shape[1]
         h = x.permute(1, 0, 2, 3, 4).reshape(B, C * self.in_ch, 32, 32)
         h = bconv2d(h, params["stem.w"], C, stride=1, padding=1)
-        h = F.relu(bgroupnorm(h, C, _GN_GROUPS,
-                              params["stem.gn.g"], params["stem.gn.b"]))
+        h = groupnorm_act(h, C, _GN_GROUPS, params["stem.gn.g"],
+                          params["stem.gn.b"], relu=True)
         for s in range(4):
             stride = 1 if s == 0 else 2
             has_down = s > 0

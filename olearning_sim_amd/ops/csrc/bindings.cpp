@@ -20,6 +20,20 @@ extern "C" void ols_cross_entropy_fwd_bwd(
     const void* logits, const int64_t* labels, float* loss, void* dlogits,
     int64_t nrows, int64_t K, float inv_n, int dtype, hipStream_t stream);
 
+extern "C" void ols_groupnorm_fwd(const void* x, const void* res, void* y,
+                                  float* mean, float* rstd, const void* gamma,
+                                  const void* beta, int B, int C, int ch,
+                                  int G, int HW, float eps, bool relu,
+                                  int dtype, hipStream_t stream);
+
+extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
+                                  const void* dy, void* dx, void* dres,
+                                  const float* mean, const float* rstd,
+                                  const void* gamma, float* dgamma,
+                                  float* dbeta, int B, int C, int ch, int G,
+                                  int HW, bool relu, int dtype,
+                                  hipStream_t stream);
+
 namespace {
 
 int dtype_code(const at::Tensor& t) {
@@ -86,6 +100,62 @@ std::tuple<at::Tensor, at::Tensor> cross_entropy_fwd_bwd(at::Tensor logits,
   return {loss, dlogits};
 }
 
+int gn_dtype(const at::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == at::kFloat || t.scalar_type() == at::kBFloat16,
+              "groupnorm: f32/bf16 only");
+  return t.scalar_type() == at::kFloat ? 0 : 1;
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> groupnorm_fwd(
+    at::Tensor x, at::Tensor res, at::Tensor gamma, at::Tensor beta,
+    int64_t clients, int64_t groups, double eps, bool relu) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
+  const int64_t B = x.size(0), cch = x.size(1),
+                HW = x.size(2) * x.size(3);
+  const int64_t ch = cch / clients;
+  TORCH_CHECK(cch % clients == 0 && ch % groups == 0);
+  TORCH_CHECK(ch / groups <= 128, "groupnorm: ch/G > 128 unsupported");
+  TORCH_CHECK(gamma.is_contiguous() && beta.is_contiguous());
+  const bool has_res = res.numel() > 0;
+  if (has_res) TORCH_CHECK(res.is_contiguous() && res.sizes() == x.sizes());
+  auto y = at::empty_like(x);
+  const int64_t ngroups = B * clients * groups;
+  auto mean = at::empty({ngroups}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({ngroups}, x.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  ols_groupnorm_fwd(x.data_ptr(), has_res ? res.data_ptr() : nullptr,
+                    y.data_ptr(), mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), gamma.data_ptr(), beta.data_ptr(),
+                    (int)B, (int)clients, (int)ch, (int)groups, (int)HW,
+                    (float)eps, relu, gn_dtype(x), stream.stream());
+  return {y, mean, rstd};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> groupnorm_bwd(
+    at::Tensor x, at::Tensor y, at::Tensor dy, at::Tensor mean,
+    at::Tensor rstd, at::Tensor gamma, int64_t clients, int64_t groups,
+    bool has_res, bool relu) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && y.is_contiguous());
+  auto dyc = dy.contiguous();
+  const int64_t B = x.size(0), cch = x.size(1),
+                HW = x.size(2) * x.size(3);
+  const int64_t ch = cch / clients;
+  auto dx = at::empty_like(x);
+  auto dres = has_res ? at::empty_like(x)
+                      : at::empty({0}, x.options());
+  auto dgamma = at::zeros({clients, ch}, x.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({clients, ch}, x.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  ols_groupnorm_bwd(x.data_ptr(), y.data_ptr(), dyc.data_ptr(),
+                    dx.data_ptr(), has_res ? dres.data_ptr() : nullptr,
+                    mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                    gamma.data_ptr(), dgamma.data_ptr<float>(),
+                    dbeta.data_ptr<float>(), (int)B, (int)clients, (int)ch,
+                    (int)groups, (int)HW, relu, gn_dtype(x),
+                    stream.stream());
+  return {dx, dres, dgamma, dbeta};
+}
+
 }  // namespace
 
 TORCH_LIBRARY(olsim_hip, m) {
@@ -94,10 +164,17 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("weighted_delta_accum_flat(Tensor(a!) delta, Tensor buf, "
         "Tensor global_flat, Tensor weights, Tensor offsets, int clients, float wsum) -> ()");
   m.def("cross_entropy_fwd_bwd(Tensor logits, Tensor labels) -> (Tensor, Tensor)");
+  m.def("groupnorm_fwd(Tensor x, Tensor res, Tensor gamma, Tensor beta, "
+        "int clients, int groups, float eps, bool relu) -> (Tensor, Tensor, Tensor)");
+  m.def("groupnorm_bwd(Tensor x, Tensor y, Tensor dy, Tensor mean, "
+        "Tensor rstd, Tensor gamma, int clients, int groups, bool has_res, "
+        "bool relu) -> (Tensor, Tensor, Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("fused_sgd_update_flat", &fused_sgd_update_flat);
   m.impl("weighted_delta_accum_flat", &weighted_delta_accum_flat);
   m.impl("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd);
+  m.impl("groupnorm_fwd", &groupnorm_fwd);
+  m.impl("groupnorm_bwd", &groupnorm_bwd);
 }

@@ -212,3 +212,50 @@ def cross_entropy_fwd_bwd(logits: torch.Tensor, labels: torch.Tensor) -> torch.T
     if logits.is_cuda:
         return _CrossEntropyFn.apply(logits, labels)
     return torch.nn.functional.cross_entropy(logits.float(), labels)
+
+
+class _GroupNormActFn(torch.autograd.Function):
+    """Fused per-client GroupNorm (+residual +ReLU) on channel-grouped
+    activations [B, C*ch, H, W] (ops/csrc/groupnorm.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, res, gamma, beta, clients, groups, eps, relu):
+        ops = load_hip_ops(required=True)
+        has_res = res is not None
+        res_in = res.contiguous() if has_res else torch.empty(0, dtype=x.dtype,
+                                                              device=x.device)
+        y, mean, rstd = ops.groupnorm_fwd(x.contiguous(), res_in, gamma,
+                                          beta, clients, groups, eps, relu)
+        ctx.save_for_backward(x, y, mean, rstd, gamma)
+        ctx.meta = (clients, groups, has_res, relu)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, mean, rstd, gamma = ctx.saved_tensors
+        clients, groups, has_res, relu = ctx.meta
+        ops = load_hip_ops(required=True)
+        dx, dres, dgamma, dbeta = ops.groupnorm_bwd(
+            x, y, dy, mean, rstd, gamma, clients, groups, has_res, relu)
+        return (dx, dres.view_as(x) if has_res else None,
+                dgamma.to(gamma.dtype), dbeta.to(gamma.dtype),
+                None, None, None, None)
+
+
+def groupnorm_act(x: torch.Tensor, clients: int, groups: int,
+                  gamma: torch.Tensor, beta: torch.Tensor,
+                  res: Optional[torch.Tensor] = None, relu: bool = False,
+                  eps: float = 1e-5) -> torch.Tensor:
+    """y = relu?(gn(x; clients, groups)*gamma + beta [+ res]).
+
+    GPU: one fused HIP kernel each way.  CPU: composed torch ops
+    (models/base.bgroupnorm semantics)."""
+    if x.is_cuda:
+        return _GroupNormActFn.apply(x, res, gamma.contiguous(),
+                                     beta.contiguous(), clients, groups,
+                                     eps, relu)
+    from ..models.base import bgroupnorm
+    y = bgroupnorm(x, clients, groups, gamma, beta, eps)
+    if res is not None:
+        y = y + res
+    return torch.nn.functional.relu(y) if relu else y

@@ -138,3 +138,45 @@ def test_gpu_engine_uses_hip_ops(monkeypatch):
                     lr=0.1, device="cuda:0", dtype="bfloat16", num_classes=4)
     LogicalEngine(job).run_round(0)
     assert calls["n"] > 0
+
+
+@pytest.mark.parametrize("shape", [(4, 5, 64, 8, 8), (3, 7, 128, 16, 16)])
+@pytest.mark.parametrize("relu,res", [(False, False), (True, False), (True, True)])
+def test_groupnorm_fused_matches_reference(shape, relu, res):
+    from olearning_sim_amd.ops.fused import groupnorm_act
+    from olearning_sim_amd.models.base import bgroupnorm
+    B, C, ch, H, W = shape
+    G = 8
+    g = torch.Generator().manual_seed(0)
+    x = torch.randn(B, C * ch, H, W, generator=g).cuda().to(torch.bfloat16)
+    gamma = (1 + 0.1 * torch.randn(C, ch, generator=g)).cuda().to(torch.bfloat16)
+    beta = (0.1 * torch.randn(C, ch, generator=g)).cuda().to(torch.bfloat16)
+    resid = (torch.randn(B, C * ch, H, W, generator=g).cuda().to(torch.bfloat16)
+             if res else None)
+
+    xf = x.detach().clone().requires_grad_(True)
+    gf = gamma.detach().clone().requires_grad_(True)
+    bf = beta.detach().clone().requires_grad_(True)
+    rf = resid.detach().clone().requires_grad_(True) if res else None
+    y = groupnorm_act(xf, C, G, gf, bf, res=rf, relu=relu)
+
+    # fp32 torch reference
+    xr = x.detach().float().requires_grad_(True)
+    gr = gamma.detach().float().requires_grad_(True)
+    br = beta.detach().float().requires_grad_(True)
+    rr = resid.detach().float().requires_grad_(True) if res else None
+    yr = bgroupnorm(xr, C, G, gr, br)
+    if res:
+        yr = yr + rr
+    if relu:
+        yr = torch.relu(yr)
+    torch.testing.assert_close(y.float(), yr, atol=6e-2, rtol=6e-2)
+
+    dy = torch.randn(y.shape, generator=g).cuda()
+    y.backward(dy.to(y.dtype))
+    yr.backward(dy)
+    torch.testing.assert_close(xf.grad.float(), xr.grad, atol=1e-1, rtol=1e-1)
+    torch.testing.assert_close(gf.grad.float(), gr.grad, atol=2e-1, rtol=5e-2)
+    torch.testing.assert_close(bf.grad.float(), br.grad, atol=2e-1, rtol=5e-2)
+    if res:
+        torch.testing.assert_close(rf.grad.float(), rr.grad, atol=1e-1, rtol=1e-1)
