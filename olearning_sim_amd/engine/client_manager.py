@@ -74,6 +74,29 @@ class FlatParams:
             self.views[k].copy_(v.float())
 
 
+def replicate_params(cast_params: Params, clients: int) -> Params:
+    """Per-parameter [C, ...] leaf replicas.
+
+    Separate leaves (not views of one flat buffer) so autograd writes
+    each gradient straight into its own tensor — a single flat leaf
+    makes every view's backward materialise a full-size zeros + copy +
+    accumulate chain, which measured ~50% of a ResNet round
+    (profiles/resnet_round_r01.md).  The fused update/delta kernels
+    take the tensor list (one small launch per parameter).
+    """
+    out: Params = {}
+    for k, v in cast_params.items():
+        # clone, not contiguous(): for clients==1 an expand of a
+        # contiguous tensor is already contiguous and .contiguous()
+        # would RETURN THE SAME STORAGE — the "replica" would alias the
+        # global master and local training would corrupt it in place.
+        rep = v.detach().unsqueeze(0).expand(clients, *v.shape) \
+               .clone(memory_format=torch.contiguous_format)
+        rep.requires_grad_(True)
+        out[k] = rep
+    return out
+
+
 def replicate_flat(cast_params: Params, clients: int) -> torch.Tensor:
     """Build ONE flat replica buffer holding every client's weights.
 

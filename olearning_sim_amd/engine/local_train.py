@@ -22,7 +22,7 @@ import torch
 
 from ..models.base import ClientBatchedModel, Params
 from ..ops import fused
-from .client_manager import FlatParams, replicate_flat, batched_views
+from .client_manager import FlatParams, replicate_params
 from .data import SyntheticFederatedData
 
 
@@ -64,28 +64,31 @@ class LocalTrainer:
         """
         C = int(client_ids.numel())
         cast = self._cast_params()
-        buf = replicate_flat(cast, C)
+        params = replicate_params(cast, C)
+        plist = list(params.values())
+        glist = [cast[k] for k in params]
         last_loss = 0.0
         for step in range(self.local_steps):
-            params = batched_views(buf, self.master.shapes, C)
             x, y = self.data.batch(client_ids, round_idx, step,
                                    self.batch_size, self.dtype)
             loss = self.model.loss(params, x, y)
             # loss is the mean over C*B rows; each client's SGD step needs
             # the gradient of ITS OWN per-client mean, i.e. d(loss*C)/dw_c
             # — this also makes training invariant to the chunking.
-            grad, = torch.autograd.grad(loss * C, [buf])
+            grads = torch.autograd.grad(loss * C, plist, allow_unused=True)
             with torch.no_grad():
-                fused.fused_sgd_update_flat(
-                    buf, grad, self._cast_flat, C,
+                live = [(p, g) for p, g in zip(plist, grads) if g is not None]
+                fused.fused_sgd_update(
+                    [p for p, _ in live], [g for _, g in live],
                     lr=self.lr, mu=self.prox_mu,
-                    offsets=self.master.offsets)
+                    global_params=[cast[k] for k, g in
+                                   zip(params, grads) if g is not None])
             if self.report_loss and step == self.local_steps - 1:
                 last_loss = float(loss.detach())
-            del grad
+            del grads
         with torch.no_grad():
-            fused.weighted_delta_accum_flat(
-                delta_flat, buf.detach(), self._cast_flat, weights, C,
-                offsets=self.master.offsets,
+            delta_views = list(self.master.views_of(delta_flat).values())
+            fused.weighted_delta_accum(
+                delta_views, [p.detach() for p in plist], glist, weights,
                 wsum=wsum if wsum is not None else float(weights.sum()))
         return {"loss": last_loss, "clients": C}

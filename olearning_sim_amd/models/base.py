@@ -53,8 +53,10 @@ class ClientBatchedModel:
         """Clone the global model into [C, ...] per-client leaves."""
         out: Params = {}
         for k, v in global_params.items():
+            # clone (not .contiguous()): at clients==1 the expand is
+            # already contiguous and would alias the source storage
             rep = v.detach().to(dtype).unsqueeze(0).expand(
-                clients, *v.shape).contiguous()
+                clients, *v.shape).clone(memory_format=torch.contiguous_format)
             rep.requires_grad_(True)
             out[k] = rep
         return out

@@ -98,12 +98,14 @@ def fused_sgd_update(params: Sequence[torch.Tensor],
 def weighted_delta_accum(delta: Sequence[torch.Tensor],
                          client_params: Sequence[torch.Tensor],
                          global_params: Sequence[torch.Tensor],
-                         weights: torch.Tensor) -> None:
+                         weights: torch.Tensor,
+                         wsum: Optional[float] = None) -> None:
     """delta[i] (fp32, shape [...]) += sum_c weights[c] * (client_params[i][c] - global[i])."""
     if not delta:
         return
     if client_params[0].is_cuda:
-        wsum = float(weights.sum())
+        if wsum is None:
+            wsum = float(weights.sum())
         for dl, cw, gw in zip(delta, client_params, global_params):
             offs = torch.tensor([0, gw.numel()], dtype=torch.int64,
                                 device=cw.device)
