@@ -1,0 +1,135 @@
+"""Client-batched BERT for federated next-word prediction
+(BASELINE config 5: federated BERT-base, 1k clients, bf16 MFMA GEMMs).
+
+Every client owns a full transformer; weights carry a leading client
+dim [C, ...] and all dense math runs as client-batched GEMMs
+(torch.bmm -> hipBLASLt MFMA kernels on gfx950).  Attention folds the
+client dim into the batch dim of scaled_dot_product_attention.  The LM
+head predicts the next token (causal mask), CE via the fused HIP kernel.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from .base import (ClientBatchedModel, Params, binit, blinear, blayernorm)
+
+
+class BertLM(ClientBatchedModel):
+    name = "bert"
+    sequence_model = True
+
+    def __init__(self, vocab_size: int = 30522, hidden: int = 768,
+                 layers: int = 12, heads: int = 12, seq_len: int = 128,
+                 intermediate: Optional[int] = None):
+        self.vocab_size = vocab_size
+        self.hidden = hidden
+        self.layers = layers
+        self.heads = heads
+        self.seq_len = seq_len
+        self.intermediate = intermediate or hidden * 4
+        self.num_classes = vocab_size
+        self.input_shape = (seq_len,)
+        self.act_elems_per_sample = seq_len * hidden * layers * 10
+
+    # ------------------------------------------------------------------
+    def init_global(self, device="cpu", dtype=torch.float32,
+                    generator: Optional[torch.Generator] = None) -> Params:
+        g = generator
+        h, inter, v = self.hidden, self.intermediate, self.vocab_size
+        p: Params = {}
+
+        def lin(name, out_f, in_f):
+            p[f"{name}.w"] = binit((out_f, in_f), in_f, device, dtype, g)
+            p[f"{name}.b"] = binit((out_f,), in_f, device, dtype, g)
+
+        def ln(name):
+            p[f"{name}.g"] = torch.ones(h, device=device, dtype=dtype)
+            p[f"{name}.b"] = torch.zeros(h, device=device, dtype=dtype)
+
+        emb = torch.empty(v, h, device=device, dtype=torch.float32)
+        emb.normal_(0, 0.02, generator=g)
+        p["emb.tok"] = emb.to(dtype)
+        pos = torch.empty(self.seq_len, h, device=device, dtype=torch.float32)
+        pos.normal_(0, 0.02, generator=g)
+        p["emb.pos"] = pos.to(dtype)
+        ln("emb.ln")
+        for i in range(self.layers):
+            pre = f"l{i}"
+            lin(f"{pre}.qkv", 3 * h, h)
+            lin(f"{pre}.attn_out", h, h)
+            ln(f"{pre}.ln1")
+            lin(f"{pre}.ffn_in", inter, h)
+            lin(f"{pre}.ffn_out", h, inter)
+            ln(f"{pre}.ln2")
+        # LM head ties to the token embedding (BERT-style decoder bias)
+        p["head.bias"] = torch.zeros(v, device=device, dtype=dtype)
+        return p
+
+    # ------------------------------------------------------------------
+    def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        # x: [C, B, L] token ids
+        C, B, L = x.shape
+        h, nh = self.hidden, self.heads
+        hd = h // nh
+        tok = params["emb.tok"]                    # [C, V, H]
+        flat_ids = (x + torch.arange(C, device=x.device)
+                    .view(C, 1, 1) * self.vocab_size).reshape(-1)
+        emb = tok.reshape(C * self.vocab_size, h)[flat_ids].view(C, B, L, h)
+        emb = emb + params["emb.pos"][:, :L].unsqueeze(1)
+        hs = blayernorm(emb, params["emb.ln.g"], params["emb.ln.b"])
+
+        causal = torch.ones(L, L, dtype=torch.bool, device=x.device).tril()
+        for i in range(self.layers):
+            pre = f"l{i}"
+            qkv = blinear(hs.view(C, B * L, h), params[f"{pre}.qkv.w"],
+                          params[f"{pre}.qkv.b"])          # [C, B*L, 3H]
+            qkv = qkv.view(C * B, L, 3, nh, hd).permute(2, 0, 3, 1, 4)
+            q, k, v = qkv[0], qkv[1], qkv[2]               # [C*B, nh, L, hd]
+            att = F.scaled_dot_product_attention(q, k, v, attn_mask=causal)
+            att = att.transpose(1, 2).reshape(C, B * L, h)
+            hs = hs + blinear(att, params[f"{pre}.attn_out.w"],
+                              params[f"{pre}.attn_out.b"]).view(C, B, L, h)
+            hs = blayernorm(hs, params[f"{pre}.ln1.g"], params[f"{pre}.ln1.b"])
+            ff = F.gelu(blinear(hs.view(C, B * L, h),
+                                params[f"{pre}.ffn_in.w"],
+                                params[f"{pre}.ffn_in.b"]))
+            hs = hs + blinear(ff, params[f"{pre}.ffn_out.w"],
+                              params[f"{pre}.ffn_out.b"]).view(C, B, L, h)
+            hs = blayernorm(hs, params[f"{pre}.ln2.g"], params[f"{pre}.ln2.b"])
+
+        # tied LM head: logits = hs @ emb^T + bias  -> [C, B, L, V]
+        logits = torch.bmm(hs.view(C, B * L, h), tok.transpose(1, 2))
+        logits = logits + params["head.bias"].unsqueeze(1)
+        return logits.view(C, B, L, self.vocab_size)
+
+    def loss(self, params: Params, x: torch.Tensor,
+             y: torch.Tensor) -> torch.Tensor:
+        # y: [C, B, L] next-token targets
+        logits = self.forward(params, x)
+        C, B, L, V = logits.shape
+        from ..ops import cross_entropy_fwd_bwd
+        return cross_entropy_fwd_bwd(logits.reshape(C * B * L, V),
+                                     y.reshape(-1))
+
+
+class BertBase(BertLM):
+    name = "bert-base"
+
+    def __init__(self, seq_len: int = 128, vocab_size: int = 30522):
+        super().__init__(vocab_size=vocab_size, hidden=768, layers=12,
+                         heads=12, seq_len=seq_len)
+
+
+class BertTiny(BertLM):
+    """4-layer/128-hidden variant for tests and CPU runs."""
+    name = "bert"
+
+    def __init__(self, seq_len: int = 32, vocab_size: int = 1000,
+                 hidden: int = 128, layers: int = 2, heads: int = 4):
+        super().__init__(vocab_size=vocab_size, hidden=hidden, layers=layers,
+                         heads=heads, seq_len=seq_len)
