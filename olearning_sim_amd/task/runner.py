@@ -1,0 +1,310 @@
+"""Task runner: turn a scheduled TaskConfig into running simulations.
+
+Parity with the reference's TaskRunner (taskMgr/task_runner.py:30-256):
+`submit` computes the hybrid allocation, assembles the per-side job
+descriptions, registers with deviceflow when the gradient house is in
+use, writes logical_target/device_target rows, and launches
+
+- the logical simulation: an in-process LogicalEngine thread (the
+  reference submits run_task.py as a Ray job, task_runner.py:41-87;
+  here the engine IS the execution plane), and
+- the device simulation: a PhoneFarmSimulator thread that models the
+  real-phone farm's completion using the reference's published cost
+  model (LAMBDA + BETA * nums / phones) — the proprietary PhoneMgr farm
+  is not reachable, so its *timing and result semantics* are simulated.
+
+Operator knobs come from the train operator's `operator_params` JSON
+(the operator-facing API of taskMgr/base/base_operator.py:12-53):
+model, lr, local_steps, batch_size, prox_mu, cohort_size, num_classes,
+dirichlet_alpha, dtype, chunk_clients, vocab_size, seq_len.
+"""
+
+from __future__ import annotations
+
+import json
+import threading
+import time
+import uuid
+from typing import Any, Dict, List, Optional
+
+from ..engine.job import EngineJob
+from ..utils.logging import Logger
+from .allocation import HybridOptimizer, DataAllocation, BETA, LAMBDA
+from .schema import TaskConfig
+from .status import JobStatus
+from .table import TaskTableRepo
+
+
+class JobHandle:
+    """Tracks one side's execution (reference: Ray job id / phone task)."""
+
+    def __init__(self, job_id: str, kind: str):
+        self.job_id = job_id
+        self.kind = kind
+        self.status = JobStatus.PENDING
+        self.error: Optional[str] = None
+        self._stop = threading.Event()
+        self.thread: Optional[threading.Thread] = None
+
+    def request_stop(self) -> None:
+        self._stop.set()
+
+    @property
+    def stop_requested(self) -> bool:
+        return self._stop.is_set()
+
+
+def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
+                         device: str = "cpu",
+                         checkpoint_dir: str = "") -> EngineJob:
+    """Build the logical-simulation EngineJob from the task config."""
+    train_op = None
+    for op in task.operatorflow.operators:
+        if op.logical_simulation.operator_code_path or \
+                op.logical_simulation.operator_entry_file:
+            train_op = op
+            break
+    params: Dict[str, Any] = {}
+    if train_op is not None and train_op.logical_simulation.operator_params:
+        try:
+            params = json.loads(train_op.logical_simulation.operator_params)
+        except Exception:
+            params = {}
+
+    clients = sum(a.logical_total() for a in allocations)
+    dynamic = sum(sum(d.total_simulation.dynamic_nums)
+                  for d in task.target.data)
+    behavior = ""
+    if train_op is not None and \
+            train_op.operation_behavior_controller.use_gradient_house:
+        behavior = train_op.operation_behavior_controller.strategy_gradient_house
+
+    first_data = task.target.data[0] if task.target.data else None
+    job = EngineJob(
+        task_id=task.task_id,
+        model_name=params.get("model", "mlp"),
+        model_kwargs=params.get("model_kwargs", {}),
+        clients=max(1, clients),
+        cohort_size=params.get("cohort_size", 0),
+        rounds=max(1, task.operatorflow.flow_setting.round),
+        local_steps=params.get("local_steps", 2),
+        batch_size=params.get("batch_size", 8),
+        lr=params.get("lr", 0.05),
+        prox_mu=params.get("prox_mu", 0.0),
+        dtype=params.get("dtype", "float32" if device == "cpu" else "bfloat16"),
+        device=device,
+        chunk_clients=params.get("chunk_clients", 0),
+        seed=params.get("seed", 1234),
+        num_classes=params.get("num_classes", 10),
+        dirichlet_alpha=params.get("dirichlet_alpha", 0.1),
+        shard_size=params.get("shard_size", 64),
+        vocab_size=params.get("vocab_size", 0),
+        seq_len=params.get("seq_len", 0),
+        behavior_strategy=behavior,
+        checkpoint_dir=checkpoint_dir,
+        save_every_round=bool(train_op and train_op.model.use_model
+                              and checkpoint_dir),
+        model_update_style=(train_op.model.model_update_style
+                            if train_op else ""),
+        data_name=first_data.name if first_data else "data_0",
+        device_tier=(first_data.total_simulation.devices[0]
+                     if first_data and first_data.total_simulation.devices
+                     else "high"),
+        dynamic_num=dynamic,
+    )
+    return job
+
+
+class TaskRunner:
+    def __init__(self, table: TaskTableRepo, device: str = "cpu",
+                 checkpoint_dir: str = "", deviceflow=None):
+        self.table = table
+        self.device = device
+        self.checkpoint_dir = checkpoint_dir
+        self.deviceflow = deviceflow   # deviceflow service facade (optional)
+        self.jobs: Dict[str, JobHandle] = {}
+        self.log = Logger.shared()
+        self._lock = threading.Lock()
+
+    # ------------------------------------------------------------------
+    def submit(self, task: TaskConfig) -> Optional[str]:
+        """Allocate, register, launch.  Returns the logical job id."""
+        allocations = HybridOptimizer(task).allocate()
+
+        logical_target = {"logical_target": [
+            {"name": a.data_name,
+             "simulation_target": {
+                 "devices": [t.tier for t in a.tiers],
+                 "nums": [t.logical for t in a.tiers],
+                 "dynamic_nums": []}}
+            for a in allocations if a.logical_total() > 0]}
+        device_target = {"device_target": [
+            {"name": a.data_name,
+             "simulation_target": {
+                 "devices": [t.tier for t in a.tiers],
+                 "nums": [t.device for t in a.tiers],
+                 "dynamic_nums": []}}
+            for a in allocations if a.device_total() > 0]}
+
+        if logical_target["logical_target"]:
+            self.table.set_item_value(task.task_id, "logical_target",
+                                      json.dumps(logical_target))
+        if device_target["device_target"]:
+            self.table.set_item_value(task.task_id, "device_target",
+                                      json.dumps(device_target))
+
+        if self.deviceflow is not None:
+            for op in task.operatorflow.operators:
+                if op.operation_behavior_controller.use_gradient_house:
+                    self.deviceflow.register_task(
+                        task.task_id, total_compute_resources=self._sides(
+                            allocations))
+                    break
+
+        job_id = None
+        if logical_target["logical_target"]:
+            job_id = self._submit_logical(task, allocations)
+        if device_target["device_target"]:
+            self._submit_phone(task, allocations)
+        return job_id
+
+    @staticmethod
+    def _sides(allocations: List[DataAllocation]) -> List[str]:
+        sides = []
+        if any(a.logical_total() > 0 for a in allocations):
+            sides.append("logical_simulation")
+        if any(a.device_total() > 0 for a in allocations):
+            sides.append("device_simulation")
+        return sides
+
+    # -- logical side ----------------------------------------------------
+    def _submit_logical(self, task: TaskConfig,
+                        allocations: List[DataAllocation]) -> str:
+        job = engine_job_from_task(task, allocations, self.device,
+                                   self.checkpoint_dir)
+        job_id = f"olsjob_{uuid.uuid4().hex[:12]}"
+        handle = JobHandle(job_id, "logical")
+
+        def run():
+            handle.status = JobStatus.RUNNING
+            try:
+                from ..engine.round_loop import LogicalEngine
+
+                def sink(row: Dict[str, Any]) -> None:
+                    self.table.set_items(
+                        task.task_id,
+                        logical_round=row["logical_round"],
+                        logical_operator=row["logical_operator"],
+                        logical_result=json.dumps(row["logical_result"]))
+
+                eng = LogicalEngine(job, result_sink=sink)
+                # cooperative stop (reference: JobSubmissionClient.stop_job)
+                orig_run_round = eng.run_round
+
+                def run_round(r):
+                    if handle.stop_requested:
+                        eng.stop_requested = True
+                    return orig_run_round(r)
+
+                eng.run_round = run_round
+                out = eng.run()
+                if handle.stop_requested:
+                    handle.status = JobStatus.STOPPED
+                elif any(rec.get("round_failed") for rec in out["records"]):
+                    handle.status = JobStatus.FAILED
+                else:
+                    handle.status = JobStatus.SUCCEEDED
+            except Exception as e:  # engine crash -> FAILED, like a Ray job
+                handle.error = str(e)
+                handle.status = JobStatus.FAILED
+                self.log.error(task.task_id, "TaskMgr", "runner",
+                               f"logical job failed: {e}")
+
+        handle.thread = threading.Thread(target=run, daemon=True)
+        with self._lock:
+            self.jobs[job_id] = handle
+        handle.thread.start()
+        return job_id
+
+    # -- device side (simulated phone farm) -----------------------------
+    def _submit_phone(self, task: TaskConfig,
+                      allocations: List[DataAllocation]) -> str:
+        job_id = f"phone_{uuid.uuid4().hex[:12]}"
+        handle = JobHandle(job_id, "device")
+        rounds = max(1, task.operatorflow.flow_setting.round)
+        phones_map: Dict[str, int] = {}
+        for rr in task.device_simulation.resource_request:
+            for tier, n in zip(rr.devices, rr.num_request):
+                phones_map[tier] = phones_map.get(tier, 0) + n
+
+        def run():
+            handle.status = JobStatus.RUNNING
+            try:
+                for r in range(rounds):
+                    per_round = 0.0
+                    for a in allocations:
+                        for t in a.tiers:
+                            if t.device > 0:
+                                per_round = max(
+                                    per_round,
+                                    LAMBDA + BETA * t.device
+                                    / max(1, phones_map.get(t.tier, 1)))
+                    # scaled-down wait: the farm timing model, compressed
+                    # so simulated phone rounds do not dominate tests
+                    waited = 0.0
+                    while waited < min(per_round * 0.01, 2.0):
+                        if handle.stop_requested:
+                            handle.status = JobStatus.STOPPED
+                            return
+                        time.sleep(0.01)
+                        waited += 0.01
+                    result = {"device_result": [
+                        {"name": a.data_name,
+                         "simulation_target": {
+                             "devices": [t.tier for t in a.tiers],
+                             "success_num": [t.device for t in a.tiers],
+                             "failed_num": [0 for _ in a.tiers]}}
+                        for a in allocations if a.device_total() > 0]}
+                    self.table.set_items(
+                        task.task_id,
+                        device_round=r + 1,
+                        device_operator="train",
+                        device_result=json.dumps(result))
+                handle.status = JobStatus.SUCCEEDED
+            except Exception as e:
+                handle.error = str(e)
+                handle.status = JobStatus.FAILED
+
+        handle.thread = threading.Thread(target=run, daemon=True)
+        with self._lock:
+            self.jobs[job_id] = handle
+        handle.thread.start()
+        return job_id
+
+    # ------------------------------------------------------------------
+    def get_job_status(self, job_id: str) -> Optional[JobStatus]:
+        h = self.jobs.get(job_id)
+        return h.status if h else None
+
+    def stop_job(self, job_id: str) -> bool:
+        h = self.jobs.get(job_id)
+        if h is None:
+            return False
+        h.request_stop()
+        return True
+
+    def stop_task(self, task_id: str) -> None:
+        job_id = self.table.get_item_value(task_id, "job_id")
+        if job_id:
+            self.stop_job(job_id)
+        for h in self.jobs.values():
+            if h.kind == "device":
+                h.request_stop()
+
+    def wait(self, job_id: str, timeout: float = 60.0) -> Optional[JobStatus]:
+        h = self.jobs.get(job_id)
+        if h is None:
+            return None
+        if h.thread is not None:
+            h.thread.join(timeout)
+        return h.status

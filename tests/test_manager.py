@@ -1,0 +1,186 @@
+"""End-to-end task lifecycle through the control plane:
+submit -> validate -> queue -> schedule -> freeze -> engine run ->
+status fusion -> release. (reference call stack SURVEY.md §3.1/§3.4)"""
+
+import json
+import time
+
+import pytest
+
+from olearning_sim_amd.resource.manager import ResourceManager
+from olearning_sim_amd.task.manager import TaskManager
+from olearning_sim_amd.task.runner import TaskRunner
+from olearning_sim_amd.task.status import TaskStatus
+from olearning_sim_amd.task.table import TaskTableRepo
+
+
+def task_json(task_id="t_e2e", rounds=2, clients=6, dynamic=1,
+              priority=0, params=None, use_gradient_house=False):
+    op_params = {
+        "model": "mlp",
+        "model_kwargs": {"in_features": 32, "hidden": 16, "num_classes": 5},
+        "lr": 0.1, "local_steps": 1, "batch_size": 4, "num_classes": 5,
+        "shard_size": 8,
+    }
+    if params:
+        op_params.update(params)
+    return json.dumps({
+        "user_id": "u1", "task_id": task_id,
+        "target": {"priority": priority, "data": [{
+            "name": "data_0", "data_path": "", "data_split_type": False,
+            "data_transfer_type": "FILE", "task_type": "classification",
+            "total_simulation": {"devices": ["high"], "nums": [clients],
+                                 "dynamic_nums": [dynamic]},
+            "allocation": {"optimization": False,
+                           "logical_simulation": [clients],
+                           "device_simulation": [0],
+                           "running_response": {"devices": [], "nums": []}}}]},
+        "operatorflow": {
+            "flow_setting": {"round": rounds,
+                             "start": {"logical_simulation": {}, "device_simulation": {}},
+                             "stop": {"logical_simulation": {}, "device_simulation": {}}},
+            "operators": [{
+                "name": "train",
+                "operation_behavior_controller": {
+                    "use_gradient_house": use_gradient_house,
+                    "strategy_gradient_house": json.dumps(
+                        {"real_time_dispatch": {"use_strategy": True}})
+                    if use_gradient_house else "",
+                    "outbound_service": ""},
+                "input": [], "use_data": True,
+                "model": {"use_model": False},
+                "logical_simulation": {
+                    "operator_transfer_type": "FILE",
+                    "operator_code_path": "builtin:fedavg",
+                    "operator_entry_file": "train.py",
+                    "operator_params": json.dumps(op_params)},
+                "device_simulation": {}}]},
+        "logical_simulation": {
+            "computation_unit": {"devices": ["high"],
+                                 "setting": [{"num_cpus": 1}]},
+            "resource_request": [{"name": "data_0", "devices": ["high"],
+                                  "num_request": [2]}]},
+        "device_simulation": {"resource_request": []},
+    })
+
+
+def make_manager(cpu=8.0):
+    table = TaskTableRepo(":memory:")
+    res = ResourceManager(":memory:", totals={"cpu": cpu, "mem": 64.0,
+                                              "gpu": 0, "hbm_gb": 0})
+    runner = TaskRunner(table, device="cpu")
+    return TaskManager(table=table, resource_mgr=res, runner=runner)
+
+
+def wait_terminal(mgr, task_id, timeout=30.0):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        st = mgr.get_task_status(task_id)
+        if st.is_terminal():
+            return st
+        time.sleep(0.05)
+    return mgr.get_task_status(task_id)
+
+
+def test_submit_validates_and_queues():
+    mgr = make_manager()
+    ok, msg = mgr.submit_task(task_json())
+    assert ok, msg
+    assert mgr.get_task_status("t_e2e") == TaskStatus.QUEUED
+    assert mgr.get_task_queue() == ["t_e2e"]
+    # duplicate rejected
+    ok2, msg2 = mgr.submit_task(task_json())
+    assert not ok2
+
+
+def test_submit_rejects_invalid():
+    mgr = make_manager()
+    bad = json.loads(task_json())
+    bad["target"]["priority"] = 99
+    ok, msg = mgr.submit_task(json.dumps(bad))
+    assert not ok and "priority" in msg
+
+
+def test_full_lifecycle_succeeds():
+    mgr = make_manager()
+    ok, _ = mgr.submit_task(task_json())
+    assert ok
+    task_id = mgr.step_schedule()
+    assert task_id == "t_e2e"
+    assert mgr.table.get_item_value("t_e2e", "resource_occupied") == 1
+    assert mgr.resources.holding("t_e2e")
+    st = wait_terminal(mgr, "t_e2e")
+    assert st == TaskStatus.SUCCEEDED
+    # release step frees the quota
+    released = mgr.step_release()
+    assert "t_e2e" in released
+    assert not mgr.resources.holding("t_e2e")
+    # results recorded with reference shapes
+    lr = json.loads(mgr.table.get_item_value("t_e2e", "logical_result"))
+    tgt = lr["logical_result"][0]["simulation_target"]
+    assert tgt["success_num"] == [6]
+    assert mgr.table.get_item_value("t_e2e", "logical_round") == 2
+
+
+def test_no_schedule_when_resources_missing():
+    mgr = make_manager(cpu=1.0)   # needs 2 cpus
+    mgr.submit_task(task_json())
+    assert mgr.step_schedule() is None
+    assert mgr.get_task_status("t_e2e") == TaskStatus.QUEUED
+
+
+def test_priority_scheduling_order():
+    mgr = make_manager()
+    mgr.submit_task(task_json(task_id="low", priority=0))
+    mgr.submit_task(task_json(task_id="high", priority=10))
+    first = mgr.step_schedule()
+    assert first == "high"
+
+
+def test_stop_queued_task():
+    mgr = make_manager(cpu=1.0)
+    mgr.submit_task(task_json())
+    ok, _ = mgr.stop_task("t_e2e")
+    assert ok
+    assert mgr.get_task_status("t_e2e") == TaskStatus.STOPPED
+
+
+def test_missing_task_status():
+    mgr = make_manager()
+    assert mgr.get_task_status("nope") == TaskStatus.MISSING
+
+
+def test_interrupt_overdue_queued_task():
+    mgr = make_manager(cpu=1.0)
+    mgr.timers["interrupt_queue_time"] = 0.0
+    mgr.submit_task(task_json())
+    time.sleep(0.02)
+    assert "t_e2e" in mgr.step_interrupt()
+    assert mgr.get_task_status("t_e2e") == TaskStatus.FAILED
+
+
+def test_round_failure_reports_failed():
+    """All clients offline + tolerance 0 -> round fails -> task FAILED."""
+    mgr = make_manager()
+    strategy = json.dumps({"offline_simulation": {"offline_probability": 1.0}})
+    tj = json.loads(task_json(task_id="t_fail", dynamic=1))
+    op = tj["operatorflow"]["operators"][0]
+    op["operation_behavior_controller"]["use_gradient_house"] = True
+    op["operation_behavior_controller"]["strategy_gradient_house"] = strategy
+    mgr.submit_task(json.dumps(tj))
+    assert mgr.step_schedule() == "t_fail"
+    st = wait_terminal(mgr, "t_fail")
+    assert st == TaskStatus.FAILED
+
+
+def test_queue_recovery_from_table():
+    table = TaskTableRepo(":memory:")
+    res = ResourceManager(":memory:", totals={"cpu": 8, "mem": 64,
+                                              "gpu": 0, "hbm_gb": 0})
+    mgr1 = TaskManager(table=table, resource_mgr=res,
+                       runner=TaskRunner(table))
+    mgr1.submit_task(task_json())
+    # simulate restart: new manager over the same table
+    mgr2 = TaskManager(table=table, resource_mgr=res,
+                       runner=TaskRunner(table))
+    assert mgr2.get_task_queue() == ["t_e2e"]
