@@ -1,0 +1,72 @@
+"""Client-batched convolution ops in client-channel-first layout.
+
+Layout [C, ch, B, H, W]: per (client, channel) the [B, H, W] block is
+contiguous — the implicit-GEMM-friendly layout of
+ops/csrc/client_conv.hip (custom MFMA kernels for 3x3), and 1x1 convs
+collapse to one batched GEMM over clients (torch.bmm -> hipBLASLt).
+
+CPU fallback: grouped F.conv2d after a layout permute (reference
+semantics, used by tests and the CPU plumbing config).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from .fused import load_hip_ops
+
+
+def _cpu_conv3x3(x: torch.Tensor, w: torch.Tensor, stride: int) -> torch.Tensor:
+    # x [C, IC, B, H, W], w [C, OC, IC, 3, 3] -> y [C, OC, B, OH, OW]
+    C, IC, B, H, W = x.shape
+    OC = w.shape[1]
+    xg = x.permute(2, 0, 1, 3, 4).reshape(B, C * IC, H, W)
+    wf = w.reshape(C * OC, IC, 3, 3)
+    y = F.conv2d(xg, wf, stride=stride, padding=1, groups=C)
+    OH, OW = y.shape[-2:]
+    return y.reshape(B, C, OC, OH, OW).permute(1, 2, 0, 3, 4).contiguous()
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, stride):
+        ops = load_hip_ops(required=True)
+        y = ops.conv3x3_fwd(x, w, stride)
+        ctx.save_for_backward(x, w)
+        ctx.stride = stride
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        ops = load_hip_ops(required=True)
+        dy = dy.contiguous()
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = ops.conv3x3_dgrad(dy, w, x.shape[3], x.shape[4], ctx.stride)
+        if ctx.needs_input_grad[1]:
+            dw = ops.conv3x3_wgrad(x, dy, ctx.stride).to(w.dtype)
+        return dx, dw, None
+
+
+def client_conv3x3(x: torch.Tensor, w: torch.Tensor,
+                   stride: int = 1) -> torch.Tensor:
+    """y[C,OC,B,OH,OW] = conv3x3(x[C,IC,B,H,W], w[C,OC,IC,3,3]), pad 1."""
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _Conv3x3Fn.apply(x.contiguous(), w.contiguous(), stride)
+    return _cpu_conv3x3(x, w, stride)
+
+
+def client_conv1x1(x: torch.Tensor, w: torch.Tensor,
+                   stride: int = 1) -> torch.Tensor:
+    """1x1 conv = one batched GEMM: y[C,OC,n] = w[C,OC,IC] @ x[C,IC,n]."""
+    C, IC, B, H, W = x.shape
+    OC = w.shape[1]
+    if stride != 1:
+        x = x[:, :, :, ::stride, ::stride].contiguous()
+        H, W = x.shape[-2:]
+    y = torch.bmm(w.reshape(C, OC, IC), x.reshape(C, IC, B * H * W))
+    return y.view(C, OC, B, H, W)

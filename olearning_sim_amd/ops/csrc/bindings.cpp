@@ -26,6 +26,19 @@ extern "C" void ols_groupnorm_fwd(const void* x, const void* res, void* y,
                                   int G, int HW, float eps, bool relu,
                                   int dtype, hipStream_t stream);
 
+extern "C" void ols_mfma_selftest(const void* A, const void* B, float* D,
+                                  hipStream_t stream);
+
+extern "C" void ols_conv3x3_fwd(const void* x, const void* w, void* y, int C,
+                                int IC, int OC, int B, int H, int W,
+                                int stride, hipStream_t stream);
+extern "C" void ols_conv3x3_dgrad(const void* dy, const void* w, void* dx,
+                                  int C, int IC, int OC, int B, int H, int W,
+                                  int stride, hipStream_t stream);
+extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
+                                  int C, int IC, int OC, int B, int H, int W,
+                                  int stride, hipStream_t stream);
+
 extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const void* dy, void* dx, void* dres,
                                   const float* mean, const float* rstd,
@@ -156,6 +169,56 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> groupnorm_bwd(
   return {dx, dres, dgamma, dbeta};
 }
 
+at::Tensor mfma_selftest(at::Tensor A, at::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({16, 32}) &&
+              B.sizes() == at::IntArrayRef({32, 16}));
+  auto D = at::empty({16, 16}, A.options().dtype(at::kFloat));
+  ols_mfma_selftest(A.contiguous().data_ptr(), B.contiguous().data_ptr(),
+                    D.data_ptr<float>(),
+                    at::cuda::getCurrentCUDAStream().stream());
+  return D;
+}
+
+// x: [C, IC, B, H, W] bf16; w: [C, OC, IC, 3, 3] bf16 -> y [C, OC, B, OH, OW]
+at::Tensor conv3x3_fwd(at::Tensor x, at::Tensor w, int64_t stride) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 5);
+  TORCH_CHECK(w.is_contiguous() && w.dim() == 5 && w.size(3) == 3);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16);
+  int C = x.size(0), IC = x.size(1), B = x.size(2), H = x.size(3),
+      W = x.size(4), OC = w.size(1);
+  TORCH_CHECK(w.size(0) == C && w.size(2) == IC);
+  int OH = (H + stride - 1) / stride, OW = (W + stride - 1) / stride;
+  auto y = at::empty({C, OC, B, OH, OW}, x.options());
+  ols_conv3x3_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), C, IC, OC, B, H,
+                  W, (int)stride, at::cuda::getCurrentCUDAStream().stream());
+  return y;
+}
+
+at::Tensor conv3x3_dgrad(at::Tensor dy, at::Tensor w, int64_t H, int64_t W,
+                         int64_t stride) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 5);
+  int C = dy.size(0), OC = dy.size(1), B = dy.size(2);
+  int IC = w.size(2);
+  auto dx = at::empty({C, IC, B, H, W}, dy.options());
+  ols_conv3x3_dgrad(dy.data_ptr(), w.contiguous().data_ptr(), dx.data_ptr(),
+                    C, IC, OC, B, (int)H, (int)W, (int)stride,
+                    at::cuda::getCurrentCUDAStream().stream());
+  return dx;
+}
+
+at::Tensor conv3x3_wgrad(at::Tensor x, at::Tensor dy, int64_t stride) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  int C = x.size(0), IC = x.size(1), B = x.size(2), H = x.size(3),
+      W = x.size(4), OC = dy.size(1);
+  auto dw = at::empty({C, OC, IC, 3, 3}, x.options().dtype(at::kFloat));
+  ols_conv3x3_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), C, IC,
+                    OC, B, H, W, (int)stride,
+                    at::cuda::getCurrentCUDAStream().stream());
+  return dw;
+}
+
 }  // namespace
 
 TORCH_LIBRARY(olsim_hip, m) {
@@ -169,6 +232,10 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("groupnorm_bwd(Tensor x, Tensor y, Tensor dy, Tensor mean, "
         "Tensor rstd, Tensor gamma, int clients, int groups, bool has_res, "
         "bool relu) -> (Tensor, Tensor, Tensor, Tensor)");
+  m.def("mfma_selftest(Tensor A, Tensor B) -> Tensor");
+  m.def("conv3x3_fwd(Tensor x, Tensor w, int stride) -> Tensor");
+  m.def("conv3x3_dgrad(Tensor dy, Tensor w, int H, int W, int stride) -> Tensor");
+  m.def("conv3x3_wgrad(Tensor x, Tensor dy, int stride) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
@@ -177,4 +244,8 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd);
   m.impl("groupnorm_fwd", &groupnorm_fwd);
   m.impl("groupnorm_bwd", &groupnorm_bwd);
+  m.impl("mfma_selftest", &mfma_selftest);
+  m.impl("conv3x3_fwd", &conv3x3_fwd);
+  m.impl("conv3x3_dgrad", &conv3x3_dgrad);
+  m.impl("conv3x3_wgrad", &conv3x3_wgrad);
 }

@@ -1,0 +1,363 @@
+// Client-batched 3x3 convolution as implicit GEMM on MFMA — the core
+// compute kernel of the simulator: every co-resident virtual client
+// convolves with ITS OWN filters, so the client dimension is a grid
+// dimension and each workgroup computes one (client, M-tile, N-tile)
+// output block on `v_mfma_f32_16x16x32_bf16` matrix cores with
+// LDS-staged tiles.  Replaces MIOpen grouped conv (which falls back to
+// per-group GEMM loops / naive kernels at thousands of groups).
+//
+// Activation layout: [C, ch, B, H, W] ("client-channel-first"): for a
+// fixed (client, channel) the [B, H, W] block is contiguous, so the
+// implicit-GEMM N dimension n=(b, oh, ow) walks nearly contiguous
+// memory and output writes are coalesced.
+//
+// GEMM views (per client, pad=1, stride s, kernel 3x3):
+//   fwd   : Y[oc][n]    = sum_k  W[oc][k]      * P[k][n]
+//           k=(ic,dh,dw), P[k][n] = x[ic][b][oh*s+dh-1][ow*s+dw-1]
+//   dgrad : dX[ic][n]   = sum_k  W'[ic][k]     * Q[k][n]
+//           k=(oc,dh,dw), W'[ic][(oc,dh,dw)] = W[oc][ic][2-dh][2-dw],
+//           Q[k][n] = dY[oc][b][oh][ow] where oh=(h+1-dh')/s if exact
+//   wgrad : dW[oc][k]   = sum_n dY[oc][n] * P[k][n]   (K = n = B*OH*OW)
+//
+// Tiles: BM=64 (4 waves x 16 rows), BN=64, BK=32 (one MFMA K per step),
+// 256 threads.  Gathered operands are staged through LDS; gathers are
+// the price of implicit im2col — they read each input element 9 times
+// through L2 instead of materialising a 9x patch matrix in HBM.
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__device__ __forceinline__ short bf16_bits(float v) {
+  __hip_bfloat16 h = from_f32<__hip_bfloat16>(v);
+  return *reinterpret_cast<short*>(&h);
+}
+
+#define CONV_BM 64
+#define CONV_BN 64
+#define CONV_BK 32
+#define CONV_THREADS 256
+
+// fragment maps for v_mfma_f32_16x16x32_bf16 (cdna_hip_programming.md §3):
+//   A[row][k]: lane l holds rows row=l&15, k = 8*(l>>4) + e   (e=0..7)
+//   B[k][col]: lane l holds col=l&15,      k = 8*(l>>4) + e
+//   C/D:       lane l, reg r -> row = (l>>4)*4 + r, col = l&15
+
+// ---------------------------------------------------------------------------
+// Tiny layout self-test: D[16][16] = A[16][32] x B[32][16] for one
+// workgroup of 64 threads — verifies the fragment maps on hardware.
+__global__ __launch_bounds__(64) void k_mfma_selftest(
+    const __hip_bfloat16* __restrict__ A, const __hip_bfloat16* __restrict__ B,
+    float* __restrict__ D) {
+  int l = threadIdx.x;
+  bf16x8 a, b;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    a[e] = *reinterpret_cast<const short*>(&A[(l & 15) * 32 + 8 * (l >> 4) + e]);
+    b[e] = *reinterpret_cast<const short*>(&B[(8 * (l >> 4) + e) * 16 + (l & 15)]);
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    D[((l >> 4) * 4 + r) * 16 + (l & 15)] = c[r];
+}
+
+// ---------------------------------------------------------------------------
+// shared gather helpers
+
+struct ConvGeom {
+  int B, H, W, OH, OW, IC, OC, stride;
+};
+
+// stage the fwd/wgrad patch tile P[k][n] (k in [k0,k0+BK), n in
+// [n0,n0+BN)) into lds[BK][BN]; k=(ic,dh,dw) w/ dw fastest.
+template <int TILE_K, int TILE_N>
+__device__ void stage_patch(const __hip_bfloat16* __restrict__ x,
+                            short* lds, const ConvGeom g,
+                            int k0, int n0, int kmax, int nmax) {
+  const int HW = g.H * g.W;
+  const int OHW = g.OH * g.OW;
+  // each thread fills (TILE_K*TILE_N)/CONV_THREADS elements, n fastest
+  for (int i = threadIdx.x; i < TILE_K * TILE_N; i += CONV_THREADS) {
+    int kk = i / TILE_N, nn = i % TILE_N;
+    int k = k0 + kk, n = n0 + nn;
+    float v = 0.f;
+    if (k < kmax && n < nmax) {
+      int ic = k / 9, r = k % 9;
+      int dh = r / 3, dw = r % 3;
+      int b = n / OHW, q = n % OHW;
+      int oh = q / g.OW, ow = q % g.OW;
+      int ih = oh * g.stride + dh - 1, iw = ow * g.stride + dw - 1;
+      if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
+        v = to_f32(x[((int64_t)ic * g.B + b) * HW + ih * g.W + iw]);
+    }
+    lds[kk * TILE_N + nn] = bf16_bits(v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// forward: grid (ntiles_n, ntiles_m, C); x[C,IC,B,H,W] w[C,OC,IC,3,3]
+// y[C,OC,B,OH,OW]
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ y, ConvGeom g) {
+  __shared__ short a_lds[CONV_BM * CONV_BK];
+  __shared__ short b_lds[CONV_BK * CONV_BN];
+  const int c = blockIdx.z;
+  const int m0 = blockIdx.y * CONV_BM;
+  const int n0 = blockIdx.x * CONV_BN;
+  const int K = g.IC * 9;
+  const int N = g.B * g.OH * g.OW;
+  const __hip_bfloat16* xc = x + (int64_t)c * g.IC * g.B * g.H * g.W;
+  const __hip_bfloat16* wc = w + (int64_t)c * g.OC * K;
+  __hip_bfloat16* yc = y + (int64_t)c * g.OC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;   // wave owns rows [m0+wave*16, +16)
+  f32x4 acc[CONV_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CONV_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K; k0 += CONV_BK) {
+    // stage A = W[m0.., k0..]
+    for (int i = threadIdx.x; i < CONV_BM * CONV_BK; i += CONV_THREADS) {
+      int mm = i / CONV_BK, kk = i % CONV_BK;
+      int m = m0 + mm, k = k0 + kk;
+      a_lds[i] = (m < g.OC && k < K)
+                     ? bf16_bits(to_f32(wc[(int64_t)m * K + k])) : (short)0;
+    }
+    stage_patch<CONV_BK, CONV_BN>(xc, b_lds, g, k0, n0, K, N);
+    __syncthreads();
+
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(
+        &a_lds[(wave * 16 + (lane & 15)) * CONV_BK + 8 * (lane >> 4)]);
+#pragma unroll
+    for (int nt = 0; nt < CONV_BN / 16; ++nt) {
+      bf16x8 b;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        b[e] = b_lds[(8 * (lane >> 4) + e) * CONV_BN + nt * 16 + (lane & 15)];
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // write D: row = m0 + wave*16 + (lane>>4)*4 + r, col = n0 + nt*16 + (lane&15)
+#pragma unroll
+  for (int nt = 0; nt < CONV_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        yc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dgrad: dX[C,IC,B,H,W] from dY[C,OC,B,OH,OW]; k=(oc,dh,dw),
+// A[ic][k] = W[oc][ic][2-dh][2-dw]
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad(
+    const __hip_bfloat16* __restrict__ dy, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ dx, ConvGeom g) {
+  __shared__ short a_lds[CONV_BM * CONV_BK];
+  __shared__ short b_lds[CONV_BK * CONV_BN];
+  const int c = blockIdx.z;
+  const int m0 = blockIdx.y * CONV_BM;           // over IC
+  const int n0 = blockIdx.x * CONV_BN;           // over B*H*W
+  const int K = g.OC * 9;
+  const int N = g.B * g.H * g.W;
+  const int OHW = g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * g.B * OHW;
+  const __hip_bfloat16* wc = w + (int64_t)c * g.OC * g.IC * 9;
+  __hip_bfloat16* dxc = dx + (int64_t)c * g.IC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  f32x4 acc[CONV_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CONV_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < K; k0 += CONV_BK) {
+    // A[ic][k=(oc,dh,dw)] = W[oc][ic][2-dh][2-dw]
+    for (int i = threadIdx.x; i < CONV_BM * CONV_BK; i += CONV_THREADS) {
+      int mm = i / CONV_BK, kk = i % CONV_BK;
+      int ic = m0 + mm, k = k0 + kk;
+      float v = 0.f;
+      if (ic < g.IC && k < K) {
+        int oc = k / 9, r = k % 9;
+        int dh = r / 3, dw = r % 3;
+        v = to_f32(wc[((int64_t)oc * g.IC + ic) * 9 + (2 - dh) * 3 + (2 - dw)]);
+      }
+      a_lds[i] = bf16_bits(v);
+    }
+    // B[k][n] = dY[oc][b][oh][ow] with h+dh-1 = oh*s (flipped offsets)
+    for (int i = threadIdx.x; i < CONV_BK * CONV_BN; i += CONV_THREADS) {
+      int kk = i / CONV_BN, nn = i % CONV_BN;
+      int k = k0 + kk, n = n0 + nn;
+      float v = 0.f;
+      if (k < K && n < N) {
+        int oc = k / 9, r = k % 9;
+        int dh = r / 3, dw = r % 3;
+        int b = n / HW, q = n % HW;
+        int h = q / g.W, wd = q % g.W;
+        int num_h = h + dh - 1, num_w = wd + dw - 1;
+        if (num_h >= 0 && num_w >= 0 && num_h % g.stride == 0
+            && num_w % g.stride == 0) {
+          int oh = num_h / g.stride, ow = num_w / g.stride;
+          if (oh < g.OH && ow < g.OW)
+            v = to_f32(dyc[((int64_t)oc * g.B + b) * OHW + oh * g.OW + ow]);
+        }
+      }
+      b_lds[kk * CONV_BN + nn] = bf16_bits(v);
+    }
+    __syncthreads();
+
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(
+        &a_lds[(wave * 16 + (lane & 15)) * CONV_BK + 8 * (lane >> 4)]);
+#pragma unroll
+    for (int nt = 0; nt < CONV_BN / 16; ++nt) {
+      bf16x8 b;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        b[e] = b_lds[(8 * (lane >> 4) + e) * CONV_BN + nt * 16 + (lane & 15)];
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CONV_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.IC)
+        dxc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad: dW[C,OC,IC*9] (fp32 out) = dY[oc][n] x P[k][n] over n.
+// grid (ntiles_k9, ntiles_oc, C); K-loop over n.
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
+    float* __restrict__ dw, ConvGeom g) {
+  __shared__ short a_lds[CONV_BM * CONV_BK];   // dY tile [oc][n]
+  __shared__ short b_lds[CONV_BK * CONV_BN];   // P^T tile [n][k9]
+  const int c = blockIdx.z;
+  const int m0 = blockIdx.y * CONV_BM;          // over OC
+  const int n0 = blockIdx.x * CONV_BN;          // over IC*9
+  const int K9 = g.IC * 9;
+  const int NN = g.B * g.OH * g.OW;             // reduction dim
+  const int OHW = g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const __hip_bfloat16* xc = x + (int64_t)c * g.IC * g.B * HW;
+  const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
+  float* dwc = dw + (int64_t)c * g.OC * K9;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  f32x4 acc[CONV_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CONV_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int q0 = 0; q0 < NN; q0 += CONV_BK) {
+    // A[oc][q] = dY[oc][q0+q]
+    for (int i = threadIdx.x; i < CONV_BM * CONV_BK; i += CONV_THREADS) {
+      int mm = i / CONV_BK, qq = i % CONV_BK;
+      int m = m0 + mm, q = q0 + qq;
+      a_lds[i] = (m < g.OC && q < NN)
+                     ? bf16_bits(to_f32(dyc[(int64_t)m * NN + q])) : (short)0;
+    }
+    // B[q][k9] = P[k9][q0+q]
+    for (int i = threadIdx.x; i < CONV_BK * CONV_BN; i += CONV_THREADS) {
+      int qq = i / CONV_BN, kk = i % CONV_BN;
+      int q = q0 + qq, k = n0 + kk;
+      float v = 0.f;
+      if (q < NN && k < K9) {
+        int ic = k / 9, r = k % 9;
+        int dh = r / 3, dw2 = r % 3;
+        int b = q / OHW, p = q % OHW;
+        int oh = p / g.OW, ow = p % g.OW;
+        int ih = oh * g.stride + dh - 1, iw = ow * g.stride + dw2 - 1;
+        if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
+          v = to_f32(xc[((int64_t)ic * g.B + b) * HW + ih * g.W + iw]);
+      }
+      b_lds[qq * CONV_BN + kk] = bf16_bits(v);
+    }
+    __syncthreads();
+
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(
+        &a_lds[(wave * 16 + (lane & 15)) * CONV_BK + 8 * (lane >> 4)]);
+#pragma unroll
+    for (int nt = 0; nt < CONV_BN / 16; ++nt) {
+      bf16x8 b;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        b[e] = b_lds[(8 * (lane >> 4) + e) * CONV_BN + nt * 16 + (lane & 15)];
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CONV_BN / 16; ++nt) {
+    int k = n0 + nt * 16 + (lane & 15);
+    if (k >= K9) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        dwc[(int64_t)m * K9 + k] = acc[nt][r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+extern "C" void ols_mfma_selftest(const void* A, const void* B, float* D,
+                                  hipStream_t stream) {
+  hipLaunchKernelGGL(k_mfma_selftest, dim3(1), dim3(64), 0, stream,
+                     (const __hip_bfloat16*)A, (const __hip_bfloat16*)B, D);
+}
+
+static inline int ceil_div(int a, int b) { return (a + b - 1) / b; }
+
+extern "C" void ols_conv3x3_fwd(const void* x, const void* w, void* y, int C,
+                                int IC, int OC, int B, int H, int W,
+                                int stride, hipStream_t stream) {
+  ConvGeom g{B, H, W, (H + stride - 1) / stride, (W + stride - 1) / stride,
+             IC, OC, stride};
+  dim3 grid(ceil_div(B * g.OH * g.OW, CONV_BN), ceil_div(OC, CONV_BM), C);
+  hipLaunchKernelGGL(k_conv3x3_fwd, grid, dim3(CONV_THREADS), 0, stream,
+                     (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
+                     (__hip_bfloat16*)y, g);
+}
+
+extern "C" void ols_conv3x3_dgrad(const void* dy, const void* w, void* dx,
+                                  int C, int IC, int OC, int B, int H, int W,
+                                  int stride, hipStream_t stream) {
+  ConvGeom g{B, H, W, (H + stride - 1) / stride, (W + stride - 1) / stride,
+             IC, OC, stride};
+  dim3 grid(ceil_div(B * H * W, CONV_BN), ceil_div(IC, CONV_BM), C);
+  hipLaunchKernelGGL(k_conv3x3_dgrad, grid, dim3(CONV_THREADS), 0, stream,
+                     (const __hip_bfloat16*)dy, (const __hip_bfloat16*)w,
+                     (__hip_bfloat16*)dx, g);
+}
+
+extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
+                                  int C, int IC, int OC, int B, int H, int W,
+                                  int stride, hipStream_t stream) {
+  ConvGeom g{B, H, W, (H + stride - 1) / stride, (W + stride - 1) / stride,
+             IC, OC, stride};
+  dim3 grid(ceil_div(IC * 9, CONV_BN), ceil_div(OC, CONV_BM), C);
+  hipLaunchKernelGGL(k_conv3x3_wgrad, grid, dim3(CONV_THREADS), 0, stream,
+                     (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, dw, g);
+}

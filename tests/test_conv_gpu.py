@@ -1,0 +1,87 @@
+"""Numerics of the hand-written client-batched conv3x3 MFMA kernels
+against the fp32 torch grouped-conv reference."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_ext():
+    from olearning_sim_amd.ops import load_hip_ops
+    load_hip_ops(required=True)
+
+
+def test_mfma_fragment_layout_selftest():
+    """16x16x32 bf16 MFMA fragment maps vs torch mm."""
+    from olearning_sim_amd.ops.fused import load_hip_ops
+    ops = load_hip_ops(required=True)
+    g = torch.Generator().manual_seed(0)
+    A = torch.randn(16, 32, generator=g).to(torch.bfloat16).cuda()
+    B = torch.randn(32, 16, generator=g).to(torch.bfloat16).cuda()
+    D = ops.mfma_selftest(A, B)
+    ref = (A.float() @ B.float())
+    torch.testing.assert_close(D, ref, atol=5e-2, rtol=5e-2)
+
+
+SHAPES = [
+    # (C, IC, OC, B, H, stride)
+    (3, 64, 64, 4, 16, 1),
+    (2, 64, 128, 4, 16, 2),
+    (2, 16, 16, 2, 8, 1),
+    (2, 3, 64, 4, 32, 1),      # stem: IC=3 (K=27, sub-tile)
+    (2, 512, 512, 2, 4, 1),    # deepest stage
+    (1, 128, 256, 3, 8, 2),
+]
+
+
+@pytest.mark.parametrize("C,IC,OC,B,H,stride", SHAPES)
+def test_conv3x3_fwd_matches_reference(C, IC, OC, B, H, stride):
+    from olearning_sim_amd.ops.conv import client_conv3x3, _cpu_conv3x3
+    g = torch.Generator().manual_seed(1)
+    x = torch.randn(C, IC, B, H, H, generator=g).to(torch.bfloat16).cuda()
+    w = (torch.randn(C, OC, IC, 3, 3, generator=g) * 0.1).to(torch.bfloat16).cuda()
+    y = client_conv3x3(x, w, stride)
+    ref = _cpu_conv3x3(x.float().cpu(), w.float().cpu(), stride)
+    torch.testing.assert_close(y.float().cpu(), ref,
+                               atol=0.1 * (IC ** 0.5) * 0.3, rtol=5e-2)
+
+
+@pytest.mark.parametrize("C,IC,OC,B,H,stride", SHAPES)
+def test_conv3x3_backward_matches_reference(C, IC, OC, B, H, stride):
+    from olearning_sim_amd.ops.conv import client_conv3x3, _cpu_conv3x3
+    g = torch.Generator().manual_seed(2)
+    x0 = torch.randn(C, IC, B, H, H, generator=g) * 0.5
+    w0 = torch.randn(C, OC, IC, 3, 3, generator=g) * 0.1
+
+    xg = x0.to(torch.bfloat16).cuda().requires_grad_(True)
+    wg = w0.to(torch.bfloat16).cuda().requires_grad_(True)
+    y = client_conv3x3(xg, wg, stride)
+    dy = torch.randn(y.shape, generator=g) * 0.1
+    y.backward(dy.to(torch.bfloat16).cuda())
+
+    xr = x0.clone().requires_grad_(True)
+    wr = w0.clone().requires_grad_(True)
+    yr = _cpu_conv3x3(xr, wr, stride)
+    yr.backward(dy)
+
+    scale = max(1.0, float(yr.abs().max()))
+    torch.testing.assert_close(xg.grad.float().cpu(), xr.grad,
+                               atol=5e-2 * (OC ** 0.5) * 0.1 + 2e-2, rtol=8e-2)
+    torch.testing.assert_close(wg.grad.float().cpu(), wr.grad,
+                               atol=2e-2 * (B * H * H) ** 0.5 * 0.1 + 2e-2,
+                               rtol=8e-2)
+
+
+def test_conv1x1_matches_reference():
+    from olearning_sim_amd.ops.conv import client_conv1x1
+    g = torch.Generator().manual_seed(3)
+    C, IC, OC, B, H = 3, 64, 128, 4, 16
+    for stride in (1, 2):
+        x = torch.randn(C, IC, B, H, H, generator=g).to(torch.bfloat16).cuda()
+        w = (torch.randn(C, OC, IC, 1, 1, generator=g) * 0.1).to(torch.bfloat16).cuda()
+        y = client_conv1x1(x, w, stride)
+        xs = x.float()[:, :, :, ::stride, ::stride]
+        ref = torch.einsum("coi,cibhw->cobhw", w.float().squeeze(-1).squeeze(-1), xs)
+        torch.testing.assert_close(y.float(), ref.cuda(), atol=5e-1, rtol=5e-2)
