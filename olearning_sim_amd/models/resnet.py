@@ -86,8 +86,47 @@ class ResNet18(ClientBatchedModel):
         return groupnorm_act(h, C, _GN_GROUPS, params[f"{pre}.gn2.g"],
                              params[f"{pre}.gn2.b"], res=sc, relu=True)
 
+    # -- fast path: client-channel-first layout + hand-written MFMA
+    # conv kernels (ops/conv.py / ops/csrc/client_conv.hip) -------------
+    def _block_cbf(self, params: Params, x: torch.Tensor, C: int, pre: str,
+                   stride: int, has_down: bool) -> torch.Tensor:
+        from ..ops.conv import client_conv3x3, client_conv1x1
+        from ..ops.fused import groupnorm_act
+        h = client_conv3x3(x, params[f"{pre}.c1.w"], stride)
+        h = groupnorm_act(h, C, _GN_GROUPS, params[f"{pre}.gn1.g"],
+                          params[f"{pre}.gn1.b"], relu=True)
+        h = client_conv3x3(h, params[f"{pre}.c2.w"], 1)
+        if has_down:
+            sc = client_conv1x1(x, params[f"{pre}.down.w"], stride)
+            sc = groupnorm_act(sc, C, _GN_GROUPS, params[f"{pre}.gndown.g"],
+                               params[f"{pre}.gndown.b"])
+        else:
+            sc = x
+        return groupnorm_act(h, C, _GN_GROUPS, params[f"{pre}.gn2.g"],
+                             params[f"{pre}.gn2.b"], res=sc, relu=True)
+
+    def forward_cbf(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.conv import client_conv3x3
+        from ..ops.fused import groupnorm_act
+        C, B = x.shape[0], x.shape[1]
+        h = x.permute(0, 2, 1, 3, 4).contiguous()    # [C, 3, B, 32, 32]
+        h = client_conv3x3(h, params["stem.w"], 1)
+        h = groupnorm_act(h, C, _GN_GROUPS, params["stem.gn.g"],
+                          params["stem.gn.b"], relu=True)
+        for s in range(4):
+            stride = 1 if s == 0 else 2
+            h = self._block_cbf(params, h, C, f"s{s}.b0", stride, s > 0)
+            h = self._block_cbf(params, h, C, f"s{s}.b1", 1, False)
+        # h: [C, 512, B, 4, 4] -> avg pool -> [C, B, 512]
+        h = h.mean(dim=(3, 4)).permute(0, 2, 1)
+        return blinear(h, params["fc.w"], params["fc.b"])
+
     def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
         from ..ops.fused import groupnorm_act
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            from ..ops.fused import hip_ops_available
+            if hip_ops_available():
+                return self.forward_cbf(params, x)
         # x: [C, B, 3, 32, 32]
         C, B = x.shape[0], x.shape[1]
         h = x.permute(1, 0, 2, 3, 4).reshape(B, C * self.in_ch, 32, 32)

@@ -24,7 +24,7 @@ extern "C" void ols_groupnorm_fwd(const void* x, const void* res, void* y,
                                   float* mean, float* rstd, const void* gamma,
                                   const void* beta, int B, int C, int ch,
                                   int G, int HW, float eps, bool relu,
-                                  int dtype, hipStream_t stream);
+                                  int layout, int dtype, hipStream_t stream);
 
 extern "C" void ols_mfma_selftest(const void* A, const void* B, float* D,
                                   hipStream_t stream);
@@ -44,7 +44,7 @@ extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const float* mean, const float* rstd,
                                   const void* gamma, float* dgamma,
                                   float* dbeta, int B, int C, int ch, int G,
-                                  int HW, bool relu, int dtype,
+                                  int HW, bool relu, int layout, int dtype,
                                   hipStream_t stream);
 
 namespace {
@@ -122,11 +122,19 @@ int gn_dtype(const at::Tensor& t) {
 std::tuple<at::Tensor, at::Tensor, at::Tensor> groupnorm_fwd(
     at::Tensor x, at::Tensor res, at::Tensor gamma, at::Tensor beta,
     int64_t clients, int64_t groups, double eps, bool relu) {
-  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4);
-  const int64_t B = x.size(0), cch = x.size(1),
-                HW = x.size(2) * x.size(3);
-  const int64_t ch = cch / clients;
-  TORCH_CHECK(cch % clients == 0 && ch % groups == 0);
+  // dim 4: [B, C*ch, H, W] (layout 0); dim 5: [C, ch, B, H, W] (layout 1)
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              (x.dim() == 4 || x.dim() == 5));
+  const int layout = x.dim() == 4 ? 0 : 1;
+  int64_t B, ch, HW;
+  if (layout == 0) {
+    B = x.size(0); ch = x.size(1) / clients; HW = x.size(2) * x.size(3);
+    TORCH_CHECK(x.size(1) % clients == 0);
+  } else {
+    TORCH_CHECK(x.size(0) == clients);
+    ch = x.size(1); B = x.size(2); HW = x.size(3) * x.size(4);
+  }
+  TORCH_CHECK(ch % groups == 0);
   TORCH_CHECK(ch / groups <= 128, "groupnorm: ch/G > 128 unsupported");
   TORCH_CHECK(gamma.is_contiguous() && beta.is_contiguous());
   const bool has_res = res.numel() > 0;
@@ -140,7 +148,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> groupnorm_fwd(
                     y.data_ptr(), mean.data_ptr<float>(),
                     rstd.data_ptr<float>(), gamma.data_ptr(), beta.data_ptr(),
                     (int)B, (int)clients, (int)ch, (int)groups, (int)HW,
-                    (float)eps, relu, gn_dtype(x), stream.stream());
+                    (float)eps, relu, layout, gn_dtype(x), stream.stream());
   return {y, mean, rstd};
 }
 
@@ -150,9 +158,13 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> groupnorm_bwd(
     bool has_res, bool relu) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && y.is_contiguous());
   auto dyc = dy.contiguous();
-  const int64_t B = x.size(0), cch = x.size(1),
-                HW = x.size(2) * x.size(3);
-  const int64_t ch = cch / clients;
+  const int layout = x.dim() == 4 ? 0 : 1;
+  int64_t B, ch, HW;
+  if (layout == 0) {
+    B = x.size(0); ch = x.size(1) / clients; HW = x.size(2) * x.size(3);
+  } else {
+    ch = x.size(1); B = x.size(2); HW = x.size(3) * x.size(4);
+  }
   auto dx = at::empty_like(x);
   auto dres = has_res ? at::empty_like(x)
                       : at::empty({0}, x.options());
@@ -164,7 +176,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> groupnorm_bwd(
                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
                     gamma.data_ptr(), dgamma.data_ptr<float>(),
                     dbeta.data_ptr<float>(), (int)B, (int)clients, (int)ch,
-                    (int)groups, (int)HW, relu, gn_dtype(x),
+                    (int)groups, (int)HW, relu, layout, gn_dtype(x),
                     stream.stream());
   return {dx, dres, dgamma, dbeta};
 }

@@ -22,21 +22,29 @@
 
 #include "common.h"
 
-template <typename T, bool HAS_RES, bool RELU>
+template <typename T, bool HAS_RES, bool RELU, int LAYOUT>
 __global__ __launch_bounds__(OLS_THREADS) void k_gn_fwd(
     const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
     float* __restrict__ mean_out, float* __restrict__ rstd_out,
     const T* __restrict__ gamma, const T* __restrict__ beta,
-    int C, int ch, int G, int HW, float eps) {
+    int Bn, int C, int ch, int G, int HW, float eps) {
   const int cg = ch / G;
   const int n = cg * HW;
-  const int group = blockIdx.x;             // (b, c, g)
-  const int g = group % G;
-  const int c = (group / G) % C;
-  const int64_t base = (int64_t)group * n -  // contiguous span start:
-      (int64_t)0;                            // [(b*C + c)*ch + g*cg] * HW
-  // NOTE: because groups are enumerated exactly in memory order
-  // ((b*C+c)*G+g), base == group * n.
+  const int group = blockIdx.x;
+  int c, g;
+  int64_t base, pstride;
+  if (LAYOUT == 0) {          // group = (b, c, g); contiguous cg*HW span
+    g = group % G;
+    c = (group / G) % C;
+    base = (int64_t)group * n;
+    pstride = HW;
+  } else {                    // group = (c, g, b); planes strided by B*HW
+    const int b = group % Bn;
+    g = (group / Bn) % G;
+    c = group / (Bn * G);
+    base = (((int64_t)c * ch + (int64_t)g * cg) * Bn + b) * HW;
+    pstride = (int64_t)Bn * HW;
+  }
   const T* xg = x + base;
   const T* rg = HAS_RES ? res + base : nullptr;
   T* yg = y + base;
@@ -44,7 +52,8 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_fwd(
   __shared__ float red[2][OLS_THREADS / WAVE];
   float s1 = 0.f, s2 = 0.f;
   for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    float v = to_f32(xg[i]);
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)(i / HW) * pstride + i % HW);
+    float v = to_f32(xg[idx]);
     s1 += v;
     s2 += v * v;
   }
@@ -64,29 +73,41 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_fwd(
   const T* bet = beta + (int64_t)c * ch + g * cg;
   for (int i = threadIdx.x; i < n; i += blockDim.x) {
     int chan = i / HW;                       // channel within the group
-    float v = (to_f32(xg[i]) - mean) * rstd;
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+    float v = (to_f32(xg[idx]) - mean) * rstd;
     v = v * to_f32(gam[chan]) + to_f32(bet[chan]);
-    if (HAS_RES) v += to_f32(rg[i]);
+    if (HAS_RES) v += to_f32(rg[idx]);
     if (RELU) v = fmaxf(v, 0.f);
-    yg[i] = from_f32<T>(v);
+    yg[idx] = from_f32<T>(v);
   }
 }
 
 #define MAX_CG 128
 
-template <typename T, bool HAS_RES, bool RELU>
+template <typename T, bool HAS_RES, bool RELU, int LAYOUT>
 __global__ __launch_bounds__(OLS_THREADS) void k_gn_bwd(
     const T* __restrict__ x, const T* __restrict__ y,
     const T* __restrict__ dy, T* __restrict__ dx, T* __restrict__ dres,
     const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
     const T* __restrict__ gamma, float* __restrict__ dgamma,
-    float* __restrict__ dbeta, int C, int ch, int G, int HW) {
+    float* __restrict__ dbeta, int Bn, int C, int ch, int G, int HW) {
   const int cg = ch / G;
   const int n = cg * HW;
   const int group = blockIdx.x;
-  const int g = group % G;
-  const int c = (group / G) % C;
-  const int64_t base = (int64_t)group * n;
+  int c, g;
+  int64_t base, pstride;
+  if (LAYOUT == 0) {
+    g = group % G;
+    c = (group / G) % C;
+    base = (int64_t)group * n;
+    pstride = HW;
+  } else {
+    const int b = group % Bn;
+    g = (group / Bn) % G;
+    c = group / (Bn * G);
+    base = (((int64_t)c * ch + (int64_t)g * cg) * Bn + b) * HW;
+    pstride = (int64_t)Bn * HW;
+  }
   const T* xg = x + base;
   const T* yg = y + base;
   const T* dyg_in = dy + base;
@@ -104,10 +125,11 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_bwd(
 
   float s1 = 0.f, s2 = 0.f;
   for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    float grad = to_f32(dyg_in[i]);
-    if (RELU) grad = to_f32(yg[i]) > 0.f ? grad : 0.f;
     int chan = i / HW;
-    float xhat = (to_f32(xg[i]) - mean) * rstd;
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+    float grad = to_f32(dyg_in[idx]);
+    if (RELU) grad = to_f32(yg[idx]) > 0.f ? grad : 0.f;
+    float xhat = (to_f32(xg[idx]) - mean) * rstd;
     float dxhat = grad * to_f32(gam[chan]);
     s1 += dxhat;
     s2 += dxhat * xhat;
@@ -124,13 +146,14 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_bwd(
   const float m1 = s1 / n, m2 = s2 / n;
 
   for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    float grad = to_f32(dyg_in[i]);
-    if (RELU) grad = to_f32(yg[i]) > 0.f ? grad : 0.f;
     int chan = i / HW;
-    float xhat = (to_f32(xg[i]) - mean) * rstd;
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+    float grad = to_f32(dyg_in[idx]);
+    if (RELU) grad = to_f32(yg[idx]) > 0.f ? grad : 0.f;
+    float xhat = (to_f32(xg[idx]) - mean) * rstd;
     float dxhat = grad * to_f32(gam[chan]);
-    dxg[i] = from_f32<T>(rstd * (dxhat - m1 - xhat * m2));
-    if (HAS_RES) drg[i] = from_f32<T>(grad);
+    dxg[idx] = from_f32<T>(rstd * (dxhat - m1 - xhat * m2));
+    if (HAS_RES) drg[idx] = from_f32<T>(grad);
   }
   __syncthreads();
   for (int i = threadIdx.x; i < cg; i += blockDim.x) {
@@ -143,16 +166,21 @@ template <typename T>
 static void launch_fwd(const T* x, const T* res, T* y, float* mean,
                        float* rstd, const T* gamma, const T* beta, int B,
                        int C, int ch, int G, int HW, float eps, bool relu,
-                       hipStream_t s) {
+                       int layout, hipStream_t s) {
   dim3 grid(B * C * G), block(OLS_THREADS);
   const bool has_res = res != nullptr;
-#define CASE(HR, RL)                                                          \
-  hipLaunchKernelGGL((k_gn_fwd<T, HR, RL>), grid, block, 0, s, x, res, y,     \
-                     mean, rstd, gamma, beta, C, ch, G, HW, eps)
-  if (has_res && relu) CASE(true, true);
-  else if (has_res) CASE(true, false);
-  else if (relu) CASE(false, true);
-  else CASE(false, false);
+#define CASE(HR, RL, LY)                                                      \
+  hipLaunchKernelGGL((k_gn_fwd<T, HR, RL, LY>), grid, block, 0, s, x, res,    \
+                     y, mean, rstd, gamma, beta, B, C, ch, G, HW, eps)
+#define PICK(LY)                                                              \
+  do {                                                                        \
+    if (has_res && relu) CASE(true, true, LY);                                \
+    else if (has_res) CASE(true, false, LY);                                  \
+    else if (relu) CASE(false, true, LY);                                     \
+    else CASE(false, false, LY);                                              \
+  } while (0)
+  if (layout == 0) PICK(0); else PICK(1);
+#undef PICK
 #undef CASE
 }
 
@@ -160,16 +188,22 @@ template <typename T>
 static void launch_bwd(const T* x, const T* y, const T* dy, T* dx, T* dres,
                        const float* mean, const float* rstd, const T* gamma,
                        float* dgamma, float* dbeta, int B, int C, int ch,
-                       int G, int HW, bool relu, hipStream_t s) {
+                       int G, int HW, bool relu, int layout, hipStream_t s) {
   dim3 grid(B * C * G), block(OLS_THREADS);
   const bool has_res = dres != nullptr;
-#define CASE(HR, RL)                                                          \
-  hipLaunchKernelGGL((k_gn_bwd<T, HR, RL>), grid, block, 0, s, x, y, dy, dx, \
-                     dres, mean, rstd, gamma, dgamma, dbeta, C, ch, G, HW)
-  if (has_res && relu) CASE(true, true);
-  else if (has_res) CASE(true, false);
-  else if (relu) CASE(false, true);
-  else CASE(false, false);
+#define CASE(HR, RL, LY)                                                      \
+  hipLaunchKernelGGL((k_gn_bwd<T, HR, RL, LY>), grid, block, 0, s, x, y, dy, \
+                     dx, dres, mean, rstd, gamma, dgamma, dbeta, B, C, ch,   \
+                     G, HW)
+#define PICK(LY)                                                              \
+  do {                                                                        \
+    if (has_res && relu) CASE(true, true, LY);                                \
+    else if (has_res) CASE(true, false, LY);                                  \
+    else if (relu) CASE(false, true, LY);                                     \
+    else CASE(false, false, LY);                                              \
+  } while (0)
+  if (layout == 0) PICK(0); else PICK(1);
+#undef PICK
 #undef CASE
 }
 
@@ -177,17 +211,17 @@ extern "C" void ols_groupnorm_fwd(const void* x, const void* res, void* y,
                                   float* mean, float* rstd, const void* gamma,
                                   const void* beta, int B, int C, int ch,
                                   int G, int HW, float eps, bool relu,
-                                  int dtype, hipStream_t stream) {
+                                  int layout, int dtype, hipStream_t stream) {
   if (dtype == 0)
     launch_fwd<float>((const float*)x, (const float*)res, (float*)y, mean,
                       rstd, (const float*)gamma, (const float*)beta, B, C, ch,
-                      G, HW, eps, relu, stream);
+                      G, HW, eps, relu, layout, stream);
   else
     launch_fwd<__hip_bfloat16>((const __hip_bfloat16*)x,
                                (const __hip_bfloat16*)res, (__hip_bfloat16*)y,
                                mean, rstd, (const __hip_bfloat16*)gamma,
                                (const __hip_bfloat16*)beta, B, C, ch, G, HW,
-                               eps, relu, stream);
+                               eps, relu, layout, stream);
 }
 
 extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
@@ -195,17 +229,17 @@ extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const float* mean, const float* rstd,
                                   const void* gamma, float* dgamma,
                                   float* dbeta, int B, int C, int ch, int G,
-                                  int HW, bool relu, int dtype,
+                                  int HW, bool relu, int layout, int dtype,
                                   hipStream_t stream) {
   if (dtype == 0)
     launch_bwd<float>((const float*)x, (const float*)y, (const float*)dy,
                       (float*)dx, (float*)dres, mean, rstd,
                       (const float*)gamma, dgamma, dbeta, B, C, ch, G, HW,
-                      relu, stream);
+                      relu, layout, stream);
   else
     launch_bwd<__hip_bfloat16>(
         (const __hip_bfloat16*)x, (const __hip_bfloat16*)y,
         (const __hip_bfloat16*)dy, (__hip_bfloat16*)dx, (__hip_bfloat16*)dres,
         mean, rstd, (const __hip_bfloat16*)gamma, dgamma, dbeta, B, C, ch, G,
-        HW, relu, stream);
+        HW, relu, layout, stream);
 }

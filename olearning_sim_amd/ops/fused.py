@@ -250,14 +250,24 @@ def groupnorm_act(x: torch.Tensor, clients: int, groups: int,
                   eps: float = 1e-5) -> torch.Tensor:
     """y = relu?(gn(x; clients, groups)*gamma + beta [+ res]).
 
-    GPU: one fused HIP kernel each way.  CPU: composed torch ops
-    (models/base.bgroupnorm semantics)."""
+    x 4-D [B, C*ch, H, W] (channel-grouped) or 5-D [C, ch, B, H, W]
+    (client-channel-first, the conv-kernel layout).  GPU: one fused HIP
+    kernel each way.  CPU: composed torch ops."""
     if x.is_cuda:
         return _GroupNormActFn.apply(x, res, gamma.contiguous(),
                                      beta.contiguous(), clients, groups,
                                      eps, relu)
-    from ..models.base import bgroupnorm
-    y = bgroupnorm(x, clients, groups, gamma, beta, eps)
+    if x.dim() == 5:
+        C, ch, B, H, W = x.shape
+        cg = ch // groups
+        xg = x.view(C, groups, cg, B, H, W)
+        mean = xg.mean(dim=(2, 4, 5), keepdim=True)
+        var = xg.var(dim=(2, 4, 5), unbiased=False, keepdim=True)
+        y = ((xg - mean) * torch.rsqrt(var + eps)).view(C, ch, B, H, W)
+        y = y * gamma.view(C, ch, 1, 1, 1) + beta.view(C, ch, 1, 1, 1)
+    else:
+        from ..models.base import bgroupnorm
+        y = bgroupnorm(x, clients, groups, gamma, beta, eps)
     if res is not None:
         y = y + res
     return torch.nn.functional.relu(y) if relu else y
