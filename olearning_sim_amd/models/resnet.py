@@ -123,7 +123,14 @@ class ResNet18(ClientBatchedModel):
 
     def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
         from ..ops.fused import groupnorm_act
-        if x.is_cuda and x.dtype == torch.bfloat16:
+        # OLSIM_CONV=custom routes 3x3 convs through the hand-written
+        # implicit-GEMM MFMA kernels (ops/csrc/client_conv.hip).  The
+        # default is MIOpen grouped conv, which currently wins at these
+        # shapes once its Find phase has run (see profiles/): custom
+        # ~70 TF/s vs CK ~110-150 TF/s steady state.
+        import os
+        if x.is_cuda and x.dtype == torch.bfloat16 and \
+                os.environ.get("OLSIM_CONV", "") == "custom":
             from ..ops.fused import hip_ops_available
             if hip_ops_available():
                 return self.forward_cbf(params, x)

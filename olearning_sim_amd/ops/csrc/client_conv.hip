@@ -69,7 +69,27 @@ __global__ __launch_bounds__(64) void k_mfma_selftest(
 
 struct ConvGeom {
   int B, H, W, OH, OW, IC, OC, stride;
+  // log2 of the spatial sizes when all are powers of two (CIFAR-style
+  // 32/16/8/4 planes) — lets the implicit-GEMM gather decompose n with
+  // shifts instead of ~40-instruction integer divisions per element.
+  int lg_ow, lg_ohw, pow2;
 };
+
+__device__ __forceinline__ void n_decomp(const ConvGeom& g, int n, int& b,
+                                         int& oh, int& ow) {
+  if (g.pow2) {
+    b = n >> g.lg_ohw;
+    int q = n & ((1 << g.lg_ohw) - 1);
+    oh = q >> g.lg_ow;
+    ow = q & ((1 << g.lg_ow) - 1);
+  } else {
+    int OHW = g.OH * g.OW;
+    b = n / OHW;
+    int q = n % OHW;
+    oh = q / g.OW;
+    ow = q % g.OW;
+  }
+}
 
 // stage the fwd/wgrad patch tile P[k][n] (k in [k0,k0+BK), n in
 // [n0,n0+BN)) into lds[BK][BN]; k=(ic,dh,dw) w/ dw fastest.
@@ -87,8 +107,8 @@ __device__ void stage_patch(const __hip_bfloat16* __restrict__ x,
     if (k < kmax && n < nmax) {
       int ic = k / 9, r = k % 9;
       int dh = r / 3, dw = r % 3;
-      int b = n / OHW, q = n % OHW;
-      int oh = q / g.OW, ow = q % g.OW;
+      int b, oh, ow;
+      n_decomp(g, n, b, oh, ow);
       int ih = oh * g.stride + dh - 1, iw = ow * g.stride + dw - 1;
       if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
         v = to_f32(x[((int64_t)ic * g.B + b) * HW + ih * g.W + iw]);
@@ -205,7 +225,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad(
         int oc = k / 9, r = k % 9;
         int dh = r / 3, dw = r % 3;
         int b = n / HW, q = n % HW;
-        int h = q / g.W, wd = q % g.W;
+        int h = q / g.W, wd = q % g.W;   // dgrad walks input planes (pow2 too but cheap enough here)
         int num_h = h + dh - 1, num_w = wd + dw - 1;
         if (num_h >= 0 && num_w >= 0 && num_h % g.stride == 0
             && num_w % g.stride == 0) {
@@ -285,8 +305,8 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad(
       if (q < NN && k < K9) {
         int ic = k / 9, r = k % 9;
         int dh = r / 3, dw2 = r % 3;
-        int b = q / OHW, p = q % OHW;
-        int oh = p / g.OW, ow = p % g.OW;
+        int b, oh, ow;
+        n_decomp(g, q, b, oh, ow);
         int ih = oh * g.stride + dh - 1, iw = ow * g.stride + dw2 - 1;
         if (ih >= 0 && ih < g.H && iw >= 0 && iw < g.W)
           v = to_f32(xc[((int64_t)ic * g.B + b) * HW + ih * g.W + iw]);
@@ -321,6 +341,126 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// forward v2: BN=128, B tile stored transposed [BN][BK+4] so a B
+// fragment is two aligned ds_read_b64, gathers hoisted per (k-row).
+#define CV2_BM 64
+#define CV2_BN 128
+#define CV2_BK 32
+#define CV2_PAD 8
+typedef __attribute__((ext_vector_type(4))) short bf16x4;
+
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v2(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ y, ConvGeom g) {
+  __shared__ short a_lds[CV2_BM * CV2_BK];
+  __shared__ short bT_lds[CV2_BN * (CV2_BK + CV2_PAD)];
+  const int c = blockIdx.z;
+  const int m0 = blockIdx.y * CV2_BM;
+  const int n0 = blockIdx.x * CV2_BN;
+  const int K = g.IC * 9;
+  const int N = g.B * g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const int OHW = g.OH * g.OW;
+  const __hip_bfloat16* xc = x + (int64_t)c * g.IC * g.B * HW;
+  const __hip_bfloat16* wc = w + (int64_t)c * g.OC * K;
+  __hip_bfloat16* yc = y + (int64_t)c * g.OC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  f32x4 acc[CV2_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV2_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  // B staging: thread stages row kk, 16 consecutive n; A staging: row amm,
+  // 8 consecutive k.  All loads are UNCONDITIONAL (clamped address +
+  // multiplicative mask) so they pipeline, and double-buffered in
+  // registers: loads for step k0+BK issue before the MFMAs of step k0.
+  const int kk = threadIdx.x % CV2_BK;
+  const int nn0 = (threadIdx.x / CV2_BK) * (CV2_BN / 8);
+  const int amm = threadIdx.x / 4;
+  const int ak0 = (threadIdx.x % 4) * 8;
+
+  // hoisted n-decomposition for this thread's 16-column segment
+  int nseg[2], ohseg[2], owseg[2];  // segments the 16 cols may span
+  // (generic per-element decomposition below; segments unused when
+  // OW>=16 — kept simple: recompute per load round, it is ALU not loads)
+
+  short areg[8];
+  short breg[CV2_BN / 8];  // 16 shorts; compiler packs pairs
+
+  auto gather = [&](int k0) {
+    // A
+    int m = m0 + amm;
+    const __hip_bfloat16* wrow = wc + (int64_t)min(m, g.OC - 1) * K;
+    bool mvalid = m < g.OC;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int k = k0 + ak0 + j;
+      float v = to_f32(wrow[min(k, K - 1)]);
+      areg[j] = bf16_bits((mvalid && k < K) ? v : 0.f);
+    }
+    // B
+    int k = k0 + kk;
+    int kc = min(k, K - 1);
+    int ic = kc / 9, r = kc % 9;
+    int dh = r / 3, dw = r % 3;
+    const __hip_bfloat16* plane = xc + (int64_t)ic * g.B * HW;
+    bool kvalid = k < K;
+#pragma unroll
+    for (int j = 0; j < CV2_BN / 8; ++j) {
+      int n = n0 + nn0 + j;
+      int nc = min(n, N - 1);
+      int b, oh, ow;
+      n_decomp(g, nc, b, oh, ow);
+      int ih = oh * g.stride + dh - 1, iw = ow * g.stride + dw - 1;
+      bool ok = kvalid && n < N && ih >= 0 && ih < g.H && iw >= 0 && iw < g.W;
+      int ihc = min(max(ih, 0), g.H - 1), iwc = min(max(iw, 0), g.W - 1);
+      float v = to_f32(plane[(int64_t)b * HW + ihc * g.W + iwc]);
+      breg[j] = ok ? bf16_bits(v) : (short)0;
+    }
+  };
+
+  auto commit = [&]() {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      a_lds[amm * CV2_BK + ak0 + j] = areg[j];
+#pragma unroll
+    for (int j = 0; j < CV2_BN / 8; ++j)
+      bT_lds[(nn0 + j) * (CV2_BK + CV2_PAD) + kk] = breg[j];
+  };
+
+  gather(0);
+  for (int k0 = 0; k0 < K; k0 += CV2_BK) {
+    commit();
+    __syncthreads();
+    if (k0 + CV2_BK < K) gather(k0 + CV2_BK);   // loads overlap the MFMAs
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(
+        &a_lds[(wave * 16 + (lane & 15)) * CV2_BK + 8 * (lane >> 4)]);
+#pragma unroll
+    for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[(nt * 16 + (lane & 15)) * (CV2_BK + CV2_PAD)
+                  + 8 * (lane >> 4)]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        yc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 extern "C" void ols_mfma_selftest(const void* A, const void* B, float* D,
                                   hipStream_t stream) {
@@ -334,9 +474,18 @@ extern "C" void ols_conv3x3_fwd(const void* x, const void* w, void* y, int C,
                                 int IC, int OC, int B, int H, int W,
                                 int stride, hipStream_t stream) {
   ConvGeom g{B, H, W, (H + stride - 1) / stride, (W + stride - 1) / stride,
-             IC, OC, stride};
-  dim3 grid(ceil_div(B * g.OH * g.OW, CONV_BN), ceil_div(OC, CONV_BM), C);
-  hipLaunchKernelGGL(k_conv3x3_fwd, grid, dim3(CONV_THREADS), 0, stream,
+             IC, OC, stride, 0, 0, 0};
+  {
+    auto is_p2 = [](int v) { return v > 0 && (v & (v - 1)) == 0; };
+    auto lg = [](int v) { int l = 0; while ((1 << l) < v) ++l; return l; };
+    if (is_p2(g.OW) && is_p2(g.OH)) {
+      g.lg_ow = lg(g.OW);
+      g.lg_ohw = lg(g.OW) + lg(g.OH);
+      g.pow2 = 1;
+    }
+  }
+  dim3 grid(ceil_div(B * g.OH * g.OW, CV2_BN), ceil_div(OC, CV2_BM), C);
+  hipLaunchKernelGGL(k_conv3x3_fwd_v2, grid, dim3(CONV_THREADS), 0, stream,
                      (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
                      (__hip_bfloat16*)y, g);
 }
@@ -345,7 +494,16 @@ extern "C" void ols_conv3x3_dgrad(const void* dy, const void* w, void* dx,
                                   int C, int IC, int OC, int B, int H, int W,
                                   int stride, hipStream_t stream) {
   ConvGeom g{B, H, W, (H + stride - 1) / stride, (W + stride - 1) / stride,
-             IC, OC, stride};
+             IC, OC, stride, 0, 0, 0};
+  {
+    auto is_p2 = [](int v) { return v > 0 && (v & (v - 1)) == 0; };
+    auto lg = [](int v) { int l = 0; while ((1 << l) < v) ++l; return l; };
+    if (is_p2(g.OW) && is_p2(g.OH)) {
+      g.lg_ow = lg(g.OW);
+      g.lg_ohw = lg(g.OW) + lg(g.OH);
+      g.pow2 = 1;
+    }
+  }
   dim3 grid(ceil_div(B * H * W, CONV_BN), ceil_div(IC, CONV_BM), C);
   hipLaunchKernelGGL(k_conv3x3_dgrad, grid, dim3(CONV_THREADS), 0, stream,
                      (const __hip_bfloat16*)dy, (const __hip_bfloat16*)w,
@@ -356,7 +514,16 @@ extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
                                   int C, int IC, int OC, int B, int H, int W,
                                   int stride, hipStream_t stream) {
   ConvGeom g{B, H, W, (H + stride - 1) / stride, (W + stride - 1) / stride,
-             IC, OC, stride};
+             IC, OC, stride, 0, 0, 0};
+  {
+    auto is_p2 = [](int v) { return v > 0 && (v & (v - 1)) == 0; };
+    auto lg = [](int v) { int l = 0; while ((1 << l) < v) ++l; return l; };
+    if (is_p2(g.OW) && is_p2(g.OH)) {
+      g.lg_ow = lg(g.OW);
+      g.lg_ohw = lg(g.OW) + lg(g.OH);
+      g.pow2 = 1;
+    }
+  }
   dim3 grid(ceil_div(IC * 9, CONV_BN), ceil_div(OC, CONV_BM), C);
   hipLaunchKernelGGL(k_conv3x3_wgrad, grid, dim3(CONV_THREADS), 0, stream,
                      (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, dw, g);
