@@ -1,0 +1,142 @@
+"""Gradient-house lifecycle + message-plane tests
+(reference deviceflow_server.py / sorter.py / dispatcher.py semantics)."""
+
+import json
+import time
+
+import pytest
+
+from olearning_sim_amd.deviceflow.service import DeviceFlowService
+
+
+def rt_strategy(batch=5, drop_p=0.0):
+    return json.dumps({"real_time_dispatch": {
+        "use_strategy": True, "dispatch_batch_sizes": [batch],
+        "drop_simulation": {"drop_probability": drop_p}}})
+
+
+def flow_strategy(total, drop_p=None):
+    spec = {"flow_dispatch": {
+        "use_strategy": True, "total_dispatch_amount": total,
+        "specific_interval": {
+            "use": True, "intervals": [[0, 5]],
+            "dispatch_rules": {"domains": [[0.0, 5.0]], "functions": ["1"]}}}}
+    if drop_p is not None:
+        spec["flow_dispatch"]["specific_interval"]["drop_simulation"] = \
+            {"drop_probability": [drop_p]}
+    return json.dumps(spec)
+
+
+def wait_until(cond, timeout=10.0):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if cond():
+            return True
+        time.sleep(0.01)
+    return False
+
+
+@pytest.fixture
+def svc():
+    s = DeviceFlowService(time_scale=0.0, seed=1)
+    yield s
+    s.shutdown()
+
+
+def test_notify_start_requires_registration(svc):
+    assert svc.notify_start("nope", "train", 0, "logical_simulation") is None
+    svc.register_task("t", ["logical_simulation"])
+    fid = svc.notify_start("t", "train", 0, "logical_simulation",
+                           strategy=rt_strategy())
+    assert fid == "t_train_0"
+
+
+def test_duplicate_registration_rejected(svc):
+    assert svc.register_task("t", ["logical_simulation"])
+    assert not svc.register_task("t", ["logical_simulation"])
+    assert svc.unregister_task("t")
+    assert svc.register_task("t", ["logical_simulation"])
+
+
+def test_sorter_discards_unstarted_flow_messages(svc):
+    svc.register_task("t", ["logical_simulation"])
+    svc.publish("t_train_0", "logical_simulation")   # before NotifyStart
+    time.sleep(0.2)
+    assert svc.shelf.depth("t_train_0") == 0
+    assert svc.outbound.qsize() == 0
+
+
+def test_real_time_flow_forwards_messages(svc):
+    svc.register_task("t", ["logical_simulation"])
+    svc.notify_start("t", "train", 0, "logical_simulation",
+                     strategy=rt_strategy(batch=5))
+    for _ in range(12):
+        svc.publish("t_train_0", "logical_simulation", payload={"g": 1})
+    assert wait_until(lambda: svc.outbound.qsize() >= 10)
+    svc.notify_complete("t", "train", 0, "logical_simulation")
+    assert wait_until(lambda: svc.check_dispatch_finished("t"))
+    assert svc.outbound.qsize() == 12    # remainder flushed on release
+    assert svc.flow_release_step() == ["t_train_0"]
+
+
+def test_real_time_drop_probability(svc):
+    svc.register_task("t", ["logical_simulation"])
+    svc.notify_start("t", "train", 0, "logical_simulation",
+                     strategy=rt_strategy(batch=10, drop_p=1.0))
+    for _ in range(20):
+        svc.publish("t_train_0", "logical_simulation")
+    assert wait_until(lambda: svc.inbound.qsize() == 0)
+    svc.notify_complete("t", "train", 0, "logical_simulation")
+    assert wait_until(lambda: svc.check_dispatch_finished("t"))
+    d = None
+    # dispatcher already reaped or still present; count via stats
+    assert svc.outbound.qsize() == 0
+
+
+def test_flow_mode_dispatches_on_complete(svc):
+    svc.register_task("t", ["logical_simulation"])
+    svc.notify_start("t", "train", 0, "logical_simulation",
+                     strategy=flow_strategy(total=50))
+    for _ in range(50):
+        svc.publish("t_train_0", "logical_simulation")
+    assert wait_until(lambda: svc.shelf.depth("t_train_0") == 50)
+    assert svc.outbound.qsize() == 0     # flow mode waits for complete
+    svc.notify_complete("t", "train", 0, "logical_simulation")
+    assert wait_until(lambda: svc.outbound.qsize() == 50)
+
+
+def test_hybrid_two_resources_must_both_complete(svc):
+    svc.register_task("t", ["logical_simulation", "device_simulation"])
+    svc.notify_start("t", "train", 0, "logical_simulation",
+                     strategy=flow_strategy(total=10))
+    svc.notify_start("t", "train", 0, "device_simulation")
+    for _ in range(10):
+        svc.publish("t_train_0", "logical_simulation")
+    # messages must be absorbed before NotifyComplete (the sorter
+    # rejects post-complete messages, like the reference's)
+    assert wait_until(lambda: svc.shelf.depth("t_train_0") == 10)
+    svc.notify_complete("t", "train", 0, "logical_simulation")
+    time.sleep(0.2)
+    assert svc.outbound.qsize() == 0     # device side not complete yet
+    svc.notify_complete("t", "train", 0, "device_simulation")
+    assert wait_until(lambda: svc.outbound.qsize() == 10)
+
+
+def test_unregister_releases_flows(svc):
+    svc.register_task("t", ["logical_simulation"])
+    svc.notify_start("t", "train", 0, "logical_simulation",
+                     strategy=rt_strategy())
+    svc.unregister_task("t")
+    assert svc.check_dispatch_finished("t")
+    assert not svc.registry.is_registered("t")
+
+
+def test_registry_persists_across_restart(tmp_path):
+    db = str(tmp_path / "df.sqlite")
+    s1 = DeviceFlowService(db_path=db, time_scale=0)
+    s1.register_task("t", ["logical_simulation"])
+    s1.shutdown()
+    s2 = DeviceFlowService(db_path=db, time_scale=0)
+    assert s2.registry.is_registered("t")
+    assert s2.registry.resources("t") == ["logical_simulation"]
+    s2.shutdown()
