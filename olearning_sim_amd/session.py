@@ -1,0 +1,88 @@
+"""Simulator session: compose the services into one process.
+
+Parity with the reference's SimulatorSession
+(ols_core/simu_session.py:25-71): one server process mounts the chosen
+services by `svc` code — 0 mounts everything, 1 TaskMgr, 2 ResourceMgr,
+3 DeviceFlow, 4 PerformanceMgr (the reference mounts PerformanceMgr for
+svc in {0,4}; its RayClusterMgr is commented out, here the
+NodeClusterManager is mounted with svc 0).  `serve()` exposes the JSON
+API over HTTP (api/server.py) — the reference's gRPC surface — while
+in-process use needs no server at all.
+"""
+
+from __future__ import annotations
+
+import os
+import threading
+from typing import Optional
+
+from .cluster import NodeClusterManager
+from .deviceflow.service import DeviceFlowService
+from .perf import PerformanceManager
+from .resource.manager import ResourceManager
+from .task.manager import TaskManager
+from .task.runner import TaskRunner
+from .task.table import TaskTableRepo
+from .utils.logging import Logger
+
+
+class SimulatorSession:
+    def __init__(self, svc: int = 0, data_dir: Optional[str] = None,
+                 device: str = "", auto_start_threads: bool = True):
+        self.svc = svc
+        self.data_dir = data_dir or os.path.join(
+            os.path.expanduser("~"), ".olearning_sim_amd")
+        os.makedirs(self.data_dir, exist_ok=True)
+        self.log = Logger.shared()
+
+        def db(name: str) -> str:
+            return os.path.join(self.data_dir, name)
+
+        if not device:
+            try:
+                import torch
+                device = "cuda:0" if torch.cuda.is_available() else "cpu"
+            except Exception:
+                device = "cpu"
+
+        self.resource_mgr = (ResourceManager(db("resmgr.sqlite"))
+                             if svc in (0, 1, 2) else None)
+        self.deviceflow = (DeviceFlowService(db("deviceflow.sqlite"))
+                           if svc in (0, 3) else None)
+        self.performance_mgr = (PerformanceManager(db("perf.sqlite"))
+                                if svc in (0, 4) else None)
+        self.cluster_mgr = NodeClusterManager() if svc == 0 else None
+        self.task_mgr = None
+        if svc in (0, 1):
+            table = TaskTableRepo(db("taskmgr.sqlite"))
+            runner = TaskRunner(
+                table, device=device,
+                checkpoint_dir=os.path.join(self.data_dir, "checkpoints"),
+                deviceflow=self.deviceflow)
+            self.task_mgr = TaskManager(
+                table=table, resource_mgr=self.resource_mgr, runner=runner,
+                deviceflow=self.deviceflow, auto_start=auto_start_threads)
+
+    def serve(self, host: str = "127.0.0.1", port: int = 60061,
+              block: bool = True):
+        """Serve the JSON API (reference: grpc server on the session
+        port).  Returns the uvicorn server when block=False."""
+        import uvicorn
+        from .api.server import build_app
+        app = build_app(self)
+        config = uvicorn.Config(app, host=host, port=port, log_level="warning")
+        server = uvicorn.Server(config)
+        if block:
+            server.run()
+            return server
+        t = threading.Thread(target=server.run, daemon=True)
+        t.start()
+        return server
+
+    def shutdown(self) -> None:
+        if self.task_mgr is not None:
+            self.task_mgr.shutdown()
+        if self.deviceflow is not None:
+            self.deviceflow.shutdown()
+        if self.cluster_mgr is not None:
+            self.cluster_mgr.shutdown()

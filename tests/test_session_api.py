@@ -1,0 +1,124 @@
+"""SimulatorSession composition + JSON API surface (reference
+simu_session.py + the six proto service surfaces)."""
+
+import json
+import time
+
+import pytest
+from fastapi.testclient import TestClient
+
+from olearning_sim_amd.session import SimulatorSession
+from olearning_sim_amd.api.server import build_app
+
+from test_manager import task_json
+
+
+@pytest.fixture
+def session(tmp_path):
+    s = SimulatorSession(svc=0, data_dir=str(tmp_path), device="cpu",
+                         auto_start_threads=False)
+    yield s
+    s.shutdown()
+
+
+@pytest.fixture
+def client(session):
+    return TestClient(build_app(session))
+
+
+def test_health_and_composition(client, session):
+    assert client.get("/health").json()["ok"]
+    assert session.task_mgr is not None
+    assert session.resource_mgr is not None
+    assert session.deviceflow is not None
+    assert session.performance_mgr is not None
+
+
+def test_svc_codes(tmp_path):
+    s = SimulatorSession(svc=2, data_dir=str(tmp_path / "a"))
+    assert s.task_mgr is None and s.resource_mgr is not None
+    s.shutdown()
+    s = SimulatorSession(svc=4, data_dir=str(tmp_path / "b"))
+    assert s.performance_mgr is not None and s.task_mgr is None
+    s.shutdown()
+
+
+def test_submit_and_run_task_via_api(client, session):
+    body = {"task": json.loads(task_json(task_id="t_api"))}
+    r = client.post("/taskmgr/submitTask", json=body).json()
+    assert r["is_success"], r
+    assert client.get("/taskmgr/getTaskQueue").json()["tasks"] == ["t_api"]
+    assert client.get("/taskmgr/getTaskStatus/t_api").json()[
+        "task_status"] == "QUEUED"
+    # drive the scheduler manually (threads off in tests)
+    assert session.task_mgr.step_schedule() == "t_api"
+    t0 = time.time()
+    while time.time() - t0 < 30:
+        st = client.get("/taskmgr/getTaskStatus/t_api").json()["task_status"]
+        if st in ("SUCCEEDED", "FAILED", "STOPPED"):
+            break
+        time.sleep(0.05)
+    assert st == "SUCCEEDED"
+
+
+def test_invalid_task_rejected_via_api(client):
+    bad = json.loads(task_json(task_id="t_bad"))
+    bad["target"]["priority"] = 99
+    r = client.post("/taskmgr/submitTask", json={"task": bad}).json()
+    assert not r["is_success"]
+
+
+def test_resource_api_cycle(client):
+    before = client.get("/resourcemgr/getResource").json()
+    assert "logical_simulation" in before
+    ok = client.post("/resourcemgr/requestResource", json={
+        "task_id": "r1", "cpu": 1.0, "mem": 1.0}).json()["is_success"]
+    assert ok
+    after = client.get("/resourcemgr/getResource").json()
+    assert after["logical_simulation"]["cpu"] == \
+        before["logical_simulation"]["cpu"] - 1.0
+    assert client.post("/resourcemgr/releaseResource/r1").json()["is_success"]
+
+
+def test_deviceflow_api_lifecycle(client):
+    assert client.post("/deviceflow/RegisterTask", json={
+        "task_id": "t", "total_compute_resources": ["logical_simulation"],
+    }).json()["is_success"]
+    r = client.post("/deviceflow/NotifyStart", json={
+        "task_id": "t", "operator_name": "train", "round": 0,
+        "compute_resource": "logical_simulation",
+        "strategy": json.dumps({"real_time_dispatch": {
+            "use_strategy": True, "dispatch_batch_sizes": [1]}})}).json()
+    assert r["is_success"] and r["flow_id"] == "t_train_0"
+    assert client.post("/deviceflow/NotifyComplete", json={
+        "task_id": "t", "operator_name": "train", "round": 0,
+        "compute_resource": "logical_simulation"}).json()["is_success"]
+    t0 = time.time()
+    while time.time() - t0 < 10:
+        if client.get("/deviceflow/CheckDeviceflowDispatchFinished/t"
+                      ).json()["is_finished"]:
+            break
+        time.sleep(0.05)
+    assert client.post("/deviceflow/UnRegisterTask/t").json()["is_success"]
+
+
+def test_cluster_api(client):
+    assert client.post("/cluster/create", json={
+        "name": "grp", "replicas": 2}).json()["is_success"]
+    info = client.get("/cluster/get/grp").json()
+    assert info["replicas"] == 2 and info["status"] == "registered"
+    assert client.post("/cluster/updateReplicas/grp/4").json()["is_success"]
+    assert client.get("/cluster/get/grp").json()["replicas"] == 4
+    assert client.get("/cluster/list").json()["clusters"] == ["grp"]
+    assert client.post("/cluster/delete/grp").json()["is_success"]
+
+
+def test_performance_api(client, session):
+    session.performance_mgr.record_round("t", 0, 1.5, 30, loss=2.0)
+    session.performance_mgr.record_round("t", 1, 1.0, 30, loss=1.5)
+    s = client.get("/performancemgr/summary/t").json()
+    assert s["metrics"]["round_time_s"]["count"] == 2
+    assert s["metrics"]["clients_per_s"]["last"] == 30.0
+    rows = client.get("/performancemgr/metrics/t",
+                      params={"metric": "loss"}).json()["metrics"]
+    assert [r["value"] for r in rows] == [2.0, 1.5]
