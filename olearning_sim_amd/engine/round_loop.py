@@ -129,14 +129,27 @@ class LogicalEngine:
         weights_all = (~active_drop).float()
         losses: List[float] = []
         trained = 0
-        chunk = self._chunk_size(max(1, int(active_ids.numel())))
-        for lo in range(0, int(active_ids.numel()), chunk):
-            cid = active_ids[lo:lo + chunk].to(self.device)
+        n_active = int(active_ids.numel())
+        # chunk size derives from the FIXED cohort, not this round's
+        # churn-varying active count, so kernel shapes stay constant
+        chunk = self._chunk_size(min(job.resolved_cohort(), job.clients))
+        for lo in range(0, n_active, chunk):
+            cid = active_ids[lo:lo + chunk]
             w_cpu = weights_all[lo:lo + chunk]
+            if cid.numel() < chunk and n_active > 0:
+                # pad the tail chunk to the full chunk size with
+                # zero-weight repeats: every chunk then has the SAME
+                # shape, so shape-specialised kernels (MIOpen Find,
+                # captured graphs) run once per job instead of once per
+                # round as churn varies the cohort
+                pad = chunk - int(cid.numel())
+                cid = torch.cat([cid, cid[-1:].repeat(pad)])
+                w_cpu = torch.cat([w_cpu, torch.zeros(pad)])
             w = w_cpu.to(self.device)
-            stats = self.trainer.train_chunk(cid, w, round_idx, self._delta,
+            stats = self.trainer.train_chunk(cid.to(self.device), w,
+                                             round_idx, self._delta,
                                              wsum=float(w_cpu.sum()))
-            trained += stats["clients"]
+            trained += min(stats["clients"], n_active - lo)
             if stats["loss"]:
                 losses.append(stats["loss"])
 

@@ -83,14 +83,17 @@ class BertLM(ClientBatchedModel):
         emb = emb + params["emb.pos"][:, :L].unsqueeze(1)
         hs = blayernorm(emb, params["emb.ln.g"], params["emb.ln.b"])
 
-        causal = torch.ones(L, L, dtype=torch.bool, device=x.device).tril()
         for i in range(self.layers):
             pre = f"l{i}"
             qkv = blinear(hs.view(C, B * L, h), params[f"{pre}.qkv.w"],
                           params[f"{pre}.qkv.b"])          # [C, B*L, 3H]
             qkv = qkv.view(C * B, L, 3, nh, hd).permute(2, 0, 3, 1, 4)
-            q, k, v = qkv[0], qkv[1], qkv[2]               # [C*B, nh, L, hd]
-            att = F.scaled_dot_product_attention(q, k, v, attn_mask=causal)
+            # contiguous q/k/v: strided slices of the permuted qkv trip
+            # ROCm SDPA kernels at large C*B
+            q = qkv[0].contiguous()
+            k = qkv[1].contiguous()
+            v = qkv[2].contiguous()                        # [C*B, nh, L, hd]
+            att = F.scaled_dot_product_attention(q, k, v, is_causal=True)
             att = att.transpose(1, 2).reshape(C, B * L, h)
             hs = hs + blinear(att, params[f"{pre}.attn_out.w"],
                               params[f"{pre}.attn_out.b"]).view(C, B, L, h)
