@@ -106,7 +106,12 @@ class BertLM(ClientBatchedModel):
             hs = blayernorm(hs, params[f"{pre}.ln2.g"], params[f"{pre}.ln2.b"])
 
         # tied LM head: logits = hs @ emb^T + bias  -> [C, B, L, V]
-        logits = torch.bmm(hs.view(C, B * L, h), tok.transpose(1, 2))
+        # NOTE: bmm with a strided-transposed B of this width segfaults
+        # in hipBLASLt (ROCm 7.0 torch: [C,768,30522] B^T view faults on
+        # gfx950) — materialise the transpose; the copy is bandwidth-
+        # cheap relative to the head GEMM itself.
+        logits = torch.bmm(hs.view(C, B * L, h),
+                           tok.transpose(1, 2).contiguous())
         logits = logits + params["head.bias"].unsqueeze(1)
         return logits.view(C, B, L, self.vocab_size)
 
