@@ -94,13 +94,16 @@ def kaiming(shape, fan_in: int, device, dtype, generator) -> torch.Tensor:
 
 def blinear(x: torch.Tensor, w: torch.Tensor,
             b: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """Per-client linear: x [C,B,in] @ w[C,out,in]^T + b[C,out] -> [C,B,out].
+    """Per-client linear: x [C,B,in] @ w[C,in,out] + b[C,out] -> [C,B,out].
 
-    Runs as one batched GEMM over the client dimension (hipBLASLt bmm on
-    ROCm; the MFMA batched-linear HIP kernel replaces this on the hot
-    path once loaded).
+    One batched GEMM over the client dimension (hipBLASLt bmm -> MFMA on
+    gfx950).  Weights are stored [in, out] so BOTH operands are
+    contiguous: bmm with a strided-transposed B operand memory-faults in
+    hipBLASLt at large batch counts on this stack (ROCm 7.0 torch,
+    gfx950) — see models/bert.py LM-head note for the one place a
+    transpose is unavoidable.
     """
-    y = torch.bmm(x, w.transpose(1, 2))
+    y = torch.bmm(x, w)
     if b is not None:
         y = y + b.unsqueeze(1)
     return y

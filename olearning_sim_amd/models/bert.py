@@ -44,7 +44,8 @@ class BertLM(ClientBatchedModel):
         p: Params = {}
 
         def lin(name, out_f, in_f):
-            p[f"{name}.w"] = binit((out_f, in_f), in_f, device, dtype, g)
+            # [in, out] layout: blinear needs no transpose (see base.py)
+            p[f"{name}.w"] = binit((in_f, out_f), in_f, device, dtype, g)
             p[f"{name}.b"] = binit((out_f,), in_f, device, dtype, g)
 
         def ln(name):

@@ -31,11 +31,11 @@ class LeNet5(ClientBatchedModel):
             "conv1.b": binit((6,), ic * 25, device, dtype, g),
             "conv2.w": kaiming((16, 6, 5, 5), 6 * 25, device, dtype, g),
             "conv2.b": binit((16,), 6 * 25, device, dtype, g),
-            "fc1.w": binit((120, 400), 400, device, dtype, g),
+            "fc1.w": binit((400, 120), 400, device, dtype, g),
             "fc1.b": binit((120,), 400, device, dtype, g),
-            "fc2.w": binit((84, 120), 120, device, dtype, g),
+            "fc2.w": binit((120, 84), 120, device, dtype, g),
             "fc2.b": binit((84,), 120, device, dtype, g),
-            "fc3.w": binit((k, 84), 84, device, dtype, g),
+            "fc3.w": binit((84, k), 84, device, dtype, g),
             "fc3.b": binit((k,), 84, device, dtype, g),
         }
 

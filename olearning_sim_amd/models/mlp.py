@@ -27,9 +27,9 @@ class MLP(ClientBatchedModel):
                     generator: Optional[torch.Generator] = None) -> Params:
         d, h, k = self.in_features, self.hidden, self.num_classes
         return {
-            "fc1.w": binit((h, d), d, device, dtype, generator),
+            "fc1.w": binit((d, h), d, device, dtype, generator),
             "fc1.b": binit((h,), d, device, dtype, generator),
-            "fc2.w": binit((k, h), h, device, dtype, generator),
+            "fc2.w": binit((h, k), h, device, dtype, generator),
             "fc2.b": binit((k,), h, device, dtype, generator),
         }
 

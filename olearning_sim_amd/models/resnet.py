@@ -64,7 +64,7 @@ class ResNet18(ClientBatchedModel):
                     conv(f"{pre}.down", out_c, ic, 1)
                     gn(f"{pre}.gndown", out_c)
             in_c = out_c
-        p["fc.w"] = binit((self.num_classes, w[3]), w[3], device, dtype, g)
+        p["fc.w"] = binit((w[3], self.num_classes), w[3], device, dtype, g)
         p["fc.b"] = binit((self.num_classes,), w[3], device, dtype, g)
         return p
 

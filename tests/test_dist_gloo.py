@@ -43,7 +43,10 @@ def _worker(rank, world, port, conn):
 
 @pytest.mark.timeout(120)
 def test_two_rank_gloo_round_aggregates():
-    port = 29801
+    import socket
+    with socket.socket() as s:      # pick a free rendezvous port
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
     ctx = mp.get_context("spawn")
     pipes, procs = [], []
     for rank in range(2):
