@@ -92,17 +92,48 @@ def kaiming(shape, fan_in: int, device, dtype, generator) -> torch.Tensor:
     return t.to(dtype)
 
 
+class _BLinearFn(torch.autograd.Function):
+    """bmm-based per-client linear with layout-safe backward.
+
+    hipBLASLt (ROCm 7.0 torch, gfx950) memory-faults when a bmm operand
+    is a strided transposed view at large batch counts — which is
+    exactly what autograd's default bmm backward produces (x^T, w^T
+    views).  This Function materialises the transposes, so every GEMM
+    the GPU sees has contiguous operands."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        y = torch.bmm(x, w)
+        if b is not None:
+            y = y + b.unsqueeze(1)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.bmm(dy, w.transpose(1, 2).contiguous())
+        if ctx.needs_input_grad[1]:
+            dw = torch.bmm(x.transpose(1, 2).contiguous(), dy)
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            db = dy.sum(dim=1)
+        return dx, dw, db
+
+
 def blinear(x: torch.Tensor, w: torch.Tensor,
             b: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Per-client linear: x [C,B,in] @ w[C,in,out] + b[C,out] -> [C,B,out].
 
     One batched GEMM over the client dimension (hipBLASLt bmm -> MFMA on
-    gfx950).  Weights are stored [in, out] so BOTH operands are
-    contiguous: bmm with a strided-transposed B operand memory-faults in
-    hipBLASLt at large batch counts on this stack (ROCm 7.0 torch,
-    gfx950) — see models/bert.py LM-head note for the one place a
-    transpose is unavoidable.
-    """
+    gfx950).  Weights are stored [in, out] so the forward needs no
+    transpose; on GPU the custom Function above also keeps the backward
+    GEMM operands contiguous (hipBLASLt strided-view fault)."""
+    if x.is_cuda:
+        return _BLinearFn.apply(x.contiguous(), w.contiguous(), b)
     y = torch.bmm(x, w)
     if b is not None:
         y = y + b.unsqueeze(1)

@@ -106,14 +106,13 @@ class BertLM(ClientBatchedModel):
                               params[f"{pre}.ffn_out.b"]).view(C, B, L, h)
             hs = blayernorm(hs, params[f"{pre}.ln2.g"], params[f"{pre}.ln2.b"])
 
-        # tied LM head: logits = hs @ emb^T + bias  -> [C, B, L, V]
-        # NOTE: bmm with a strided-transposed B of this width segfaults
-        # in hipBLASLt (ROCm 7.0 torch: [C,768,30522] B^T view faults on
-        # gfx950) — materialise the transpose; the copy is bandwidth-
-        # cheap relative to the head GEMM itself.
-        logits = torch.bmm(hs.view(C, B * L, h),
-                           tok.transpose(1, 2).contiguous())
-        logits = logits + params["head.bias"].unsqueeze(1)
+        # tied LM head: logits = hs @ emb^T + bias  -> [C, B, L, V].
+        # The transpose is materialised and the GEMM goes through
+        # blinear so fwd AND bwd operands are contiguous (hipBLASLt
+        # strided-view fault, models/base.py _BLinearFn).
+        tok_t = tok.transpose(1, 2).contiguous()
+        logits = blinear(hs.view(C, B * L, h), tok_t,
+                         params["head.bias"])
         return logits.view(C, B, L, self.vocab_size)
 
     def loss(self, params: Params, x: torch.Tensor,
