@@ -20,8 +20,34 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import shutil
 import sys
+import tempfile
 import time
+
+
+def _seed_miopen_find_db() -> None:
+    """Seed MIOpen's user find-db from the in-repo tuned copy so a fresh
+    process skips the ~40 s per-shape Find phase (the db ships the
+    gfx950 solver choices for the preset conv shapes; MIOpen wants the
+    directory writable, so it is copied to a temp dir)."""
+    if os.environ.get("MIOPEN_USER_DB_PATH"):
+        return
+    src = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                       "olearning_sim_amd", "ops", "miopen_udb")
+    if not os.path.isdir(src):
+        return
+    dst = os.path.join(tempfile.gettempdir(),
+                       f"olsim_miopen_udb_{os.getuid()}")
+    os.makedirs(dst, exist_ok=True)
+    for f in os.listdir(src):
+        target = os.path.join(dst, f)
+        if not os.path.exists(target):
+            shutil.copy2(os.path.join(src, f), target)
+    os.environ["MIOPEN_USER_DB_PATH"] = dst
+
+
+_seed_miopen_find_db()
 
 import torch
 
