@@ -43,6 +43,13 @@ class EngineJob:
     # behaviour simulation (deviceflow): arrival/offline/drop shaping
     behavior_strategy: str = ""     # gradient-house strategy JSON ("" = none)
 
+    # operator-flow round gates (reference flow_setting.start/stop)
+    flow_start_strategy: str = ""
+    flow_stop_strategy: str = ""
+    flow_wait_interval: float = 1.0
+    flow_total_timeout: float = 0.0
+    flow_work_dir: str = ""
+
     # bookkeeping / checkpointing
     checkpoint_dir: str = ""
     model_update_style: str = ""    # e.g. "{task_id}_{current_round}_result_model.safetensors"

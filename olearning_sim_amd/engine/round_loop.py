@@ -76,6 +76,14 @@ class LogicalEngine:
                                        device=str(self.device))
         self.behavior = behavior
 
+        from .operatorflow import OperatorFlow
+        self.flow = OperatorFlow(
+            job.task_id,
+            start_strategy=job.flow_start_strategy,
+            stop_strategy=job.flow_stop_strategy,
+            wait_interval=job.flow_wait_interval,
+            total_timeout=job.flow_total_timeout,
+            work_dir=job.flow_work_dir or job.checkpoint_dir or ".")
         self._delta = self.master.zeros_like_flat()
         self._scalar = torch.zeros(3, dtype=torch.float64, device=self.device)
         # totals across rounds (reference logical_result accounting)
@@ -201,7 +209,9 @@ class LogicalEngine:
         for r in range(job.rounds):
             if self.stop_requested:
                 break
+            self.flow.start(r)         # operator-flow start gate
             rec = self.run_round(r)
+            self.flow.stop(r)          # operator-flow stop gate
             records.append(rec)
             if self.result_sink is not None:
                 self.result_sink(self._round_result(rec))

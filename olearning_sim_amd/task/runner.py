@@ -79,8 +79,15 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
             train_op.operation_behavior_controller.use_gradient_house:
         behavior = train_op.operation_behavior_controller.strategy_gradient_house
 
+    fs = task.operatorflow.flow_setting
     first_data = task.target.data[0] if task.target.data else None
     job = EngineJob(
+        flow_start_strategy=fs.start.logical_simulation.strategy,
+        flow_stop_strategy=fs.stop.logical_simulation.strategy,
+        flow_wait_interval=max(1, fs.stop.logical_simulation.wait_interval
+                               or fs.start.logical_simulation.wait_interval or 1),
+        flow_total_timeout=max(fs.start.logical_simulation.total_timeout,
+                               fs.stop.logical_simulation.total_timeout),
         task_id=task.task_id,
         model_name=params.get("model", "mlp"),
         model_kwargs=params.get("model_kwargs", {}),

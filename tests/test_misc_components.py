@@ -132,3 +132,44 @@ def test_logger_writes_table(tmp_path):
     rows = lg._table.get_rows_where({"task_id": "t1"})
     assert len(rows) == 2
     assert {r["log_type"] for r in rows} == {"info", "error"}
+
+
+def test_operatorflow_gates(tmp_path):
+    import threading, time
+    from olearning_sim_amd.engine.operatorflow import OperatorFlow, GateTimeout
+
+    # empty strategy = immediate
+    OperatorFlow("t").start(0)
+    OperatorFlow("t").stop(0)
+
+    # selection-service gate: stop waits until the round advances
+    state = {"round": 0}
+    flow = OperatorFlow("t", stop_strategy="waiting_for_global_aggregation",
+                        wait_interval=0.02,
+                        selection_round_fn=lambda: state["round"])
+    t = threading.Thread(target=lambda: (time.sleep(0.1),
+                                         state.update(round=1)))
+    t.start()
+    flow.stop(0)          # returns once service round > 0
+    t.join()
+    assert state["round"] == 1
+
+    # flag-file handshake
+    flow = OperatorFlow("t", stop_strategy="sample_and_aggregation",
+                        wait_interval=0.02, work_dir=str(tmp_path))
+    def aggregator():
+        while not (tmp_path / "simulation_finished.txt").exists():
+            time.sleep(0.01)
+        (tmp_path / "aggregation_finished.txt").write_text("done")
+    t = threading.Thread(target=aggregator)
+    t.start()
+    flow.stop(0)
+    t.join()
+    assert not (tmp_path / "aggregation_finished.txt").exists()  # consumed
+
+    # timeout honoured
+    flow = OperatorFlow("t", stop_strategy="waiting_for_global_aggregation",
+                        wait_interval=0.01, total_timeout=0.05,
+                        selection_round_fn=lambda: 0)
+    with pytest.raises(GateTimeout):
+        flow.stop(0)
