@@ -173,3 +173,61 @@ def test_operatorflow_gates(tmp_path):
                         selection_round_fn=lambda: 0)
     with pytest.raises(GateTimeout):
         flow.stop(0)
+
+
+def test_fragment_repo_json_and_tensor():
+    from olearning_sim_amd.deviceflow.rooms import OutboundRoom, Message
+    from olearning_sim_amd.utils.fragments import (JsonFragmentRepo,
+                                                   TensorFragmentRepo)
+    out = OutboundRoom()
+    out.send(Message("t_train_0", "logical_simulation",
+                     payload='{"grad_norm": 1.5}'))
+    import base64
+    out.send(Message("t_train_0", "logical_simulation",
+                     payload=base64.b64encode(b'{"grad_norm": 2.5}')))
+    frags = list(JsonFragmentRepo(out).fragments())
+    assert [f["payload"]["grad_norm"] for f in frags] == [1.5, 2.5]
+
+    out2 = OutboundRoom()
+    t = torch.ones(3)
+    out2.send(Message("t_train_0", "logical_simulation", payload=t))
+    frags = list(TensorFragmentRepo(out2).fragments())
+    assert torch.equal(frags[0]["payload"], t)
+
+
+def test_hybrid_data_splitter(tmp_path):
+    from olearning_sim_amd.utils.data_split import HybridDataSplitter
+    data = tmp_path / "data"
+    data.mkdir()
+    rows = "\n".join(f"{i},{i*2}" for i in range(100))
+    (data / "train.csv").write_text("a,b\n" + rows + "\n")
+    (data / "meta.txt").write_text("labels")
+    sp = HybridDataSplitter(seed=3)
+    nl, nd = sp.split_dir(str(data), str(tmp_path / "lg"), str(tmp_path / "dv"),
+                          device_fraction=0.3)
+    assert nl == 70 and nd == 30
+    lg = (tmp_path / "lg" / "train.csv").read_text().splitlines()
+    dv = (tmp_path / "dv" / "train.csv").read_text().splitlines()
+    assert lg[0] == dv[0] == "a,b"
+    assert len(lg) - 1 == 70 and len(dv) - 1 == 30
+    # every row lands exactly once
+    assert sorted(lg[1:] + dv[1:]) == sorted(rows.split("\n"))
+    assert (tmp_path / "lg" / "meta.txt").exists()
+    assert (tmp_path / "dv" / "meta.txt").exists()
+    # archives round-trip
+    z = sp.archive(str(tmp_path / "lg"), str(tmp_path / "lg.zip"))
+    out = sp.extract(z, str(tmp_path / "back"))
+    assert (tmp_path / "back" / "train.csv").exists()
+
+
+def test_simulator_config_load(tmp_path):
+    from olearning_sim_amd.config import SimulatorConfig
+    cfg = SimulatorConfig.load(None)
+    assert cfg.scheduler_sleep_time == 5.0
+    assert cfg.interrupt_queue_time == 3600.0
+    y = tmp_path / "conf.yaml"
+    y.write_text("scheduler_sleep_time: 1.5\nphone_pool:\n  u1:\n    high: 4\n")
+    cfg = SimulatorConfig.load(str(y))
+    assert cfg.scheduler_sleep_time == 1.5
+    assert cfg.phone_pool == {"u1": {"high": 4}}
+    assert cfg.timers()["release_sleep_time"] == 10.0
