@@ -43,7 +43,10 @@ def _seed_miopen_find_db() -> None:
     for f in os.listdir(src):
         target = os.path.join(dst, f)
         if not os.path.exists(target):
-            shutil.copy2(os.path.join(src, f), target)
+            # atomic publish: 8 ranks race to seed the same directory
+            tmp = target + f".tmp{os.getpid()}"
+            shutil.copy2(os.path.join(src, f), tmp)
+            os.replace(tmp, target)
     os.environ["MIOPEN_USER_DB_PATH"] = dst
 
 

@@ -57,6 +57,21 @@ def hip_ops_available() -> bool:
     return load_hip_ops() is not None
 
 
+_OFFS_CACHE = {}
+_EMPTY_OFFS = None
+
+
+def _cached_offsets(n: int, device) -> torch.Tensor:
+    """[0, n] block table, cached — avoids a tiny H2D copy per
+    parameter per step in the list-form fused ops."""
+    key = (n, str(device))
+    t = _OFFS_CACHE.get(key)
+    if t is None:
+        t = torch.tensor([0, n], dtype=torch.int64, device=device)
+        _OFFS_CACHE[key] = t
+    return t
+
+
 def _gpu_ops(t: torch.Tensor):
     """Return the HIP op namespace for a GPU tensor; raise if unavailable."""
     if t.is_cuda:
@@ -83,7 +98,8 @@ def fused_sgd_update(params: Sequence[torch.Tensor],
             gl = (global_params[i].reshape(-1) if (global_params is not None
                                                   and mu != 0.0)
                   else torch.empty(0, dtype=w.dtype, device=w.device))
-            offs = torch.tensor([0, n], dtype=torch.int64, device=w.device)
+            offs = (_cached_offsets(n, w.device) if mu != 0.0
+                    else _cached_offsets(0, w.device)[:0])
             fused_sgd_update_flat(w.reshape(-1), g.reshape(-1), gl,
                                   w.shape[0], lr, mu, offs)
         return
@@ -107,8 +123,7 @@ def weighted_delta_accum(delta: Sequence[torch.Tensor],
         if wsum is None:
             wsum = float(weights.sum())
         for dl, cw, gw in zip(delta, client_params, global_params):
-            offs = torch.tensor([0, gw.numel()], dtype=torch.int64,
-                                device=cw.device)
+            offs = _cached_offsets(gw.numel(), cw.device)
             weighted_delta_accum_flat(dl.reshape(-1), cw.reshape(-1),
                                       gw.reshape(-1), weights, cw.shape[0],
                                       offs, wsum)
