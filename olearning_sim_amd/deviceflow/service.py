@@ -181,6 +181,19 @@ class DeviceFlowService:
                 payload=None) -> None:
         self.inbound.publish(Message(routing_key, compute_resource, payload))
 
+    def drain_inbound(self, timeout: float = 5.0) -> bool:
+        """Wait until the sorter has absorbed the inbound queue — a
+        producer calls this before NotifyComplete so its own messages
+        are not discarded by the completion flip (the reference relies
+        on Pulsar consume ordering for the same guarantee)."""
+        import time as _t
+        t0 = _t.time()
+        while self.inbound.qsize() > 0:
+            if _t.time() - t0 > timeout:
+                return False
+            _t.sleep(0.002)
+        return True
+
     def _should_put(self, msg: Message) -> bool:
         """Admission: flow exists, resource known, between start and
         complete (sorter.py:56-69)."""

@@ -50,6 +50,12 @@ class EngineJob:
     flow_total_timeout: float = 0.0
     flow_work_dir: str = ""
 
+    # global-model evaluation (the reference's post-train operators):
+    # every N rounds run the aggregated model on a held-out synthetic
+    # batch and report loss/accuracy (0 = off)
+    eval_every: int = 0
+    eval_batch: int = 64
+
     # bookkeeping / checkpointing
     checkpoint_dir: str = ""
     model_update_style: str = ""    # e.g. "{task_id}_{current_round}_result_model.safetensors"

@@ -40,7 +40,15 @@ class LogicalEngine:
     def __init__(self, job: EngineJob,
                  dist_ctx: Optional[pdist.DistContext] = None,
                  behavior: Optional[BehaviorFn] = None,
-                 result_sink: Optional[Callable[[Dict[str, Any]], None]] = None):
+                 result_sink: Optional[Callable[[Dict[str, Any]], None]] = None,
+                 deviceflow=None, perf=None):
+        # deviceflow: gradient-house service handle — the engine drives
+        # the reference's NotifyStart/NotifyComplete lifecycle per round
+        # (run_task.py:234-308) and publishes per-chunk summary messages
+        # to the inbound room for outbound consumers; perf: a
+        # PerformanceManager recording per-round metrics.
+        self.deviceflow = deviceflow
+        self.perf = perf
         self.job = job
         self.ctx = dist_ctx or pdist.DistContext(device=job.device)
         self.device = torch.device(self.ctx.device if dist_ctx else job.device)
@@ -117,10 +125,34 @@ class LogicalEngine:
         return ids.to(torch.int64)
 
     # ------------------------------------------------------------------
+    def evaluate_global(self, round_idx: int) -> Dict[str, float]:
+        """Run the aggregated global model on a held-out synthetic batch
+        (the reference's evaluate-style operator)."""
+        job = self.job
+        with torch.no_grad():
+            cast = {k: v.to(self.dtype).unsqueeze(0)
+                    for k, v in self.master.views.items()}
+            eval_ids = torch.zeros(1, dtype=torch.int64)
+            x, y = self.data.batch(eval_ids, round_idx, 10_000,
+                                   job.eval_batch, self.dtype)
+            logits = self.model.forward(cast, x)
+            k = logits.shape[-1]
+            flat = logits.reshape(-1, k).float()
+            labels = y.reshape(-1)
+            loss = float(torch.nn.functional.cross_entropy(flat, labels))
+            acc = float((flat.argmax(-1) == labels).float().mean())
+        return {"eval_loss": loss, "eval_acc": acc}
+
     def run_round(self, round_idx: int) -> Dict[str, Any]:
         job = self.job
+        t_round = time.time()
         ids = self.select_cohort(round_idx)
         cohort = int(ids.numel())
+        flow_id = None
+        if self.deviceflow is not None and job.behavior_strategy:
+            flow_id = self.deviceflow.notify_start(
+                job.task_id, "train", round_idx, "logical_simulation",
+                strategy=job.behavior_strategy)
 
         if self.behavior is not None:
             offline, dropped = self.behavior(round_idx, cohort)
@@ -160,6 +192,14 @@ class LogicalEngine:
             trained += min(stats["clients"], n_active - lo)
             if stats["loss"]:
                 losses.append(stats["loss"])
+            if flow_id is not None:
+                # per-chunk summary into the gradient house (the per-
+                # client tensors aggregate in-engine; the message plane
+                # carries chunk summaries for outbound consumers)
+                self.deviceflow.publish(flow_id, "logical_simulation",
+                                        payload={"round": round_idx,
+                                                 "clients": stats["clients"],
+                                                 "loss": stats["loss"]})
 
         local_weight = float(weights_all.sum())
         success_local = int(active_ids.numel())
@@ -187,6 +227,10 @@ class LogicalEngine:
         self.success_total += success
         self.failed_total += failed
 
+        if flow_id is not None:
+            self.deviceflow.drain_inbound()
+            self.deviceflow.notify_complete(job.task_id, "train", round_idx,
+                                            "logical_simulation")
         record = {
             "round": round_idx,
             "success": success,
@@ -195,6 +239,15 @@ class LogicalEngine:
             "loss": sum(losses) / len(losses) if losses else None,
             "round_failed": failed > job.dynamic_num,
         }
+        if job.eval_every > 0 and (round_idx + 1) % job.eval_every == 0:
+            record.update(self.evaluate_global(round_idx))
+        if self.perf is not None:
+            self.perf.record_round(job.task_id, round_idx,
+                                   time.time() - t_round, success,
+                                   loss=record["loss"])
+            if "eval_acc" in record:
+                self.perf.record(job.task_id, "eval_acc",
+                                 record["eval_acc"], round_idx)
         if job.save_every_round and job.checkpoint_dir and self.ctx.rank == 0:
             record["checkpoint"] = save_checkpoint(
                 job.checkpoint_dir, job.task_id, round_idx,

@@ -58,7 +58,7 @@ class SimulatorSession:
             runner = TaskRunner(
                 table, device=device,
                 checkpoint_dir=os.path.join(self.data_dir, "checkpoints"),
-                deviceflow=self.deviceflow)
+                deviceflow=self.deviceflow, perf=self.performance_mgr)
             self.task_mgr = TaskManager(
                 table=table, resource_mgr=self.resource_mgr, runner=runner,
                 deviceflow=self.deviceflow, auto_start=auto_start_threads)

@@ -124,11 +124,12 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
 
 class TaskRunner:
     def __init__(self, table: TaskTableRepo, device: str = "cpu",
-                 checkpoint_dir: str = "", deviceflow=None):
+                 checkpoint_dir: str = "", deviceflow=None, perf=None):
         self.table = table
         self.device = device
         self.checkpoint_dir = checkpoint_dir
         self.deviceflow = deviceflow   # deviceflow service facade (optional)
+        self.perf = perf               # PerformanceManager (optional)
         self.jobs: Dict[str, JobHandle] = {}
         self.log = Logger.shared()
         self._lock = threading.Lock()
@@ -204,7 +205,9 @@ class TaskRunner:
                         logical_operator=row["logical_operator"],
                         logical_result=json.dumps(row["logical_result"]))
 
-                eng = LogicalEngine(job, result_sink=sink)
+                eng = LogicalEngine(job, result_sink=sink,
+                                    deviceflow=self.deviceflow,
+                                    perf=self.perf)
                 # cooperative stop (reference: JobSubmissionClient.stop_job)
                 orig_run_round = eng.run_round
 

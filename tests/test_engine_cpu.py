@@ -80,3 +80,47 @@ def test_round_failed_when_over_tolerance():
     out = eng.run()
     assert out["rounds"] == 1  # early stop: failed > dynamic_num
     assert out["records"][0]["round_failed"]
+
+
+def test_engine_drives_deviceflow_lifecycle():
+    """use_gradient_house: the engine notifies start/complete per round
+    and chunk summaries flow through sorter -> dispatcher -> outbound."""
+    import json
+    import time as _time
+    from olearning_sim_amd.deviceflow.service import DeviceFlowService
+    strategy = json.dumps({"real_time_dispatch": {
+        "use_strategy": True, "dispatch_batch_sizes": [1]}})
+    svc = DeviceFlowService(time_scale=0.0, seed=1)
+    svc.register_task("t_df", ["logical_simulation"])
+    eng = LogicalEngine(_job(task_id="t_df", rounds=2, chunk_clients=5,
+                             behavior_strategy=strategy),
+                        deviceflow=svc)
+    eng.run()
+    t0 = _time.time()
+    while _time.time() - t0 < 10 and not svc.check_dispatch_finished("t_df"):
+        _time.sleep(0.02)
+    assert svc.check_dispatch_finished("t_df")
+    msgs = svc.outbound.drain()
+    # 2 rounds x 2 chunks of 5 clients
+    assert len(msgs) == 4
+    assert {m.routing_key for m in msgs} == {"t_df_train_0", "t_df_train_1"}
+    assert all(m.payload["clients"] == 5 for m in msgs)
+    svc.shutdown()
+
+
+def test_engine_eval_operator():
+    eng = LogicalEngine(_job(rounds=2, eval_every=2))
+    recs = eng.run()["records"]
+    assert "eval_acc" not in recs[0]
+    assert 0.0 <= recs[1]["eval_acc"] <= 1.0
+    assert recs[1]["eval_loss"] > 0
+
+
+def test_engine_records_perf_metrics():
+    from olearning_sim_amd.perf import PerformanceManager
+    pm = PerformanceManager()
+    eng = LogicalEngine(_job(task_id="t_perf", rounds=3), perf=pm)
+    eng.run()
+    summ = pm.summary("t_perf")
+    assert summ["metrics"]["round_time_s"]["count"] == 3
+    assert summ["metrics"]["clients_per_s"]["last"] > 0
