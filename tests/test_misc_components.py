@@ -231,3 +231,36 @@ def test_simulator_config_load(tmp_path):
     assert cfg.scheduler_sleep_time == 1.5
     assert cfg.phone_pool == {"u1": {"high": 4}}
     assert cfg.timers()["release_sleep_time"] == 10.0
+
+
+def test_operator_code_staging(tmp_path):
+    from olearning_sim_amd.task.staging import (stage_operator_code,
+                                                OperatorStagingError)
+    # builtin: no staging
+    assert stage_operator_code("builtin:fedavg", "train.py", "train",
+                               str(tmp_path / "w")) is None
+    # directory copy + entry validation
+    src = tmp_path / "opsrc"
+    src.mkdir()
+    (src / "train.py").write_text("print('hi')")
+    dst = stage_operator_code(str(src), "train.py", "train",
+                              str(tmp_path / "w1"))
+    assert os.path.exists(os.path.join(dst, "train.py"))
+    with pytest.raises(OperatorStagingError):
+        stage_operator_code(str(src), "missing.py", "train",
+                            str(tmp_path / "w2"))
+    # zip with a wrapping top-level dir gets flattened
+    import zipfile
+    z = tmp_path / "op.zip"
+    with zipfile.ZipFile(z, "w") as zf:
+        zf.writestr("mypkg/train.py", "print('hi')")
+    dst = stage_operator_code(str(z), "train.py", "train",
+                              str(tmp_path / "w3"))
+    assert os.path.exists(os.path.join(dst, "train.py"))
+    # file-repo key
+    from olearning_sim_amd.utils.file_repo import LocalFileRepo
+    repo = LocalFileRepo(root=str(tmp_path / "store"))
+    repo.upload_file(str(z), "operators", "remote/op.zip")
+    dst = stage_operator_code("remote/op.zip", "train.py", "train",
+                              str(tmp_path / "w4"), repo=repo)
+    assert os.path.exists(os.path.join(dst, "train.py"))
