@@ -461,6 +461,116 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v2(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// forward v5 — the pipelined variant for the common shapes
+// (IC*9 % 32 == 0, OW a power of two >= 16):
+//   - A fragments load DIRECTLY from global: each lane reads its own
+//     16-byte W row slice per K-step (L2-shared across the client''s
+//     n-tile workgroups) — no A staging, no A barrier;
+//   - B tile double-buffered in LDS with ONE barrier per K-step: the
+//     gather for step k+1 issues before the MFMAs of step k, and the
+//     commit of step k+1''s tile happens after them, so global latency
+//     hides under compute (PMC on v2: 44% of wave time parked);
+//   - the gather hoists the (b, oh) decomposition per 16-column row
+//     segment (always within one output row when OW >= 16).
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v5(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ y, ConvGeom g) {
+  __shared__ short bT_lds[2][CV2_BN * (CV2_BK + CV2_PAD)];
+  const int c = blockIdx.z;
+  const int m0 = blockIdx.y * CV2_BM;
+  const int n0 = blockIdx.x * CV2_BN;
+  const int K = g.IC * 9;
+  const int N = g.B * g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const __hip_bfloat16* xc = x + (int64_t)c * g.IC * g.B * HW;
+  const __hip_bfloat16* wc = w + (int64_t)c * g.OC * K;
+  __hip_bfloat16* yc = y + (int64_t)c * g.OC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int arow = m0 + wave * 16 + (lane & 15);     // this lane's W row
+  const __hip_bfloat16* wrow = wc + (int64_t)min(arow, g.OC - 1) * K;
+  const bool arow_ok = arow < g.OC;
+
+  f32x4 acc[CV2_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV2_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kk = threadIdx.x % CV2_BK;
+  const int nn0 = (threadIdx.x / CV2_BK) * (CV2_BN / 8);
+  short breg[CV2_BN / 8];
+
+  auto gather = [&](int k0) {
+    int k = k0 + kk;                       // k < K by construction
+    int ic = k / 9, r = k % 9;
+    int dh = r / 3, dw = r % 3;
+    const __hip_bfloat16* plane = xc + (int64_t)ic * g.B * HW;
+    // the 16-column segment lies inside ONE output row (OW >= 16 pow2)
+    int n = n0 + nn0;
+    int b = n >> g.lg_ohw;
+    int q = n & ((1 << g.lg_ohw) - 1);
+    int oh = q >> g.lg_ow;
+    int ow0 = q & ((1 << g.lg_ow) - 1);
+    int ih = oh * g.stride + dh - 1;
+    bool row_ok = (ih >= 0) && (ih < g.H) && (n < N);
+    const __hip_bfloat16* row =
+        plane + (int64_t)b * HW + (int64_t)max(0, min(ih, g.H - 1)) * g.W;
+#pragma unroll
+    for (int j = 0; j < CV2_BN / 8; ++j) {
+      int iw = (ow0 + j) * g.stride + dw - 1;
+      bool ok = row_ok && iw >= 0 && iw < g.W;
+      float v = to_f32(row[max(0, min(iw, g.W - 1))]);
+      breg[j] = ok ? bf16_bits(v) : (short)0;
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < CV2_BN / 8; ++j)
+      bT_lds[buf][(nn0 + j) * (CV2_BK + CV2_PAD) + kk] = breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += CV2_BK) {
+    __syncthreads();                       // buf[cur] ready for everyone
+    if (k0 + CV2_BK < K) gather(k0 + CV2_BK);   // loads fly over the MFMAs
+    // A fragment direct from global (16 B per lane; k0 8-aligned)
+    bf16x8 a;
+    {
+      uint4 av = *reinterpret_cast<const uint4*>(wrow + k0 + 8 * (lane >> 4));
+      a = *reinterpret_cast<const bf16x8*>(&av);
+      if (!arow_ok) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) a[e] = 0;
+      }
+    }
+#pragma unroll
+    for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur][(nt * 16 + (lane & 15)) * (CV2_BK + CV2_PAD)
+                       + 8 * (lane >> 4)]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    if (k0 + CV2_BK < K) commit(cur ^ 1);  // waits the gather's loads
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        yc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 extern "C" void ols_mfma_selftest(const void* A, const void* B, float* D,
                                   hipStream_t stream) {
@@ -485,9 +595,16 @@ extern "C" void ols_conv3x3_fwd(const void* x, const void* w, void* y, int C,
     }
   }
   dim3 grid(ceil_div(B * g.OH * g.OW, CV2_BN), ceil_div(OC, CV2_BM), C);
-  hipLaunchKernelGGL(k_conv3x3_fwd_v2, grid, dim3(CONV_THREADS), 0, stream,
-                     (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
-                     (__hip_bfloat16*)y, g);
+  const bool v5_ok = g.pow2 && g.OW >= 16 && ((IC * 9) % CV2_BK == 0)
+                     && ((IC * 9) % 8 == 0);
+  if (v5_ok)
+    hipLaunchKernelGGL(k_conv3x3_fwd_v5, grid, dim3(CONV_THREADS), 0, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
+                       (__hip_bfloat16*)y, g);
+  else
+    hipLaunchKernelGGL(k_conv3x3_fwd_v2, grid, dim3(CONV_THREADS), 0, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
+                       (__hip_bfloat16*)y, g);
 }
 
 extern "C" void ols_conv3x3_dgrad(const void* dy, const void* w, void* dx,

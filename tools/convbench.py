@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 """Kernel-level conv benchmark: custom MFMA vs MIOpen grouped."""
 import time, torch, torch.nn.functional as F
 from olearning_sim_amd.ops import load_hip_ops

@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time, torch
 from olearning_sim_amd.ops import load_hip_ops
 ops = load_hip_ops(required=True)
