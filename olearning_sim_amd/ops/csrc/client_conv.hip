@@ -474,6 +474,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v2(
 //     hides under compute (PMC on v2: 44% of wave time parked);
 //   - the gather hoists the (b, oh) decomposition per 16-column row
 //     segment (always within one output row when OW >= 16).
+template <bool ROW_HOIST>
 __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v5(
     const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
     __hip_bfloat16* __restrict__ y, ConvGeom g) {
@@ -507,22 +508,42 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v5(
     int ic = k / 9, r = k % 9;
     int dh = r / 3, dw = r % 3;
     const __hip_bfloat16* plane = xc + (int64_t)ic * g.B * HW;
-    // the 16-column segment lies inside ONE output row (OW >= 16 pow2)
-    int n = n0 + nn0;
-    int b = n >> g.lg_ohw;
-    int q = n & ((1 << g.lg_ohw) - 1);
-    int oh = q >> g.lg_ow;
-    int ow0 = q & ((1 << g.lg_ow) - 1);
-    int ih = oh * g.stride + dh - 1;
-    bool row_ok = (ih >= 0) && (ih < g.H) && (n < N);
-    const __hip_bfloat16* row =
-        plane + (int64_t)b * HW + (int64_t)max(0, min(ih, g.H - 1)) * g.W;
+    if (ROW_HOIST) {
+      // the whole segment lies inside ONE output row: hoist (b, oh)
+      int n = n0 + nn0;
+      int b = n >> g.lg_ohw;
+      int q = n & ((1 << g.lg_ohw) - 1);
+      int oh = q >> g.lg_ow;
+      int ow0 = q & ((1 << g.lg_ow) - 1);
+      int ih = oh * g.stride + dh - 1;
+      bool row_ok = (ih >= 0) && (ih < g.H) && (n < N);
+      const __hip_bfloat16* row =
+          plane + (int64_t)b * HW + (int64_t)max(0, min(ih, g.H - 1)) * g.W;
 #pragma unroll
-    for (int j = 0; j < CV2_BN / 8; ++j) {
-      int iw = (ow0 + j) * g.stride + dw - 1;
-      bool ok = row_ok && iw >= 0 && iw < g.W;
-      float v = to_f32(row[max(0, min(iw, g.W - 1))]);
-      breg[j] = ok ? bf16_bits(v) : (short)0;
+      for (int j = 0; j < CV2_BN / 8; ++j) {
+        int iw = (ow0 + j) * g.stride + dw - 1;
+        bool ok = row_ok && iw >= 0 && iw < g.W;
+        float v = to_f32(row[max(0, min(iw, g.W - 1))]);
+        breg[j] = ok ? bf16_bits(v) : (short)0;
+      }
+    } else {
+      // small planes (OW 4/8): per-element shift decomposition under
+      // the same pipeline
+#pragma unroll
+      for (int j = 0; j < CV2_BN / 8; ++j) {
+        int n = n0 + nn0 + j;
+        int nc = min(n, N - 1);
+        int b = nc >> g.lg_ohw;
+        int q = nc & ((1 << g.lg_ohw) - 1);
+        int oh = q >> g.lg_ow;
+        int ow = q & ((1 << g.lg_ow) - 1);
+        int ih = oh * g.stride + dh - 1, iw = ow * g.stride + dw - 1;
+        bool ok = n < N && ih >= 0 && ih < g.H && iw >= 0 && iw < g.W;
+        float v = to_f32(plane[(int64_t)b * HW
+                               + (int64_t)max(0, min(ih, g.H - 1)) * g.W
+                               + max(0, min(iw, g.W - 1))]);
+        breg[j] = ok ? bf16_bits(v) : (short)0;
+      }
     }
   };
   auto commit = [&](int buf) {
@@ -595,12 +616,15 @@ extern "C" void ols_conv3x3_fwd(const void* x, const void* w, void* y, int C,
     }
   }
   dim3 grid(ceil_div(B * g.OH * g.OW, CV2_BN), ceil_div(OC, CV2_BM), C);
-  const bool v5_ok = g.pow2 && g.OW >= 16 && ((IC * 9) % CV2_BK == 0)
-                     && ((IC * 9) % 8 == 0);
-  if (v5_ok)
-    hipLaunchKernelGGL(k_conv3x3_fwd_v5, grid, dim3(CONV_THREADS), 0, stream,
-                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
-                       (__hip_bfloat16*)y, g);
+  const bool v5_ok = g.pow2 && ((IC * 9) % CV2_BK == 0);
+  if (v5_ok && g.OW >= CV2_BN / 8)
+    hipLaunchKernelGGL((k_conv3x3_fwd_v5<true>), grid, dim3(CONV_THREADS), 0,
+                       stream, (const __hip_bfloat16*)x,
+                       (const __hip_bfloat16*)w, (__hip_bfloat16*)y, g);
+  else if (v5_ok)
+    hipLaunchKernelGGL((k_conv3x3_fwd_v5<false>), grid, dim3(CONV_THREADS), 0,
+                       stream, (const __hip_bfloat16*)x,
+                       (const __hip_bfloat16*)w, (__hip_bfloat16*)y, g);
   else
     hipLaunchKernelGGL(k_conv3x3_fwd_v2, grid, dim3(CONV_THREADS), 0, stream,
                        (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
