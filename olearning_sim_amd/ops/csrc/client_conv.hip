@@ -592,6 +592,245 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v5(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// dgrad v5: same pipeline as fwd v5 — A (flipped W) double-buffered in
+// LDS, B (dy gather) register-double-buffered, ONE barrier per K-step.
+template <int STRIDE>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v5(
+    const __hip_bfloat16* __restrict__ dy, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ dx, ConvGeom g) {
+  __shared__ short a_lds[2][CV2_BM * CV2_BK];
+  __shared__ short bT_lds[2][CV2_BN * (CV2_BK + CV2_PAD)];
+  const int c = blockIdx.z;
+  const int m0 = blockIdx.y * CV2_BM;            // over IC
+  const int n0 = blockIdx.x * CV2_BN;            // over B*H*W
+  const int K = g.OC * 9;
+  const int N = g.B * g.H * g.W;
+  const int OHW = g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const int lg_w = g.lg_ow + (STRIDE == 2 ? 1 : 0);   // log2(W) = log2(OW*s)
+  const int lg_hw = lg_w * 2;                          // H == W planes
+  const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * g.B * OHW;
+  const __hip_bfloat16* wc = w + (int64_t)c * g.OC * g.IC * 9;
+  __hip_bfloat16* dxc = dx + (int64_t)c * g.IC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  f32x4 acc[CV2_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV2_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kk = threadIdx.x % CV2_BK;
+  const int nn0 = (threadIdx.x / CV2_BK) * (CV2_BN / 8);
+  const int amm = threadIdx.x / 4;               // A staging row (ic)
+  const int ak0 = (threadIdx.x % 4) * 8;
+  short breg[CV2_BN / 8];
+  short areg[8];
+
+  auto gather = [&](int k0) {
+    {
+      int ic = m0 + amm;
+      int icc = min(ic, g.IC - 1);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = k0 + ak0 + j;                    // < K by construction
+        int oc = k / 9, r2 = k % 9;
+        int dh = r2 / 3, dw2 = r2 % 3;
+        float v = to_f32(wc[((int64_t)oc * g.IC + icc) * 9
+                            + (2 - dh) * 3 + (2 - dw2)]);
+        areg[j] = (ic < g.IC) ? bf16_bits(v) : (short)0;
+      }
+    }
+    {
+      int k = k0 + kk;
+      int oc = k / 9, r2 = k % 9;
+      int dh = r2 / 3, dw2 = r2 % 3;
+      const __hip_bfloat16* plane = dyc + (int64_t)oc * g.B * OHW;
+      // the segment lies inside ONE input row when W >= seg width
+      int n = n0 + nn0;
+      int b = n >> lg_hw;
+      int q = n & ((1 << lg_hw) - 1);
+      int h = q >> lg_w;
+      int w0 = q & ((1 << lg_w) - 1);
+      int num_h = h + dh - 1;
+      bool row_ok = (n < N) && num_h >= 0 && num_h % STRIDE == 0
+                    && (num_h / STRIDE) < g.OH;
+      const __hip_bfloat16* row = plane
+          + (int64_t)b * OHW
+          + (int64_t)max(0, min(num_h / STRIDE, g.OH - 1)) * g.OW;
+      if (g.W >= CV2_BN / 8) {
+#pragma unroll
+        for (int j = 0; j < CV2_BN / 8; ++j) {
+          int num_w = w0 + j + dw2 - 1;
+          bool ok = row_ok && num_w >= 0 && num_w % STRIDE == 0
+                    && (num_w / STRIDE) < g.OW;
+          float v = to_f32(row[max(0, min(num_w / STRIDE, g.OW - 1))]);
+          breg[j] = ok ? bf16_bits(v) : (short)0;
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < CV2_BN / 8; ++j) {
+          int nj = min(n0 + nn0 + j, N - 1);
+          int bj = nj >> lg_hw;
+          int qj = nj & ((1 << lg_hw) - 1);
+          int hj = qj >> lg_w;
+          int wj = qj & ((1 << lg_w) - 1);
+          int nh = hj + dh - 1, nw = wj + dw2 - 1;
+          bool ok = (n0 + nn0 + j) < N && nh >= 0 && nw >= 0
+                    && nh % STRIDE == 0 && nw % STRIDE == 0
+                    && (nh / STRIDE) < g.OH && (nw / STRIDE) < g.OW;
+          float v = to_f32(plane[(int64_t)bj * OHW
+              + (int64_t)max(0, min(nh / STRIDE, g.OH - 1)) * g.OW
+              + max(0, min(nw / STRIDE, g.OW - 1))]);
+          breg[j] = ok ? bf16_bits(v) : (short)0;
+        }
+      }
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      a_lds[buf][amm * CV2_BK + ak0 + j] = areg[j];
+#pragma unroll
+    for (int j = 0; j < CV2_BN / 8; ++j)
+      bT_lds[buf][(nn0 + j) * (CV2_BK + CV2_PAD) + kk] = breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += CV2_BK) {
+    __syncthreads();
+    if (k0 + CV2_BK < K) gather(k0 + CV2_BK);
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(
+        &a_lds[cur][(wave * 16 + (lane & 15)) * CV2_BK + 8 * (lane >> 4)]);
+#pragma unroll
+    for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur][(nt * 16 + (lane & 15)) * (CV2_BK + CV2_PAD)
+                       + 8 * (lane >> 4)]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    if (k0 + CV2_BK < K) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.IC)
+        dxc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad v5: A = dY rows DIRECT from global (contiguous, like fwd's W);
+// B = patch tile register-double-buffered; one barrier per q-step.
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v5(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
+    float* __restrict__ dw, ConvGeom g) {
+  __shared__ short bT_lds[2][CV2_BN * (CV2_BK + CV2_PAD)];
+  const int c = blockIdx.z;
+  const int m0 = blockIdx.y * CV2_BM;          // over OC
+  const int n0 = blockIdx.x * CV2_BN;          // over IC*9
+  const int K9 = g.IC * 9;
+  const int NN = g.B * g.OH * g.OW;            // reduction dim (pow2 x B)
+  const int OHW = g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const __hip_bfloat16* xc = x + (int64_t)c * g.IC * g.B * HW;
+  const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
+  float* dwc = dw + (int64_t)c * g.OC * K9;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int arow = m0 + wave * 16 + (lane & 15);   // dY row (oc)
+  const __hip_bfloat16* dyrow = dyc + (int64_t)min(arow, g.OC - 1) * NN;
+  const bool arow_ok = arow < g.OC;
+
+  f32x4 acc[CV2_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV2_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int qq = threadIdx.x % CV2_BK;
+  const int nn0 = (threadIdx.x / CV2_BK) * (CV2_BN / 8);
+  short breg[CV2_BN / 8];
+
+  auto gather = [&](int q0) {
+    int q = q0 + qq;
+    int qc = min(q, NN - 1);
+    int b = qc >> g.lg_ohw;
+    int p = qc & ((1 << g.lg_ohw) - 1);
+    int oh = p >> g.lg_ow;
+    int ow = p & ((1 << g.lg_ow) - 1);
+    bool q_ok = q < NN;
+#pragma unroll
+    for (int j = 0; j < CV2_BN / 8; ++j) {
+      int k = n0 + nn0 + j;
+      float v = 0.f;
+      bool ok = q_ok && k < K9;
+      int kc = min(k, K9 - 1);
+      int ic = kc / 9, r2 = kc % 9;
+      int dh = r2 / 3, dw2 = r2 % 3;
+      int ih = oh * g.stride + dh - 1, iw = ow * g.stride + dw2 - 1;
+      ok = ok && ih >= 0 && ih < g.H && iw >= 0 && iw < g.W;
+      v = to_f32(xc[((int64_t)ic * g.B + b) * HW
+                    + (int64_t)max(0, min(ih, g.H - 1)) * g.W
+                    + max(0, min(iw, g.W - 1))]);
+      breg[j] = ok ? bf16_bits(v) : (short)0;
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < CV2_BN / 8; ++j)
+      bT_lds[buf][(nn0 + j) * (CV2_BK + CV2_PAD) + qq] = breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int q0 = 0; q0 < NN; q0 += CV2_BK) {
+    __syncthreads();
+    if (q0 + CV2_BK < NN) gather(q0 + CV2_BK);
+    bf16x8 a;
+    {
+      // NN is B * pow2 — multiple of 32, so q0 + 8*(l>>4) is 8-aligned
+      uint4 av = *reinterpret_cast<const uint4*>(dyrow + q0 + 8 * (lane >> 4));
+      a = *reinterpret_cast<const bf16x8*>(&av);
+      if (!arow_ok) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) a[e] = 0;
+      }
+    }
+#pragma unroll
+    for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur][(nt * 16 + (lane & 15)) * (CV2_BK + CV2_PAD)
+                       + 8 * (lane >> 4)]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    if (q0 + CV2_BK < NN) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV2_BN / 16; ++nt) {
+    int k = n0 + nt * 16 + (lane & 15);
+    if (k >= K9) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        dwc[(int64_t)m * K9 + k] = acc[nt][r];
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 extern "C" void ols_mfma_selftest(const void* A, const void* B, float* D,
                                   hipStream_t stream) {
@@ -645,10 +884,24 @@ extern "C" void ols_conv3x3_dgrad(const void* dy, const void* w, void* dx,
       g.pow2 = 1;
     }
   }
-  dim3 grid(ceil_div(B * H * W, CONV_BN), ceil_div(IC, CONV_BM), C);
-  hipLaunchKernelGGL(k_conv3x3_dgrad, grid, dim3(CONV_THREADS), 0, stream,
-                     (const __hip_bfloat16*)dy, (const __hip_bfloat16*)w,
-                     (__hip_bfloat16*)dx, g);
+  const bool v5_ok = g.pow2 && ((OC * 9) % CV2_BK == 0)
+                     && (stride == 1 || stride == 2) && H == W;
+  if (v5_ok) {
+    dim3 grid5(ceil_div(B * H * W, CV2_BN), ceil_div(IC, CV2_BM), C);
+    if (stride == 1)
+      hipLaunchKernelGGL((k_conv3x3_dgrad_v5<1>), grid5, dim3(CONV_THREADS),
+                         0, stream, (const __hip_bfloat16*)dy,
+                         (const __hip_bfloat16*)w, (__hip_bfloat16*)dx, g);
+    else
+      hipLaunchKernelGGL((k_conv3x3_dgrad_v5<2>), grid5, dim3(CONV_THREADS),
+                         0, stream, (const __hip_bfloat16*)dy,
+                         (const __hip_bfloat16*)w, (__hip_bfloat16*)dx, g);
+  } else {
+    dim3 grid(ceil_div(B * H * W, CONV_BN), ceil_div(IC, CONV_BM), C);
+    hipLaunchKernelGGL(k_conv3x3_dgrad, grid, dim3(CONV_THREADS), 0, stream,
+                       (const __hip_bfloat16*)dy, (const __hip_bfloat16*)w,
+                       (__hip_bfloat16*)dx, g);
+  }
 }
 
 extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
@@ -665,7 +918,17 @@ extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
       g.pow2 = 1;
     }
   }
-  dim3 grid(ceil_div(IC * 9, CONV_BN), ceil_div(OC, CONV_BM), C);
-  hipLaunchKernelGGL(k_conv3x3_wgrad, grid, dim3(CONV_THREADS), 0, stream,
-                     (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy, dw, g);
+  const int NN = B * g.OH * g.OW;
+  const bool v5_ok = g.pow2 && (NN % CV2_BK == 0);
+  if (v5_ok) {
+    dim3 grid5(ceil_div(IC * 9, CV2_BN), ceil_div(OC, CV2_BM), C);
+    hipLaunchKernelGGL(k_conv3x3_wgrad_v5, grid5, dim3(CONV_THREADS), 0,
+                       stream, (const __hip_bfloat16*)x,
+                       (const __hip_bfloat16*)dy, dw, g);
+  } else {
+    dim3 grid(ceil_div(IC * 9, CONV_BN), ceil_div(OC, CONV_BM), C);
+    hipLaunchKernelGGL(k_conv3x3_wgrad, grid, dim3(CONV_THREADS), 0, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
+                       dw, g);
+  }
 }
