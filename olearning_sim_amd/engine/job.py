@@ -9,7 +9,7 @@ tests) or from a TaskConfig by task/runner.py.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, Optional
+from typing import Any, Dict, List, Optional
 
 
 @dataclass
@@ -42,6 +42,11 @@ class EngineJob:
 
     # behaviour simulation (deviceflow): arrival/offline/drop shaping
     behavior_strategy: str = ""     # gradient-house strategy JSON ("" = none)
+
+    # ordered operator list executed each round (reference
+    # operatorflow.operators, run_task.py:228): (name, kind) with kind
+    # in {"train", "evaluate", "checkpoint"}
+    operators: List[Any] = field(default_factory=lambda: [("train", "train")])
 
     # operator-flow round gates (reference flow_setting.start/stop)
     flow_start_strategy: str = ""

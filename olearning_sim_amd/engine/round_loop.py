@@ -97,6 +97,7 @@ class LogicalEngine:
         # totals across rounds (reference logical_result accounting)
         self.success_total = 0
         self.failed_total = 0
+        self.last_operator = "train"
 
     # ------------------------------------------------------------------
     def _chunk_size(self, cohort: int) -> int:
@@ -143,15 +144,16 @@ class LogicalEngine:
             acc = float((flat.argmax(-1) == labels).float().mean())
         return {"eval_loss": loss, "eval_acc": acc}
 
-    def run_round(self, round_idx: int) -> Dict[str, Any]:
+    def _op_train(self, round_idx: int,
+                  op_name: str = "train") -> Dict[str, Any]:
+        """The training operator: one cohort pass + aggregation."""
         job = self.job
-        t_round = time.time()
         ids = self.select_cohort(round_idx)
         cohort = int(ids.numel())
         flow_id = None
         if self.deviceflow is not None and job.behavior_strategy:
             flow_id = self.deviceflow.notify_start(
-                job.task_id, "train", round_idx, "logical_simulation",
+                job.task_id, op_name, round_idx, "logical_simulation",
                 strategy=job.behavior_strategy)
 
         if self.behavior is not None:
@@ -229,26 +231,49 @@ class LogicalEngine:
 
         if flow_id is not None:
             self.deviceflow.drain_inbound()
-            self.deviceflow.notify_complete(job.task_id, "train", round_idx,
+            self.deviceflow.notify_complete(job.task_id, op_name, round_idx,
                                             "logical_simulation")
-        record = {
-            "round": round_idx,
+        return {
             "success": success,
             "failed": failed,
             "trained": trained,
             "loss": sum(losses) / len(losses) if losses else None,
             "round_failed": failed > job.dynamic_num,
         }
-        if job.eval_every > 0 and (round_idx + 1) % job.eval_every == 0:
+
+    def run_round(self, round_idx: int) -> Dict[str, Any]:
+        """Execute the round's ordered operator list
+        (reference run_task.py:228-311: per operator -> deviceflow
+        NotifyStart, work, NotifyComplete, result accumulation)."""
+        job = self.job
+        t_round = time.time()
+        record: Dict[str, Any] = {"round": round_idx, "success": 0,
+                                  "failed": 0, "trained": 0, "loss": None,
+                                  "round_failed": False}
+        ops = job.operators or [("train", "train")]
+        for entry in ops:
+            name, kind = (entry if isinstance(entry, (tuple, list))
+                          else (entry, "train"))
+            if kind == "train":
+                record.update(self._op_train(round_idx, name))
+            elif kind == "evaluate":
+                record.update(self.evaluate_global(round_idx))
+            elif kind == "checkpoint":
+                if self.ctx.rank == 0 and job.checkpoint_dir:
+                    record["checkpoint"] = save_checkpoint(
+                        job.checkpoint_dir, job.task_id, round_idx,
+                        self.master.state_dict(), job.model_update_style)
+            self.last_operator = name
+        if job.eval_every > 0 and (round_idx + 1) % job.eval_every == 0                 and "eval_acc" not in record:
             record.update(self.evaluate_global(round_idx))
         if self.perf is not None:
             self.perf.record_round(job.task_id, round_idx,
-                                   time.time() - t_round, success,
+                                   time.time() - t_round, record["success"],
                                    loss=record["loss"])
             if "eval_acc" in record:
                 self.perf.record(job.task_id, "eval_acc",
                                  record["eval_acc"], round_idx)
-        if job.save_every_round and job.checkpoint_dir and self.ctx.rank == 0:
+        if job.save_every_round and job.checkpoint_dir                 and self.ctx.rank == 0 and "checkpoint" not in record:
             record["checkpoint"] = save_checkpoint(
                 job.checkpoint_dir, job.task_id, round_idx,
                 self.master.state_dict(), job.model_update_style)
@@ -295,7 +320,7 @@ class LogicalEngine:
         return {
             "task_id": job.task_id,
             "logical_round": rec["round"] + 1,
-            "logical_operator": "train",
+            "logical_operator": self.last_operator,
             "logical_result": {
                 "logical_result": [{
                     "name": job.data_name,

@@ -79,9 +79,37 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
             train_op.operation_behavior_controller.use_gradient_house:
         behavior = train_op.operation_behavior_controller.strategy_gradient_house
 
+    # ordered operator list for the round loop: kind from each
+    # operator's params JSON ("kind") or inferred from its name
+    op_entries = []
+    for op in task.operatorflow.operators:
+        if not (op.logical_simulation.operator_code_path
+                or op.logical_simulation.operator_entry_file):
+            continue            # device-simulation-only operator
+    # (second pass keeps params parsing in one place)
+    for op in task.operatorflow.operators:
+        if not (op.logical_simulation.operator_code_path
+                or op.logical_simulation.operator_entry_file):
+            continue
+        try:
+            op_params = json.loads(op.logical_simulation.operator_params or "{}")
+        except Exception:
+            op_params = {}
+        kind = op_params.get("kind")
+        if kind is None:
+            nm = op.name.lower()
+            if "eval" in nm:
+                kind = "evaluate"
+            elif "checkpoint" in nm or "save" in nm:
+                kind = "checkpoint"
+            else:
+                kind = "train"
+        op_entries.append((op.name, kind))
+
     fs = task.operatorflow.flow_setting
     first_data = task.target.data[0] if task.target.data else None
     job = EngineJob(
+        operators=op_entries or [("train", "train")],
         flow_start_strategy=fs.start.logical_simulation.strategy,
         flow_stop_strategy=fs.stop.logical_simulation.strategy,
         flow_wait_interval=max(1, fs.stop.logical_simulation.wait_interval

@@ -184,3 +184,30 @@ def test_queue_recovery_from_table():
     mgr2 = TaskManager(table=table, resource_mgr=res,
                        runner=TaskRunner(table))
     assert mgr2.get_task_queue() == ["t_e2e"]
+
+
+def test_multi_operator_task_lifecycle():
+    """train -> evaluate operator list: fusion requires the LAST
+    operator (reference calculate_conditions last-operator check)."""
+    mgr = make_manager()
+    tj = json.loads(task_json(task_id="t_multi", rounds=2))
+    ops = tj["operatorflow"]["operators"]
+    ops.append({
+        "name": "evaluate",
+        "operation_behavior_controller": {"use_gradient_house": False,
+                                          "strategy_gradient_house": "",
+                                          "outbound_service": ""},
+        "input": ["train"], "use_data": False,
+        "model": {"use_model": False},
+        "logical_simulation": {
+            "operator_transfer_type": "FILE",
+            "operator_code_path": "builtin:evaluate",
+            "operator_entry_file": "eval.py",
+            "operator_params": json.dumps({"kind": "evaluate"})},
+        "device_simulation": {}})
+    ok, msg = mgr.submit_task(json.dumps(tj))
+    assert ok, msg
+    assert mgr.step_schedule() == "t_multi"
+    st = wait_terminal(mgr, "t_multi")
+    assert st == TaskStatus.SUCCEEDED
+    assert mgr.table.get_item_value("t_multi", "logical_operator") == "evaluate"
