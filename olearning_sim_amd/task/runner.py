@@ -85,11 +85,6 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
     for op in task.operatorflow.operators:
         if not (op.logical_simulation.operator_code_path
                 or op.logical_simulation.operator_entry_file):
-            continue            # device-simulation-only operator
-    # (second pass keeps params parsing in one place)
-    for op in task.operatorflow.operators:
-        if not (op.logical_simulation.operator_code_path
-                or op.logical_simulation.operator_entry_file):
             continue
         try:
             op_params = json.loads(op.logical_simulation.operator_params or "{}")
