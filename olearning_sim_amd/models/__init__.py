@@ -19,9 +19,16 @@ def build_model(name: str, **kwargs: Any) -> ClientBatchedModel:
     if name == "bert-base":
         from .bert import BertBase
         return BertBase(**kwargs)
+    if ":" in name:
+        # user model plug-in "pkg.module:ClassName" (the analogue of the
+        # reference's user-supplied operator code)
+        import importlib
+        mod_name, cls_name = name.split(":", 1)
+        cls = getattr(importlib.import_module(mod_name), cls_name)
+        return cls(**kwargs)
     if name not in _REGISTRY:
         raise KeyError(f"unknown model {name!r}; known: {sorted(_REGISTRY)} "
-                       f"+ ['bert', 'bert-base']")
+                       f"+ ['bert', 'bert-base'] or 'pkg.module:ClassName'")
     return _REGISTRY[name](**kwargs)
 
 

@@ -213,12 +213,17 @@ class TaskManager:
         for task_id in running:
             status = self.get_task_status(task_id)
             if status.is_terminal():
-                self.scheduler.release(task_id)
                 if self.deviceflow is not None:
                     try:
+                        # reference releaseResource thread: resources are
+                        # held until the gradient house has drained
+                        # (utils.py check_deviceflow_dispatch_finished)
+                        if not self.deviceflow.check_dispatch_finished(task_id):
+                            continue
                         self.deviceflow.unregister_task(task_id)
                     except Exception:
                         pass
+                self.scheduler.release(task_id)
                 self.table.set_items(task_id,
                                      resource_occupied=0,
                                      finish_task_time=time.time(),

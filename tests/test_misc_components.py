@@ -264,3 +264,50 @@ def test_operator_code_staging(tmp_path):
     dst = stage_operator_code("remote/op.zip", "train.py", "train",
                               str(tmp_path / "w4"), repo=repo)
     assert os.path.exists(os.path.join(dst, "train.py"))
+
+
+def test_custom_model_plugin(tmp_path, monkeypatch):
+    import sys
+    (tmp_path / "user_models.py").write_text(
+        "from olearning_sim_amd.models.mlp import MLP\n"
+        "class TinyNet(MLP):\n"
+        "    def __init__(self):\n"
+        "        super().__init__(in_features=8, hidden=4, num_classes=2)\n")
+    monkeypatch.syspath_prepend(str(tmp_path))
+    from olearning_sim_amd.models import build_model
+    m = build_model("user_models:TinyNet")
+    assert m.num_classes == 2
+    gp = m.init_global()
+    assert gp["fc1.w"].shape == (8, 4)
+
+
+def test_release_waits_for_deviceflow_drain():
+    import time as _time
+    from olearning_sim_amd.deviceflow.service import DeviceFlowService
+    from test_manager import make_manager, task_json, wait_terminal
+    from olearning_sim_amd.resource.manager import ResourceManager
+    from olearning_sim_amd.task.manager import TaskManager
+    from olearning_sim_amd.task.runner import TaskRunner
+    from olearning_sim_amd.task.table import TaskTableRepo
+    svc = DeviceFlowService(time_scale=0.0, seed=0)
+    table = TaskTableRepo(":memory:")
+    res = ResourceManager(":memory:", totals={"cpu": 8, "mem": 64,
+                                              "gpu": 0, "hbm_gb": 0})
+    runner = TaskRunner(table, deviceflow=svc)
+    mgr = TaskManager(table=table, resource_mgr=res, runner=runner,
+                      deviceflow=svc)
+    tj = task_json(task_id="t_drain", use_gradient_house=True)
+    ok, msg = mgr.submit_task(tj)
+    assert ok, msg
+    assert mgr.step_schedule() == "t_drain"
+    st = wait_terminal(mgr, "t_drain")
+    from olearning_sim_amd.task.status import TaskStatus
+    assert st == TaskStatus.SUCCEEDED
+    t0 = _time.time()
+    while _time.time() - t0 < 15:
+        if "t_drain" in mgr.step_release():
+            break
+        _time.sleep(0.05)
+    assert not mgr.resources.holding("t_drain")
+    assert not svc.registry.is_registered("t_drain")
+    svc.shutdown()
