@@ -22,7 +22,7 @@ from __future__ import annotations
 
 import json
 import math
-from typing import Dict, Optional, Tuple
+from typing import Dict, Tuple
 
 import torch
 

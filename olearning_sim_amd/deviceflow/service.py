@@ -29,7 +29,6 @@ from ..utils.logging import Logger
 from .dispatcher import Dispatcher
 from .registry import TaskOrientedDeviceFlowRegistry
 from .rooms import InboundRoom, Message, OutboundRoom, ShelfRoom
-from .strategy import Strategy
 
 
 class FlowState:

@@ -11,7 +11,7 @@ instead of dozens of small tensors.
 
 from __future__ import annotations
 
-from typing import Dict, Iterator, List, Tuple
+from typing import Dict, Iterator, List
 
 import torch
 

@@ -15,7 +15,7 @@ transport is gone.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

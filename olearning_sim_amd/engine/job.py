@@ -9,7 +9,7 @@ tests) or from a TaskConfig by task/runner.py.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 
 @dataclass

@@ -26,7 +26,7 @@ from __future__ import annotations
 
 import os
 import time
-from typing import Any, Callable, Dict, Optional
+from typing import Callable, Dict, Optional
 
 from ..utils.logging import Logger
 

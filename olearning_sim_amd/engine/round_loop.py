@@ -17,7 +17,6 @@ matching deviceflow semantics where drops happen after NotifyComplete).
 
 from __future__ import annotations
 
-import json
 import time
 from typing import Any, Callable, Dict, List, Optional
 
@@ -27,7 +26,7 @@ from ..models import build_model
 from ..ops import fused
 from ..parallel import dist as pdist
 from .checkpoint import save_checkpoint
-from .client_manager import FlatParams, chunk_ids
+from .client_manager import FlatParams
 from .data import SyntheticFederatedData
 from .job import EngineJob
 from .local_train import LocalTrainer

@@ -10,7 +10,6 @@ head predicts the next token (causal mask), CE via the fused HIP kernel.
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

@@ -13,10 +13,9 @@ from __future__ import annotations
 from typing import Optional
 
 import torch
-import torch.nn.functional as F
 
 from .base import (ClientBatchedModel, Params, binit, kaiming,
-                   blinear, bconv2d, bgroupnorm)
+                   blinear, bconv2d)
 
 _STAGES = (64, 128, 256, 512)
 _GN_GROUPS = 8

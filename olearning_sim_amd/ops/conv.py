@@ -11,8 +11,6 @@ semantics, used by tests and the CPU plumbing config).
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn.functional as F
 

@@ -19,7 +19,7 @@ run here.  A CUDA tensor with no extension raises (no silent fallback).
 from __future__ import annotations
 
 import os
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import torch
 

@@ -15,7 +15,7 @@ import os
 import random
 import shutil
 import zipfile
-from typing import List, Optional, Tuple
+from typing import Tuple
 
 
 class HybridDataSplitter:

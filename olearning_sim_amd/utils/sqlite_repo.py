@@ -16,7 +16,7 @@ import json
 import os
 import sqlite3
 import threading
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Any, Dict, List, Optional
 
 
 class SqlTableRepo:
