@@ -122,3 +122,33 @@ def test_performance_api(client, session):
     rows = client.get("/performancemgr/metrics/t",
                       params={"metric": "loss"}).json()["metrics"]
     assert [r["value"] for r in rows] == [2.0, 1.5]
+
+
+@pytest.mark.timeout(420)
+def test_cluster_manager_launches_distributed_worker(tmp_path):
+    """NodeClusterManager runs the engine worker group (2 CPU ranks over
+    gloo) — the control-plane path to multi-GPU execution."""
+    import json as _json
+    from olearning_sim_amd.cluster import NodeClusterManager, WorkerGroupSpec
+    job = {"task_id": "grp_job", "model_name": "mlp",
+           "model_kwargs": {"in_features": 16, "hidden": 8, "num_classes": 4},
+           "clients": 8, "rounds": 2, "local_steps": 1, "batch_size": 2,
+           "lr": 0.1, "num_classes": 4, "shard_size": 8, "seed": 5,
+           "dtype": "float32"}
+    job_file = tmp_path / "job.json"
+    job_file.write_text(_json.dumps(job))
+    result = tmp_path / "result.json"
+    cm = NodeClusterManager()
+    assert cm.create_cluster(WorkerGroupSpec(
+        name="eng", replicas=2,
+        entry_module="olearning_sim_amd.engine.worker",
+        args=["--job-json", str(job_file), "--result-json", str(result)]))
+    assert cm.wait_until_running("eng", timeout=60)
+    rec = cm._clusters["eng"]
+    rec.proc.wait(timeout=300)
+    assert rec.status() == "succeeded"
+    out = _json.loads(result.read_text())
+    assert out["rounds"] == 2
+    assert out["world_size"] == 2
+    assert out["success_total"] == 2 * 8   # all shards, both rounds
+    cm.delete_cluster("eng")
