@@ -36,6 +36,27 @@ def save_checkpoint(directory: str, task_id: str, current_round: int,
     return path
 
 
+def latest_round(directory: str, task_id: str,
+                 model_update_style: str = "") -> int:
+    """Highest round with a saved artifact, or -1 (used for crash
+    resume: the reference's actors fetch round r-1's model when
+    round > 0, utils_run_task.py:327-397)."""
+    import re
+    style = model_update_style or DEFAULT_STYLE
+    best = -1
+    if not os.path.isdir(directory):
+        return best
+    pat = re.escape(style).replace(
+        re.escape("{task_id}"), re.escape(task_id)).replace(
+        re.escape("{current_round}"), r"(\d+)")
+    rx = re.compile("^" + pat + "$")
+    for f in os.listdir(directory):
+        m = rx.match(f)
+        if m:
+            best = max(best, int(m.group(1)))
+    return best
+
+
 def load_checkpoint(directory: str, task_id: str, current_round: int,
                     model_update_style: str = "",
                     device: str = "cpu") -> Optional[Dict[str, torch.Tensor]]:

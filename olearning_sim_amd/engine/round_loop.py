@@ -25,7 +25,7 @@ import torch
 from ..models import build_model
 from ..ops import fused
 from ..parallel import dist as pdist
-from .checkpoint import save_checkpoint
+from .checkpoint import save_checkpoint, load_checkpoint, latest_round
 from .client_manager import FlatParams
 from .data import SyntheticFederatedData
 from .job import EngineJob
@@ -97,6 +97,20 @@ class LogicalEngine:
         self.success_total = 0
         self.failed_total = 0
         self.last_operator = "train"
+
+        # crash resume: load the newest per-round artifact and continue
+        # from the next round (reference model_update_style download)
+        self.start_round = 0
+        if job.checkpoint_dir and job.save_every_round:
+            r = latest_round(job.checkpoint_dir, job.task_id,
+                             job.model_update_style)
+            if r >= 0:
+                sd = load_checkpoint(job.checkpoint_dir, job.task_id, r,
+                                     job.model_update_style,
+                                     device=str(self.device))
+                if sd is not None:
+                    self.master.load_state_dict(sd)
+                    self.start_round = r + 1
 
     # ------------------------------------------------------------------
     def _chunk_size(self, cohort: int) -> int:
@@ -283,7 +297,7 @@ class LogicalEngine:
         job = self.job
         records: List[Dict[str, Any]] = []
         t0 = time.time()
-        for r in range(job.rounds):
+        for r in range(self.start_round, job.rounds):
             if self.stop_requested:
                 break
             self.flow.start(r)         # operator-flow start gate

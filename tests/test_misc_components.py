@@ -311,3 +311,46 @@ def test_release_waits_for_deviceflow_drain():
     assert not mgr.resources.holding("t_drain")
     assert not svc.registry.is_registered("t_drain")
     svc.shutdown()
+
+
+def test_engine_resumes_from_latest_checkpoint(tmp_path):
+    from olearning_sim_amd.engine import EngineJob, LogicalEngine
+
+    def job(rounds):
+        return EngineJob(task_id="rz", model_name="mlp",
+                         model_kwargs={"in_features": 16, "hidden": 8,
+                                       "num_classes": 4},
+                         clients=4, rounds=rounds, local_steps=1,
+                         batch_size=2, lr=0.1, num_classes=4, seed=0,
+                         checkpoint_dir=str(tmp_path),
+                         save_every_round=True)
+
+    e1 = LogicalEngine(job(2))
+    e1.run()
+    after2 = e1.master.flat.clone()
+    # a "crashed and restarted" engine resumes at round 2, not 0
+    e2 = LogicalEngine(job(4))
+    assert e2.start_round == 2
+    torch.testing.assert_close(e2.master.flat, after2)
+    out = e2.run()
+    assert out["rounds"] == 2                 # only rounds 2 and 3 ran
+    assert os.path.exists(tmp_path / "rz_3_result_model.safetensors")
+
+
+def test_drop_all_leaves_master_unchanged():
+    import json as _json
+    from olearning_sim_amd.engine import EngineJob, LogicalEngine
+    strategy = _json.dumps({"real_time_dispatch": {
+        "use_strategy": True,
+        "drop_simulation": {"drop_probability": 1.0}}})
+    job = EngineJob(task_id="dz", model_name="mlp",
+                    model_kwargs={"in_features": 16, "hidden": 8,
+                                  "num_classes": 4},
+                    clients=4, rounds=1, local_steps=1, batch_size=2,
+                    lr=0.5, num_classes=4, seed=0,
+                    behavior_strategy=strategy, dynamic_num=100)
+    eng = LogicalEngine(job)
+    before = eng.master.flat.clone()
+    rec = eng.run_round(0)
+    assert rec["success"] == 4            # clients trained (drops are
+    torch.testing.assert_close(eng.master.flat, before)  # post-train)
