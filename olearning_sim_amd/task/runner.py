@@ -154,6 +154,7 @@ class TaskRunner:
         self.deviceflow = deviceflow   # deviceflow service facade (optional)
         self.perf = perf               # PerformanceManager (optional)
         self.jobs: Dict[str, JobHandle] = {}
+        self.task_jobs: Dict[str, List[str]] = {}   # task_id -> job ids
         self.log = Logger.shared()
         self._lock = threading.Lock()
 
@@ -256,6 +257,7 @@ class TaskRunner:
         handle.thread = threading.Thread(target=run, daemon=True)
         with self._lock:
             self.jobs[job_id] = handle
+            self.task_jobs.setdefault(task.task_id, []).append(job_id)
         handle.thread.start()
         return job_id
 
@@ -311,6 +313,7 @@ class TaskRunner:
         handle.thread = threading.Thread(target=run, daemon=True)
         with self._lock:
             self.jobs[job_id] = handle
+            self.task_jobs.setdefault(task.task_id, []).append(job_id)
         handle.thread.start()
         return job_id
 
@@ -327,12 +330,16 @@ class TaskRunner:
         return True
 
     def stop_task(self, task_id: str) -> None:
+        """Stop every job (logical + simulated device side) this task
+        launched — and ONLY this task's (reference stopTask stops the
+        Ray job by id and the phone task by task_id)."""
+        with self._lock:
+            ids = list(self.task_jobs.get(task_id, []))
         job_id = self.table.get_item_value(task_id, "job_id")
-        if job_id:
-            self.stop_job(job_id)
-        for h in self.jobs.values():
-            if h.kind == "device":
-                h.request_stop()
+        if job_id and job_id not in ids:
+            ids.append(job_id)
+        for jid in ids:
+            self.stop_job(jid)
 
     def wait(self, job_id: str, timeout: float = 60.0) -> Optional[JobStatus]:
         h = self.jobs.get(job_id)
