@@ -211,3 +211,23 @@ def test_multi_operator_task_lifecycle():
     st = wait_terminal(mgr, "t_multi")
     assert st == TaskStatus.SUCCEEDED
     assert mgr.table.get_item_value("t_multi", "logical_operator") == "evaluate"
+
+
+def test_stop_running_task_reports_stopped():
+    mgr = make_manager()
+    # long-ish task: 60 rounds of a small model
+    tj = json.loads(task_json(task_id="t_stop", rounds=60, clients=6))
+    mgr.submit_task(json.dumps(tj))
+    assert mgr.step_schedule() == "t_stop"
+    # let it get going, then stop
+    t0 = time.time()
+    while time.time() - t0 < 10:
+        if (mgr.table.get_item_value("t_stop", "logical_round") or 0) >= 1:
+            break
+        time.sleep(0.02)
+    ok, _ = mgr.stop_task("t_stop")
+    assert ok
+    st = wait_terminal(mgr, "t_stop", timeout=30)
+    assert st == TaskStatus.STOPPED
+    assert "t_stop" in mgr.step_release()
+    assert not mgr.resources.holding("t_stop")
