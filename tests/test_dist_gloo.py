@@ -41,12 +41,7 @@ def _worker(rank, world, port, conn):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
-def test_two_rank_gloo_round_aggregates():
-    import socket
-    with socket.socket() as s:      # pick a free rendezvous port
-        s.bind(("127.0.0.1", 0))
-        port = s.getsockname()[1]
+def _spawn_round(port):
     ctx = mp.get_context("spawn")
     pipes, procs = [], []
     for rank in range(2):
@@ -55,10 +50,34 @@ def test_two_rank_gloo_round_aggregates():
         p.start()
         pipes.append(parent)
         procs.append(p)
-    results = [pipe.recv() for pipe in pipes]
+    results = []
+    ok = True
+    for pipe, p in zip(pipes, procs):
+        if pipe.poll(60):
+            results.append(pipe.recv())
+        else:
+            ok = False
     for p in procs:
         p.join(30)
-        assert p.exitcode == 0
+        if p.exitcode != 0:
+            ok = False
+    return ok, results
+
+
+@pytest.mark.timeout(240)
+def test_two_rank_gloo_round_aggregates():
+    import socket
+    # the TCP-store rendezvous can race with port reuse under load:
+    # retry with a fresh port (the production path is torchrun, which
+    # owns the rendezvous; see test_bench_contract.py)
+    for attempt in range(3):
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        ok, results = _spawn_round(port)
+        if ok:
+            break
+    assert ok, "gloo rendezvous failed 3 times"
     r0, r1 = sorted(results, key=lambda r: r["rank"])
     # both ranks hold the same aggregated global model
     torch.testing.assert_close(r0["master"], r1["master"])
