@@ -32,12 +32,15 @@ def _worker(rank, world, port, conn):
                     shard_size=8, seed=77)  # same model seed on all ranks
     eng = LogicalEngine(job, dist_ctx=ctx)
     out = eng.run()
+    # plain bytes, NOT a tensor: torch tensors go through the Pipe by
+    # shared-memory FD passing, which races with child exit
     conn.send({
         "rank": rank,
-        "master": eng.master.flat.clone(),
+        "master_bytes": eng.master.flat.numpy().tobytes(),
         "success_total": out["success_total"],
         "shard": (lo, hi),
     })
+    conn.recv()                     # wait for the parent's ack
     dist.destroy_process_group()
 
 
@@ -55,6 +58,7 @@ def _spawn_round(port):
     for pipe, p in zip(pipes, procs):
         if pipe.poll(60):
             results.append(pipe.recv())
+            pipe.send("ack")
         else:
             ok = False
     for p in procs:
@@ -80,8 +84,10 @@ def test_two_rank_gloo_round_aggregates():
     assert ok, "gloo rendezvous failed 3 times"
     r0, r1 = sorted(results, key=lambda r: r["rank"])
     # both ranks hold the same aggregated global model
-    torch.testing.assert_close(r0["master"], r1["master"])
+    m0 = torch.frombuffer(bytearray(r0["master_bytes"]), dtype=torch.float32)
+    m1 = torch.frombuffer(bytearray(r1["master_bytes"]), dtype=torch.float32)
+    torch.testing.assert_close(m0, m1)
     # success counts were all-reduced: both report the global total
     assert r0["success_total"] == r1["success_total"] == 2 * 8
     assert r0["shard"] == (0, 4) and r1["shard"] == (4, 8)
-    assert r0["master"].isfinite().all()
+    assert m0.isfinite().all()
