@@ -511,7 +511,9 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v5(
     if (ROW_HOIST) {
       // the whole segment lies inside ONE output row: hoist (b, oh)
       int n = n0 + nn0;
-      int b = n >> g.lg_ohw;
+      // clamp b: the unconditional clamped-address load must stay in
+      // bounds even for fully-masked (n >= N) tail segments
+      int b = min(n >> g.lg_ohw, g.B - 1);
       int q = n & ((1 << g.lg_ohw) - 1);
       int oh = q >> g.lg_ow;
       int ow0 = q & ((1 << g.lg_ow) - 1);
@@ -649,7 +651,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v5(
       const __hip_bfloat16* plane = dyc + (int64_t)oc * g.B * OHW;
       // the segment lies inside ONE input row when W >= seg width
       int n = n0 + nn0;
-      int b = n >> lg_hw;
+      int b = min(n >> lg_hw, g.B - 1);   // masked tails stay in bounds
       int q = n & ((1 << lg_hw) - 1);
       int h = q >> lg_w;
       int w0 = q & ((1 << lg_w) - 1);
@@ -672,7 +674,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v5(
 #pragma unroll
         for (int j = 0; j < CV2_BN / 8; ++j) {
           int nj = min(n0 + nn0 + j, N - 1);
-          int bj = nj >> lg_hw;
+          int bj = min(nj >> lg_hw, g.B - 1);
           int qj = nj & ((1 << lg_hw) - 1);
           int hj = qj >> lg_w;
           int wj = qj & ((1 << lg_w) - 1);
