@@ -354,3 +354,28 @@ def test_drop_all_leaves_master_unchanged():
     rec = eng.run_round(0)
     assert rec["success"] == 4            # clients trained (drops are
     torch.testing.assert_close(eng.master.flat, before)  # post-train)
+
+
+@pytest.mark.timeout(180)
+def test_cli_submit_wait_and_status(tmp_path):
+    import subprocess, sys
+    from test_manager import task_json
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    tf = tmp_path / "task.json"
+    tf.write_text(task_json(task_id="t_cli"))
+    out = subprocess.run(
+        [sys.executable, "-m", "olearning_sim_amd",
+         "--data-dir", str(tmp_path / "data"),
+         "submit", str(tf), "--wait"],
+        cwd=repo, capture_output=True, text=True, timeout=150)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [json.loads(l) for l in out.stdout.strip().splitlines()]
+    assert lines[0]["is_success"]
+    assert lines[1]["task_status"] == "SUCCEEDED"
+    # status over the SAME data dir sees the terminal state
+    out2 = subprocess.run(
+        [sys.executable, "-m", "olearning_sim_amd",
+         "--data-dir", str(tmp_path / "data"), "status", "t_cli"],
+        cwd=repo, capture_output=True, text=True, timeout=60)
+    assert json.loads(out2.stdout.strip().splitlines()[-1])[
+        "task_status"] == "SUCCEEDED"
