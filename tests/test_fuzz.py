@@ -1,0 +1,90 @@
+"""Property-based robustness tests (hypothesis): the public parsers and
+validators must never crash on arbitrary input — they return False /
+empty schedules instead (the reference's swallow-and-log policy)."""
+
+import json
+
+from hypothesis import given, settings, strategies as st
+
+from olearning_sim_amd.deviceflow.strategy import Strategy
+from olearning_sim_amd.deviceflow.validate import ValidateStrategy
+from olearning_sim_amd.task.schema import json2taskconfig, taskconfig2json
+from olearning_sim_amd.task.validate import ValidateParameters
+
+json_scalars = st.one_of(st.none(), st.booleans(),
+                         st.integers(-10**6, 10**6),
+                         st.floats(allow_nan=False, allow_infinity=False,
+                                   width=32),
+                         st.text(max_size=20))
+json_values = st.recursive(
+    json_scalars,
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(st.text(max_size=12), children, max_size=4)),
+    max_leaves=12)
+json_objects = st.dictionaries(st.text(max_size=16), json_values, max_size=6)
+
+
+@settings(max_examples=150, deadline=None)
+@given(json_objects)
+def test_task_parsing_and_validation_never_crash(raw):
+    """json2taskconfig on structurally arbitrary dicts either parses or
+    raises cleanly; validation returns a bool, never raises."""
+    text = json.dumps(raw)
+    try:
+        cfg = json2taskconfig(text)
+    except (TypeError, AttributeError, ValueError, KeyError):
+        return   # malformed section types rejected at parse: acceptable
+    v = ValidateParameters()
+    ok = v.validate_task_parameters(raw, cfg)
+    assert isinstance(ok, bool)
+    if ok:
+        # anything accepted must round-trip
+        json.loads(taskconfig2json(cfg))
+
+
+@settings(max_examples=150, deadline=None)
+@given(json_objects)
+def test_strategy_validation_never_crashes(raw):
+    v = ValidateStrategy()
+    assert isinstance(v.check(json.dumps(raw)), bool)
+
+
+@settings(max_examples=100, deadline=None)
+@given(json_objects)
+def test_flow_analysis_never_crashes_on_garbage(raw):
+    out = Strategy.flow_strategy_analysis(json.dumps(raw), "t_op_0")
+    assert isinstance(out, tuple) and len(out) == 3
+
+
+@settings(max_examples=60, deadline=None)
+@given(total=st.integers(1, 5000),
+       length=st.integers(1, 30),
+       d0=st.floats(0, 5, allow_nan=False),
+       dlen=st.floats(0.1, 10, allow_nan=False))
+def test_interval_schedule_preserves_total(total, length, d0, dlen):
+    """Whatever the interval/domain geometry, a valid positive-rate
+    schedule apportions exactly total_dispatch_amount."""
+    s = json.dumps({"flow_dispatch": {
+        "use_strategy": True, "total_dispatch_amount": total,
+        "specific_interval": {
+            "use": True, "intervals": [[0, length]],
+            "dispatch_rules": {"domains": [[d0, d0 + dlen]],
+                               "functions": ["t*0 + 1"]}}}})
+    timing, amounts, drops = Strategy.flow_strategy_analysis(s, "t_op_0")
+    assert sum(amounts) == total
+    assert len(amounts) == length
+    assert all(a >= 0 for a in amounts)
+    assert len(drops) == length
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.integers(0, 50), min_size=1, max_size=20),
+       st.floats(0, 1, allow_nan=False))
+def test_drop_list_bounds(amounts, p):
+    drops = Strategy.generate_drop_list(
+        amounts, {"drop_probability": [p] * len(amounts)})
+    assert len(drops) == len(amounts)
+    for d, a in zip(drops, amounts):
+        assert all(0 <= i < max(1, a) for i in d)
+        assert len(d) <= a
