@@ -82,6 +82,34 @@ def build_app(session) -> FastAPI:
         def get_task_queue():
             return {"tasks": tm.get_task_queue()}
 
+        @app.get("/taskmgr/listTasks")
+        def list_tasks():
+            rows = tm.table.all_rows()
+            return {"tasks": [{
+                "task_id": r["task_id"], "user_id": r.get("user_id"),
+                "task_status": r.get("task_status"),
+                "logical_round": r.get("logical_round"),
+                "in_queue_time": r.get("in_queue_time"),
+                "finish_task_time": r.get("finish_task_time"),
+            } for r in rows]}
+
+        @app.get("/taskmgr/getTaskResult/{task_id}")
+        def get_task_result(task_id: str):
+            import json as _json
+            row = tm.table.get_row(task_id)
+            if row is None:
+                return {"error": "task not found"}
+            return {
+                "task_id": task_id,
+                "task_status": tm.get_task_status(task_id).value,
+                "logical_round": row.get("logical_round"),
+                "logical_operator": row.get("logical_operator"),
+                "logical_result": _json.loads(row["logical_result"])
+                if row.get("logical_result") else None,
+                "device_result": _json.loads(row["device_result"])
+                if row.get("device_result") else None,
+            }
+
     # -- ResourceMgr service (resourceService.proto) ---------------------
     if session.resource_mgr is not None:
         rm = session.resource_mgr

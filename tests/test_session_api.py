@@ -152,3 +152,22 @@ def test_cluster_manager_launches_distributed_worker(tmp_path):
     assert out["world_size"] == 2
     assert out["success_total"] == 2 * 8   # all shards, both rounds
     cm.delete_cluster("eng")
+
+
+def test_list_tasks_and_result_api(client, session):
+    body = {"task": json.loads(task_json(task_id="t_res"))}
+    assert client.post("/taskmgr/submitTask", json=body).json()["is_success"]
+    assert session.task_mgr.step_schedule() == "t_res"
+    t0 = time.time()
+    while time.time() - t0 < 30:
+        st = client.get("/taskmgr/getTaskStatus/t_res").json()["task_status"]
+        if st in ("SUCCEEDED", "FAILED", "STOPPED"):
+            break
+        time.sleep(0.05)
+    tasks = client.get("/taskmgr/listTasks").json()["tasks"]
+    assert any(t["task_id"] == "t_res" for t in tasks)
+    res = client.get("/taskmgr/getTaskResult/t_res").json()
+    assert res["task_status"] == "SUCCEEDED"
+    tgt = res["logical_result"]["logical_result"][0]["simulation_target"]
+    assert tgt["success_num"] == [6]
+    assert client.get("/taskmgr/getTaskResult/nope").json()["error"]
