@@ -302,8 +302,10 @@ class TaskManager:
             result["device_result"] = (json.loads(dr)["device_result"]
                                        if dr else [])
             finished = False
-            for h in self.runner.jobs.values():
-                if h.kind == "device":
+            # only THIS task's simulated phone job (runner.task_jobs)
+            for jid in self.runner.task_jobs.get(task_id, []):
+                h = self.runner.jobs.get(jid)
+                if h is not None and h.kind == "device":
                     finished = h.status in (JobStatus.SUCCEEDED,
                                             JobStatus.FAILED,
                                             JobStatus.STOPPED)

@@ -231,3 +231,35 @@ def test_stop_running_task_reports_stopped():
     assert st == TaskStatus.STOPPED
     assert "t_stop" in mgr.step_release()
     assert not mgr.resources.holding("t_stop")
+
+
+def hybrid_task_json(task_id, logical=4, device=2, phones=3):
+    tj = json.loads(task_json(task_id=task_id, clients=logical + device,
+                              dynamic=1))
+    d = tj["target"]["data"][0]
+    d["allocation"]["logical_simulation"] = [logical]
+    d["allocation"]["device_simulation"] = [device]
+    tj["device_simulation"]["resource_request"] = [
+        {"name": "data_0", "devices": ["high"], "num_request": [phones]}]
+    return json.dumps(tj)
+
+
+def test_hybrid_task_with_simulated_phone_farm():
+    """Hybrid allocation: logical engine + simulated phone farm both run
+    and the fused status combines their per-tier successes."""
+    from olearning_sim_amd.resource.manager import ResourceManager
+    table = TaskTableRepo(":memory:")
+    res = ResourceManager(":memory:", totals={"cpu": 8, "mem": 64,
+                                              "gpu": 0, "hbm_gb": 0},
+                          phone_pool={"u1": {"high": 8}})
+    mgr = TaskManager(table=table, resource_mgr=res,
+                      runner=TaskRunner(table))
+    ok, msg = mgr.submit_task(hybrid_task_json("t_hybrid"))
+    assert ok, msg
+    assert mgr.step_schedule() == "t_hybrid"
+    st = wait_terminal(mgr, "t_hybrid", timeout=60)
+    assert st == TaskStatus.SUCCEEDED
+    dr = json.loads(table.get_item_value("t_hybrid", "device_result"))
+    assert dr["device_result"][0]["simulation_target"]["success_num"] == [2]
+    lr = json.loads(table.get_item_value("t_hybrid", "logical_result"))
+    assert lr["logical_result"][0]["simulation_target"]["success_num"] == [4]
