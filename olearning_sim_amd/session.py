@@ -28,10 +28,14 @@ from .utils.logging import Logger
 
 class SimulatorSession:
     def __init__(self, svc: int = 0, data_dir: Optional[str] = None,
-                 device: str = "", auto_start_threads: bool = True):
+                 device: str = "", auto_start_threads: bool = True,
+                 config=None):
+        from .config import SimulatorConfig
+        self.config = config or SimulatorConfig.load(None)
         self.svc = svc
-        self.data_dir = data_dir or os.path.join(
-            os.path.expanduser("~"), ".olearning_sim_amd")
+        self.data_dir = (data_dir or self.config.data_dir
+                         or os.path.join(os.path.expanduser("~"),
+                                         ".olearning_sim_amd"))
         os.makedirs(self.data_dir, exist_ok=True)
         self.log = Logger.shared()
 
@@ -39,15 +43,21 @@ class SimulatorSession:
             return os.path.join(self.data_dir, name)
 
         if not device:
+            device = self.config.device
+        if not device:
             try:
                 import torch
                 device = "cuda:0" if torch.cuda.is_available() else "cpu"
             except Exception:
                 device = "cpu"
 
-        self.resource_mgr = (ResourceManager(db("resmgr.sqlite"))
+        self.resource_mgr = (ResourceManager(
+                                 db("resmgr.sqlite"),
+                                 phone_pool=self.config.phone_pool)
                              if svc in (0, 1, 2) else None)
-        self.deviceflow = (DeviceFlowService(db("deviceflow.sqlite"))
+        self.deviceflow = (DeviceFlowService(
+                               db("deviceflow.sqlite"),
+                               time_scale=self.config.deviceflow_time_scale)
                            if svc in (0, 3) else None)
         self.performance_mgr = (PerformanceManager(db("perf.sqlite"))
                                 if svc in (0, 4) else None)
@@ -61,7 +71,8 @@ class SimulatorSession:
                 deviceflow=self.deviceflow, perf=self.performance_mgr)
             self.task_mgr = TaskManager(
                 table=table, resource_mgr=self.resource_mgr, runner=runner,
-                deviceflow=self.deviceflow, auto_start=auto_start_threads)
+                deviceflow=self.deviceflow, timers=self.config.timers(),
+                auto_start=auto_start_threads)
 
     def serve(self, host: str = "127.0.0.1", port: int = 60061,
               block: bool = True):

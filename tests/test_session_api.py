@@ -171,3 +171,17 @@ def test_list_tasks_and_result_api(client, session):
     tgt = res["logical_result"]["logical_result"][0]["simulation_target"]
     assert tgt["success_num"] == [6]
     assert client.get("/taskmgr/getTaskResult/nope").json()["error"]
+
+
+def test_session_honours_config(tmp_path):
+    from olearning_sim_amd.config import SimulatorConfig
+    cfg = SimulatorConfig(scheduler_sleep_time=0.7,
+                          phone_pool={"u9": {"high": 2}},
+                          deviceflow_time_scale=0.0)
+    s = SimulatorSession(svc=0, data_dir=str(tmp_path), device="cpu",
+                         auto_start_threads=False, config=cfg)
+    assert s.task_mgr.timers["scheduler_sleep_time"] == 0.7
+    avail = s.resource_mgr.get_resource("u9")["device_simulation"]["u9"]
+    assert avail == {"high": 2}
+    assert s.deviceflow.time_scale == 0.0
+    s.shutdown()

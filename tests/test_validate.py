@@ -97,3 +97,25 @@ def test_type_rejections():
     assert not ok
     ok, err = check(mutate(("operatorflow", "operators", 0, "use_data"), "yes"))
     assert not ok
+
+
+def test_non_minio_data_path_must_be_zip():
+    raw = mutate(("target", "data", 0, "data_transfer_type"), "S3")
+    raw["target"]["data"][0]["data_path"] = "bucket/plaindir"
+    ok, err = check(raw)
+    assert not ok and "zip" in err.lower()
+    raw["target"]["data"][0]["data_path"] = "bucket/data.zip"
+    ok, err = check(raw)
+    assert ok, err
+
+
+def test_dir_code_path_requires_file_transfer(tmp_path):
+    d = tmp_path / "opdir"
+    d.mkdir()
+    raw = mutate(("operatorflow", "operators", 0, "logical_simulation"),
+                 {"operator_transfer_type": "S3",
+                  "operator_code_path": str(d),
+                  "operator_entry_file": "train.py",
+                  "operator_params": "{}"})
+    ok, err = check(raw)
+    assert not ok and "FILE" in err
