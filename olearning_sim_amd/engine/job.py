@@ -65,6 +65,10 @@ class EngineJob:
     checkpoint_dir: str = ""
     model_update_style: str = ""    # e.g. "{task_id}_{current_round}_result_model.safetensors"
     save_every_round: bool = False
+    # resume from the newest saved round artifact (crash recovery of an
+    # interrupted run).  OFF by default: a fresh submission of the same
+    # task id must start from round 0 like the reference's run_task.
+    resume: bool = False
 
     # failure-tolerance accounting (reference total_simulation semantics)
     data_name: str = "data_0"

@@ -101,7 +101,7 @@ class LogicalEngine:
         # crash resume: load the newest per-round artifact and continue
         # from the next round (reference model_update_style download)
         self.start_round = 0
-        if job.checkpoint_dir and job.save_every_round:
+        if job.resume and job.checkpoint_dir and job.save_every_round:
             r = latest_round(job.checkpoint_dir, job.task_id,
                              job.model_update_style)
             if r >= 0:

@@ -328,8 +328,13 @@ def test_engine_resumes_from_latest_checkpoint(tmp_path):
     e1 = LogicalEngine(job(2))
     e1.run()
     after2 = e1.master.flat.clone()
-    # a "crashed and restarted" engine resumes at round 2, not 0
-    e2 = LogicalEngine(job(4))
+    # a fresh engine does NOT resume by default (fresh submissions of a
+    # task id start at round 0, reference semantics) ...
+    assert LogicalEngine(job(4)).start_round == 0
+    # ... an explicit crash-recovery restart does
+    j = job(4)
+    j.resume = True
+    e2 = LogicalEngine(j)
     assert e2.start_round == 2
     torch.testing.assert_close(e2.master.flat, after2)
     out = e2.run()
