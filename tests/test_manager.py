@@ -263,3 +263,26 @@ def test_hybrid_task_with_simulated_phone_farm():
     assert dr["device_result"][0]["simulation_target"]["success_num"] == [2]
     lr = json.loads(table.get_item_value("t_hybrid", "logical_result"))
     assert lr["logical_result"][0]["simulation_target"]["success_num"] == [4]
+
+
+def test_three_tasks_contend_for_resources_and_all_finish():
+    """Only 2 tasks fit at once (4 cpus / 2 each): the scheduler runs
+    them as resources free up; all three succeed."""
+    mgr = make_manager(cpu=4.0)
+    for tid in ("a1", "a2", "a3"):
+        ok, msg = mgr.submit_task(task_json(task_id=tid, rounds=1))
+        assert ok, msg
+    done = set()
+    t0 = time.time()
+    while len(done) < 3 and time.time() - t0 < 60:
+        mgr.step_schedule()
+        for tid in ("a1", "a2", "a3"):
+            st = mgr.get_task_status(tid)
+            if st.is_terminal():
+                done.add(tid)
+        mgr.step_release()
+        time.sleep(0.05)
+    assert done == {"a1", "a2", "a3"}
+    for tid in done:
+        assert mgr.get_task_status(tid) == TaskStatus.SUCCEEDED
+        assert not mgr.resources.holding(tid)
