@@ -185,3 +185,41 @@ def test_session_honours_config(tmp_path):
     assert avail == {"high": 2}
     assert s.deviceflow.time_scale == 0.0
     s.shutdown()
+
+
+@pytest.mark.timeout(120)
+def test_real_http_server_round_trip(tmp_path):
+    """serve() over a real socket (uvicorn), not just the TestClient."""
+    import socket
+    import urllib.request
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        port = sk.getsockname()[1]
+    s = SimulatorSession(svc=0, data_dir=str(tmp_path), device="cpu",
+                         auto_start_threads=False)
+    server = s.serve(host="127.0.0.1", port=port, block=False)
+    try:
+        t0 = time.time()
+        while time.time() - t0 < 30:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/health", timeout=5) as r:
+                    body = json.loads(r.read())
+                break
+            except Exception:
+                time.sleep(0.2)
+        assert body["ok"]
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{port}/taskmgr/submitTask",
+            data=json.dumps({"task": json.loads(
+                task_json(task_id="t_http"))}).encode(),
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=10) as r:
+            assert json.loads(r.read())["is_success"]
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/taskmgr/getTaskQueue",
+                timeout=5) as r:
+            assert json.loads(r.read())["tasks"] == ["t_http"]
+    finally:
+        server.should_exit = True
+        s.shutdown()
