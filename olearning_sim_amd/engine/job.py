@@ -73,7 +73,12 @@ class EngineJob:
     # failure-tolerance accounting (reference total_simulation semantics)
     data_name: str = "data_0"
     device_tier: str = "high"
-    dynamic_num: int = 0            # tolerated failures per round
+    dynamic_num: int = 0            # tolerated failures per round (total)
+    # per-tier populations for this rank: [(tier, clients), ...] —
+    # client ids are assigned to tiers in order (prefix ranges); empty
+    # means a single job.device_tier tier of all clients
+    tier_counts: List[Any] = field(default_factory=list)
+    dynamic_nums: List[int] = field(default_factory=list)  # per tier
 
     def resolved_cohort(self) -> int:
         return self.cohort_size if self.cohort_size > 0 else self.clients

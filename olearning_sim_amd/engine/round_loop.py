@@ -97,6 +97,15 @@ class LogicalEngine:
         self.success_total = 0
         self.failed_total = 0
         self.last_operator = "train"
+        # per-tier client-id prefix ranges (reference device tiers)
+        tiers = job.tier_counts or [(job.device_tier, job.clients)]
+        self.tier_names = [t for t, _ in tiers]
+        bounds = [0]
+        for _, n in tiers:
+            bounds.append(bounds[-1] + int(n))
+        self.tier_bounds = bounds          # len = T+1; covers [0, clients)
+        self.tier_dynamic = (list(job.dynamic_nums)
+                             if job.dynamic_nums else [job.dynamic_num])
 
         # crash resume: load the newest per-round artifact and continue
         # from the next round (reference model_update_style download)
@@ -217,24 +226,35 @@ class LogicalEngine:
                                                  "loss": stats["loss"]})
 
         local_weight = float(weights_all.sum())
-        success_local = int(active_ids.numel())
-        failed_local = int(offline.sum())
+        # per-tier success/failed (client ids map to tiers by prefix range)
+        T = len(self.tier_names)
+        succ_t = [0] * T
+        fail_t = [0] * T
+        for t in range(T):
+            lo, hi = self.tier_bounds[t], self.tier_bounds[t + 1]
+            in_tier = (ids >= lo) & (ids < hi)
+            fail_t[t] = int((in_tier & offline).sum())
+            succ_t[t] = int((in_tier & ~offline).sum())
+        success_local = sum(succ_t)
+        failed_local = sum(fail_t)
 
         # cross-GPU aggregation: one RCCL all-reduce of the flat delta +
-        # one small all-reduce carrying (total_weight, success, failed)
+        # one small all-reduce carrying (weight, per-tier succ/fail)
         if self.ctx.enabled:
-            self._scalar[0] = local_weight
-            self._scalar[1] = success_local
-            self._scalar[2] = failed_local
+            stats = torch.tensor([local_weight] + succ_t + fail_t,
+                                 dtype=torch.float64, device=self.device)
             work = pdist.all_reduce_flat(self._delta, async_op=True)
-            pdist.all_reduce_flat(self._scalar)
+            pdist.all_reduce_flat(stats)
             if work is not None:
                 work.wait()
-            total_weight = float(self._scalar[0])
-            success = int(self._scalar[1])
-            failed = int(self._scalar[2])
+            total_weight = float(stats[0])
+            succ_t = [int(x) for x in stats[1:1 + T]]
+            fail_t = [int(x) for x in stats[1 + T:1 + 2 * T]]
+            success = sum(succ_t)
+            failed = sum(fail_t)
         else:
-            total_weight, success, failed = local_weight, success_local, failed_local
+            total_weight = local_weight
+            success, failed = success_local, failed_local
 
         if total_weight > 0:
             fused.apply_aggregate([self.master.flat], [self._delta], total_weight)
@@ -246,12 +266,19 @@ class LogicalEngine:
             self.deviceflow.drain_inbound()
             self.deviceflow.notify_complete(job.task_id, op_name, round_idx,
                                             "logical_simulation")
+        dyn = self.tier_dynamic
+        if len(dyn) != T:
+            dyn = [job.dynamic_num] * T
+        round_failed = (failed > job.dynamic_num if T == 1
+                        else any(f > d for f, d in zip(fail_t, dyn)))
         return {
             "success": success,
             "failed": failed,
+            "success_per_tier": succ_t,
+            "failed_per_tier": fail_t,
             "trained": trained,
             "loss": sum(losses) / len(losses) if losses else None,
-            "round_failed": failed > job.dynamic_num,
+            "round_failed": round_failed,
         }
 
     def run_round(self, round_idx: int) -> Dict[str, Any]:
@@ -262,7 +289,9 @@ class LogicalEngine:
         t_round = time.time()
         record: Dict[str, Any] = {"round": round_idx, "success": 0,
                                   "failed": 0, "trained": 0, "loss": None,
-                                  "round_failed": False}
+                                  "round_failed": False,
+                                  "success_per_tier": [0] * len(self.tier_names),
+                                  "failed_per_tier": [0] * len(self.tier_names)}
         ops = job.operators or [("train", "train")]
         for entry in ops:
             name, kind = (entry if isinstance(entry, (tuple, list))
@@ -338,9 +367,11 @@ class LogicalEngine:
                 "logical_result": [{
                     "name": job.data_name,
                     "simulation_target": {
-                        "devices": [job.device_tier],
-                        "success_num": [rec["success"]],
-                        "failed_num": [rec["failed"]],
+                        "devices": list(self.tier_names),
+                        "success_num": rec.get("success_per_tier",
+                                               [rec["success"]]),
+                        "failed_num": rec.get("failed_per_tier",
+                                              [rec["failed"]]),
                     }}]},
             "round_failed": rec["round_failed"],
             "loss": rec["loss"],

@@ -141,6 +141,10 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
                      if first_data and first_data.total_simulation.devices
                      else "high"),
         dynamic_num=dynamic,
+        tier_counts=[(t.tier, t.logical) for a in allocations[:1]
+                     for t in a.tiers],
+        dynamic_nums=(list(first_data.total_simulation.dynamic_nums)
+                      if first_data else []),
     )
     return job
 

@@ -140,3 +140,45 @@ def test_federated_training_converges():
     rec = out["records"][-1]
     assert rec["eval_acc"] > 0.35      # chance = 0.2 on 5 classes
     assert rec["eval_loss"] < 1.55     # below ln(5) ~ 1.61
+
+
+def test_per_tier_success_failed_accounting():
+    """Client ids map to device tiers by prefix ranges (reference
+    High/Middle/Low tiers, task_manager.py logical_result vectors)."""
+    import json
+    strategy = json.dumps({"offline_simulation": {"offline_probability": 1.0}})
+    eng = LogicalEngine(_job(
+        behavior_strategy=strategy, dynamic_num=100,
+        tier_counts=[("high", 6), ("low", 4)], dynamic_nums=[100, 100]))
+    rec = eng.run_round(0)
+    assert rec["failed_per_tier"] == [6, 4]
+    assert rec["success_per_tier"] == [0, 0]
+
+
+def test_per_tier_result_rows_and_tolerance():
+    rows = []
+    eng = LogicalEngine(_job(rounds=1,
+                             tier_counts=[("high", 7), ("middle", 3)],
+                             dynamic_nums=[0, 0]),
+                        result_sink=rows.append)
+    out = eng.run()
+    rec = out["records"][0]
+    assert rec["success_per_tier"] == [7, 3]
+    assert not rec["round_failed"]
+    tgt = rows[0]["logical_result"]["logical_result"][0]["simulation_target"]
+    assert tgt["devices"] == ["high", "middle"]
+    assert tgt["success_num"] == [7, 3]
+    assert tgt["failed_num"] == [0, 0]
+
+
+def test_per_tier_tolerance_fails_on_one_tier():
+    """round_failed is per tier: a tier over its own dynamic_num fails
+    the round even if the total stays under the summed tolerance."""
+    import json
+    strategy = json.dumps({"offline_simulation": {"offline_probability": 1.0}})
+    eng = LogicalEngine(_job(behavior_strategy=strategy, rounds=1,
+                             dynamic_num=100,
+                             tier_counts=[("high", 6), ("low", 4)],
+                             dynamic_nums=[100, 2]))
+    rec = eng.run_round(0)
+    assert rec["round_failed"]  # low tier: 4 failed > 2 tolerated
