@@ -21,14 +21,23 @@ non_grpc/deviceflow.py:29-197):
 
 from __future__ import annotations
 
+import json
 import random
 import threading
 from typing import Dict, List, Optional
 
 from ..utils.logging import Logger
+from ..utils.sqlite_repo import SqlTableRepo
 from .dispatcher import Dispatcher
 from .registry import TaskOrientedDeviceFlowRegistry
 from .rooms import InboundRoom, Message, OutboundRoom, ShelfRoom
+
+_FLOW_COLUMNS = {
+    "flow_id": "TEXT", "task_id": "TEXT", "strategy": "TEXT",
+    "outbound_service": "TEXT", "resources": "TEXT",
+    "notify_start_called": "TEXT", "notify_complete_called": "TEXT",
+    "is_finished": "INTEGER",
+}
 
 
 class FlowState:
@@ -53,6 +62,30 @@ class FlowState:
     def all_completed(self) -> bool:
         return all(self.notify_complete_called.values())
 
+    def to_row(self) -> Dict[str, object]:
+        return {
+            "flow_id": self.flow_id, "task_id": self.task_id,
+            "strategy": self.strategy,
+            "outbound_service": self.outbound_service,
+            "resources": json.dumps(self.resources),
+            "notify_start_called": json.dumps(self.notify_start_called),
+            "notify_complete_called": json.dumps(self.notify_complete_called),
+            "is_finished": int(self.is_finished),
+        }
+
+    @classmethod
+    def from_row(cls, row: Dict[str, object]) -> "FlowState":
+        f = cls(str(row["task_id"]), str(row["flow_id"]),
+                str(row["strategy"] or ""),
+                str(row["outbound_service"] or ""),
+                json.loads(str(row["resources"] or "[]")))
+        f.notify_start_called.update(
+            json.loads(str(row["notify_start_called"] or "{}")))
+        f.notify_complete_called.update(
+            json.loads(str(row["notify_complete_called"] or "{}")))
+        f.is_finished = bool(row.get("is_finished"))
+        return f
+
 
 class DeviceFlowService:
     def __init__(self, db_path: str = ":memory:", time_scale: float = 1.0,
@@ -69,8 +102,36 @@ class DeviceFlowService:
         self._stop = threading.Event()
         self.log = Logger.shared()
         self._sorter_thread: Optional[threading.Thread] = None
+        self._flow_repo = SqlTableRepo(db_path, "deviceflow_flow_table",
+                                       _FLOW_COLUMNS, primary_key="flow_id")
+        self._initiate_from_repo()
         if auto_start:
             self.start()
+
+    def _initiate_from_repo(self) -> None:
+        """Crash recovery: rebuild unfinished flows (and their
+        dispatchers) from the persisted flow table, like the reference's
+        initiate_from_repo (deviceflow_server.py:83-164).  Only tasks
+        still present in the registry are revived — a released task's
+        rows are stale and are dropped."""
+        for row in self._flow_repo.get_all_rows():
+            flow = FlowState.from_row(row)
+            if flow.is_finished or not self.registry.is_registered(flow.task_id):
+                self._flow_repo.delete_item("flow_id", flow.flow_id)
+                continue
+            self.flows[flow.flow_id] = flow
+            self.shelf.ensure_shelf(flow.flow_id)
+            if flow.strategy:
+                d = Dispatcher(flow.flow_id, flow.strategy, self.shelf,
+                               self.outbound, self.time_scale, self._rng)
+                self.dispatchers[flow.flow_id] = d
+                d.start()
+                if flow.all_completed():
+                    flow.to_sort = False
+                    d.release_event.set()
+
+    def _persist_flow(self, flow: FlowState) -> None:
+        self._flow_repo.upsert_item("flow_id", flow.to_row())
 
     # -- lifecycle --------------------------------------------------------
     def start(self) -> None:
@@ -122,6 +183,7 @@ class DeviceFlowService:
                                self.outbound, self.time_scale, self._rng)
                 self.dispatchers[flow_id] = d
                 d.start()
+            self._persist_flow(flow)
         return flow_id
 
     def notify_complete(self, task_id: str, operator_name: str,
@@ -137,6 +199,7 @@ class DeviceFlowService:
                 d = self.dispatchers.get(flow_id)
                 if d is not None:
                     d.release_event.set()
+            self._persist_flow(flow)
         return True
 
     def check_dispatch_finished(self, task_id: str) -> bool:
@@ -173,6 +236,7 @@ class DeviceFlowService:
         flow = self.flows.pop(flow_id, None)
         if flow is not None:
             flow.is_finished = True
+        self._flow_repo.delete_item("flow_id", flow_id)
         self.shelf.remove_shelf(flow_id)
 
     # -- data plane -------------------------------------------------------

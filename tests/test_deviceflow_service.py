@@ -140,3 +140,46 @@ def test_registry_persists_across_restart(tmp_path):
     assert s2.registry.is_registered("t")
     assert s2.registry.resources("t") == ["logical_simulation"]
     s2.shutdown()
+
+
+def test_crash_recovery_rebuilds_flows_from_repo(tmp_path):
+    """A restarted service revives unfinished flows and their
+    dispatchers from the persisted flow table (reference
+    deviceflow_server.py:83-164 initiate_from_repo)."""
+    db = str(tmp_path / "df.sqlite")
+    s1 = DeviceFlowService(db_path=db, time_scale=0.0, seed=1)
+    s1.register_task("t_rec", ["logical_simulation", "device_simulation"])
+    fid = s1.notify_start("t_rec", "train", 0, "logical_simulation",
+                          strategy=rt_strategy(batch=1))
+    assert fid == "t_rec_train_0"
+    s1.publish(fid, "logical_simulation", {"grad": 1})
+    assert s1.drain_inbound()
+    s1.shutdown()  # simulated crash: in-memory flows lost
+
+    s2 = DeviceFlowService(db_path=db, time_scale=0.0, seed=1)
+    try:
+        # flow revived with its per-resource start map intact
+        assert fid in s2.flows
+        flow = s2.flows[fid]
+        assert flow.notify_start_called["logical_simulation"]
+        assert not flow.notify_start_called["device_simulation"]
+        assert fid in s2.dispatchers
+        # lifecycle continues where it left off
+        assert s2.notify_start("t_rec", "train", 0, "device_simulation") == fid
+        s2.publish(fid, "device_simulation", {"grad": 2})
+        assert s2.drain_inbound()
+        assert s2.notify_complete("t_rec", "train", 0, "logical_simulation")
+        assert s2.notify_complete("t_rec", "train", 0, "device_simulation")
+        assert wait_until(lambda: s2.check_dispatch_finished("t_rec"))
+    finally:
+        s2.shutdown()
+
+    # released/unregistered task rows are dropped, not revived
+    s2b = DeviceFlowService(db_path=db, time_scale=0.0, auto_start=False)
+    s2b.unregister_task("t_rec")
+    s2b.shutdown()
+    s3 = DeviceFlowService(db_path=db, time_scale=0.0)
+    try:
+        assert fid not in s3.flows
+    finally:
+        s3.shutdown()
