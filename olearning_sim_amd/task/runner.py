@@ -31,6 +31,7 @@ from ..engine.job import EngineJob
 from ..utils.logging import Logger
 from .allocation import HybridOptimizer, DataAllocation, BETA, LAMBDA
 from .schema import TaskConfig
+from .submitter import JobSubmitter, fix_device_task_json, json2deviceconfig
 from .status import JobStatus
 from .table import TaskTableRepo
 
@@ -188,6 +189,21 @@ class TaskRunner:
         if device_target["device_target"]:
             self.table.set_item_value(task.task_id, "device_target",
                                       json.dumps(device_target))
+
+        # per-side assembled task JSONs (reference JobSubmitter,
+        # utils_runner.py:478-628) — persisted so the exact config each
+        # side ran under is auditable from the task table
+        sub = JobSubmitter(task, allocations)
+        side_logical = sub.assemble_info_logical_simulation()
+        if side_logical is not None:
+            self.table.set_item_value(task.task_id, "logical_task_params",
+                                      json.dumps(side_logical))
+        side_device = sub.assemble_info_device_simulation()
+        if side_device is not None:
+            device_cfg = fix_device_task_json(
+                json2deviceconfig(side_device), task.task_id)
+            self.table.set_item_value(task.task_id, "device_task_params",
+                                      json.dumps(device_cfg))
 
         if self.deviceflow is not None:
             for op in task.operatorflow.operators:

@@ -22,6 +22,8 @@ TASK_COLUMNS = {
                                     #        data_name_list, total_simulation}
     "logical_target": "TEXT",       # JSON: {"logical_target": [...]}
     "device_target": "TEXT",
+    "logical_task_params": "TEXT",  # per-side assembled task JSON
+    "device_task_params": "TEXT",   # lowered device-farm config JSON
     "logical_result": "TEXT",       # JSON: {"logical_result": [...]}
     "device_result": "TEXT",
     "logical_round": "INTEGER",
