@@ -101,7 +101,7 @@ def test_runner_persists_per_side_params():
     assert ok, msg
     assert mgr.step_schedule() == "t_side"
     st = wait_terminal(mgr, "t_side")
-    assert st.value == "SUCCEED"
+    assert st.value == "SUCCEEDED"
     raw = mgr.table.get_item_value("t_side", "logical_task_params")
     assert raw
     side = json.loads(raw)
