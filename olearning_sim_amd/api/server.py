@@ -47,6 +47,12 @@ class RegisterBody(BaseModel):
     total_compute_resources: list
 
 
+class PublishBody(BaseModel):
+    routing_key: str
+    compute_resource: str
+    payload: Optional[Any] = None
+
+
 class ClusterBody(BaseModel):
     name: str
     replicas: int = 1
@@ -158,6 +164,29 @@ def build_app(session) -> FastAPI:
         @app.get("/deviceflow/CheckDeviceflowDispatchFinished/{task_id}")
         def check_finished(task_id: str):
             return {"is_finished": df.check_dispatch_finished(task_id)}
+
+        # connection-info RPCs (GetDeviceflowPulsarClient /
+        # GetDeviceflowWebsocket analogues) + the in-process data plane
+        @app.get("/deviceflow/GetInboundInfo")
+        def get_inbound_info():
+            return df.inbound_info()
+
+        @app.get("/deviceflow/GetOutboundInfo")
+        def get_outbound_info():
+            return df.outbound_info()
+
+        @app.post("/deviceflow/publish")
+        def df_publish(body: PublishBody):
+            df.publish(body.routing_key, body.compute_resource, body.payload)
+            return {"is_success": True}
+
+        @app.get("/deviceflow/outbound")
+        def df_outbound(max_items: int = 100):
+            return {"messages": [
+                {"routing_key": m.routing_key,
+                 "compute_resource": m.compute_resource,
+                 "payload": m.payload}
+                for m in df.outbound.drain(max_items)]}
 
     # -- PerformanceMgr service (performanceService.proto) ---------------
     if session.performance_mgr is not None:

@@ -239,6 +239,21 @@ class DeviceFlowService:
         self._flow_repo.delete_item("flow_id", flow_id)
         self.shelf.remove_shelf(flow_id)
 
+    # -- connection info (GetDeviceflowPulsarClient /
+    # GetDeviceflowWebsocket, deviceflow_server.py:167-179) --------------
+    def inbound_info(self) -> Dict[str, object]:
+        """How producers reach the gradient house.  The reference hands
+        out a Pulsar url+topic; here the inbound is the in-process
+        queue, published to via the /deviceflow/publish route."""
+        return {"kind": "inproc", "endpoint": "/deviceflow/publish",
+                "queue_depth": self.inbound.qsize()}
+
+    def outbound_info(self) -> Dict[str, object]:
+        """Where aggregated fragments leave the gradient house (the
+        reference's websocket/Pulsar outbound)."""
+        return {"kind": "inproc", "endpoint": "/deviceflow/outbound",
+                "queue_depth": self.outbound.qsize()}
+
     # -- data plane -------------------------------------------------------
     def publish(self, routing_key: str, compute_resource: str,
                 payload=None) -> None:

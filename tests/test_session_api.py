@@ -223,3 +223,42 @@ def test_real_http_server_round_trip(tmp_path):
     finally:
         server.should_exit = True
         s.shutdown()
+
+
+def test_deviceflow_data_plane_over_http(client):
+    """Publish -> sorter -> dispatcher -> outbound, all through the JSON
+    routes (GetDeviceflowPulsarClient/GetDeviceflowWebsocket analogues
+    plus the in-process inbound/outbound endpoints)."""
+    import json as _json
+    import time as _time
+    assert client.get("/deviceflow/GetInboundInfo").json()["kind"] == "inproc"
+    assert client.get("/deviceflow/GetOutboundInfo").json()[
+        "endpoint"] == "/deviceflow/outbound"
+
+    strategy = _json.dumps({"real_time_dispatch": {
+        "use_strategy": True, "dispatch_batch_sizes": [1]}})
+    assert client.post("/deviceflow/RegisterTask", json={
+        "task_id": "t_dp",
+        "total_compute_resources": ["logical_simulation"]}).json()["is_success"]
+    fid = client.post("/deviceflow/NotifyStart", json={
+        "task_id": "t_dp", "operator_name": "train", "round": 0,
+        "compute_resource": "logical_simulation",
+        "strategy": strategy}).json()["flow_id"]
+    for i in range(3):
+        assert client.post("/deviceflow/publish", json={
+            "routing_key": fid, "compute_resource": "logical_simulation",
+            "payload": {"i": i}}).json()["is_success"]
+    t0 = _time.time()
+    while _time.time() - t0 < 10:
+        if client.get("/deviceflow/GetInboundInfo").json()["queue_depth"] == 0:
+            break
+        _time.sleep(0.01)
+    assert client.post("/deviceflow/NotifyComplete", json={
+        "task_id": "t_dp", "operator_name": "train", "round": 0,
+        "compute_resource": "logical_simulation"}).json()["is_success"]
+    got = []
+    t0 = _time.time()
+    while _time.time() - t0 < 10 and len(got) < 3:
+        got += client.get("/deviceflow/outbound").json()["messages"]
+        _time.sleep(0.01)
+    assert sorted(m["payload"]["i"] for m in got) == [0, 1, 2]
