@@ -40,7 +40,10 @@ class LogicalEngine:
                  dist_ctx: Optional[pdist.DistContext] = None,
                  behavior: Optional[BehaviorFn] = None,
                  result_sink: Optional[Callable[[Dict[str, Any]], None]] = None,
-                 deviceflow=None, perf=None):
+                 deviceflow=None, perf=None, script_ops=None):
+        # script_ops: {operator_name: ScriptOperator} — user script-file
+        # operators executed as sharded subprocesses (engine/script_op.py)
+        self.script_ops = script_ops or {}
         # deviceflow: gradient-house service handle — the engine drives
         # the reference's NotifyStart/NotifyComplete lifecycle per round
         # (run_task.py:234-308) and publishes per-chunk summary messages
@@ -281,6 +284,30 @@ class LogicalEngine:
             "round_failed": round_failed,
         }
 
+    def _op_script(self, round_idx: int, name: str) -> Dict[str, Any]:
+        """User script-file operator: sharded subprocess execution of
+        the staged entry file (reference Actor loop_run,
+        utils_run_task.py:481-514).  Counts feed the same per-tier
+        tolerance accounting as the in-process train path."""
+        from .script_op import per_tier_counts
+        op = self.script_ops.get(name)
+        if op is None:
+            raise RuntimeError(f"script operator {name!r} not staged")
+        res = op.run_round(round_idx)
+        succ_t, fail_t = per_tier_counts(res["failed_ranges"],
+                                         self.tier_bounds)
+        self.success_total += res["success"]
+        self.failed_total += res["failed"]
+        dyn = self.tier_dynamic
+        if len(dyn) != len(succ_t):
+            dyn = [self.job.dynamic_num] * len(succ_t)
+        round_failed = (res["failed"] > self.job.dynamic_num
+                        if len(succ_t) == 1
+                        else any(f > d for f, d in zip(fail_t, dyn)))
+        return {"success": res["success"], "failed": res["failed"],
+                "success_per_tier": succ_t, "failed_per_tier": fail_t,
+                "round_failed": round_failed}
+
     def run_round(self, round_idx: int) -> Dict[str, Any]:
         """Execute the round's ordered operator list
         (reference run_task.py:228-311: per operator -> deviceflow
@@ -298,6 +325,8 @@ class LogicalEngine:
                           else (entry, "train"))
             if kind == "train":
                 record.update(self._op_train(round_idx, name))
+            elif kind == "script":
+                record.update(self._op_script(round_idx, name))
             elif kind == "evaluate":
                 record.update(self.evaluate_global(round_idx))
             elif kind == "checkpoint":

@@ -21,6 +21,7 @@ dirichlet_alpha, dtype, chunk_clients, vocab_size, seq_len.
 
 from __future__ import annotations
 
+import os
 import json
 import threading
 import time
@@ -93,8 +94,11 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
             op_params = {}
         kind = op_params.get("kind")
         if kind is None:
+            code = op.logical_simulation.operator_code_path
             nm = op.name.lower()
-            if "eval" in nm:
+            if code and not code.startswith("builtin:"):
+                kind = "script"     # user script-file operator
+            elif "eval" in nm:
                 kind = "evaluate"
             elif "checkpoint" in nm or "save" in nm:
                 kind = "checkpoint"
@@ -220,6 +224,48 @@ class TaskRunner:
             self._submit_phone(task, allocations)
         return job_id
 
+    def _stage_script_operators(self, task: TaskConfig, job: EngineJob):
+        """Stage user script-file operator code and build executors
+        (reference get_operator_code + Actor, utils_runner.py:684-782,
+        utils_run_task.py:146-577)."""
+        script_names = {name for name, kind in job.operators
+                        if kind == "script"}
+        if not script_names:
+            return {}
+        import tempfile
+        from ..engine.script_op import ScriptOperator
+        from .staging import stage_operator_code
+        work_root = os.path.join(self.checkpoint_dir or tempfile.gettempdir(),
+                                 f"opwork_{task.task_id}")
+        first_data = task.target.data[0] if task.target.data else None
+        data_info = ({"name": first_data.name,
+                      "data_path": first_data.data_path,
+                      "data_split_type": first_data.data_split_type,
+                      "task_type": first_data.task_type}
+                     if first_data else {})
+        ops = {}
+        for op in task.operatorflow.operators:
+            if op.name not in script_names:
+                continue
+            sim = op.logical_simulation
+            staged = stage_operator_code(
+                sim.operator_code_path, sim.operator_entry_file, op.name,
+                work_root)
+            model_info = {
+                "use_model": op.model.use_model,
+                "model_path": op.model.model_path,
+                "model_update_style": op.model.model_update_style,
+            }
+            ops[op.name] = ScriptOperator(
+                name=op.name, staged_dir=staged,
+                entry_file=sim.operator_entry_file,
+                operator_params=sim.operator_params,
+                task_id=task.task_id, work_dir=work_root,
+                clients=job.clients,
+                shards=min(8, max(1, job.clients // 4)),
+                data_info=data_info, model_info=model_info)
+        return ops
+
     @staticmethod
     def _sides(allocations: List[DataAllocation]) -> List[str]:
         sides = []
@@ -234,6 +280,7 @@ class TaskRunner:
                         allocations: List[DataAllocation]) -> str:
         job = engine_job_from_task(task, allocations, self.device,
                                    self.checkpoint_dir)
+        script_ops = self._stage_script_operators(task, job)
         job_id = f"olsjob_{uuid.uuid4().hex[:12]}"
         handle = JobHandle(job_id, "logical")
 
@@ -251,7 +298,7 @@ class TaskRunner:
 
                 eng = LogicalEngine(job, result_sink=sink,
                                     deviceflow=self.deviceflow,
-                                    perf=self.perf)
+                                    perf=self.perf, script_ops=script_ops)
                 # cooperative stop (reference: JobSubmissionClient.stop_job)
                 orig_run_round = eng.run_round
 
