@@ -175,6 +175,10 @@ def build_app(session) -> FastAPI:
         def get_outbound_info():
             return df.outbound_info()
 
+        @app.get("/deviceflow/dispatchCurve/{task_id}")
+        def dispatch_curve(task_id: str):
+            return {"flows": df.dispatch_curve(task_id)}
+
         @app.post("/deviceflow/publish")
         def df_publish(body: PublishBody):
             df.publish(body.routing_key, body.compute_resource, body.payload)

@@ -43,7 +43,16 @@ class Dispatcher:
         self.finished = threading.Event()
         self.forwarded = 0
         self.dropped = 0
+        # per-slot dispatch curve: (t_virtual_s, sent, dropped, cumulative)
+        # — the reference's operation_amount_table /
+        # accumulated_amount_table demo rows (dispatcher.py:254-395)
+        self.history: List[tuple] = []
+        self._t_virtual = 0.0
         self._thread: Optional[threading.Thread] = None
+
+    def _record_slot(self, wait_s: float, sent: int, dropped: int) -> None:
+        self._t_virtual += wait_s
+        self.history.append((self._t_virtual, sent, dropped, self.forwarded))
 
     # ------------------------------------------------------------------
     def start(self) -> None:
@@ -75,12 +84,18 @@ class Dispatcher:
                                    timeout=0.01)
             batch.extend(msgs)
             if len(batch) >= target and target > 0:
+                n0, d0 = self.forwarded, self.dropped
                 self._forward_with_drop(batch, drop_p)
+                self._record_slot(0.0, self.forwarded - n0, self.dropped - d0)
                 batch = []
                 batch_idx += 1
             elif self.release_event.is_set() and \
                     self.shelf.depth(self.flow_id) == 0:
+                n0, d0 = self.forwarded, self.dropped
                 self._forward_with_drop(batch, drop_p)   # flush remainder
+                if len(batch):
+                    self._record_slot(0.0, self.forwarded - n0,
+                                      self.dropped - d0)
                 return
 
     def _forward_with_drop(self, batch: List[Message], drop_p: float) -> None:
@@ -105,12 +120,14 @@ class Dispatcher:
                 time.sleep(wait_s * self.time_scale)
             msgs = self.shelf.take(self.flow_id, amount, timeout=0.05)
             dropset = set(drop_idx)
+            n0, d0 = self.forwarded, self.dropped
             for i, m in enumerate(msgs):
                 if i in dropset:
                     self.dropped += 1
                 else:
                     self.outbound.send(m)
                     self.forwarded += 1
+            self._record_slot(wait_s, self.forwarded - n0, self.dropped - d0)
         # clean_remain_message (dispatcher.py:244-252)
         for m in self.shelf.take(self.flow_id, 1_000_000):
             self.outbound.send(m)

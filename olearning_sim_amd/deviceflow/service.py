@@ -24,7 +24,7 @@ from __future__ import annotations
 import json
 import random
 import threading
-from typing import Dict, List, Optional
+from typing import Dict, List, Optional, Tuple
 
 from ..utils.logging import Logger
 from ..utils.sqlite_repo import SqlTableRepo
@@ -96,6 +96,9 @@ class DeviceFlowService:
         self.outbound = OutboundRoom()
         self.flows: Dict[str, FlowState] = {}
         self.dispatchers: Dict[str, Dispatcher] = {}
+        # released flows keep their dispatch curves for dashboards
+        # (reference operation/accumulated_amount_table demo rows)
+        self.dispatch_history: Dict[str, List[tuple]] = {}
         self.time_scale = time_scale
         self._rng = random.Random(seed)
         self._lock = threading.RLock()
@@ -233,11 +236,29 @@ class DeviceFlowService:
         if d is not None:
             d.stop_event.set()
             d.release_event.set()
+            if d.history:
+                self.dispatch_history[flow_id] = list(d.history)
         flow = self.flows.pop(flow_id, None)
         if flow is not None:
             flow.is_finished = True
         self._flow_repo.delete_item("flow_id", flow_id)
         self.shelf.remove_shelf(flow_id)
+
+    def dispatch_curve(self, task_id: str) -> Dict[str, List[Dict[str, float]]]:
+        """Per-flow dispatch curves: the per-slot (virtual time, sent,
+        dropped, cumulative) rows the reference wrote to its demo
+        operation_amount / accumulated_amount tables
+        (dispatcher.py:254-395)."""
+        out: Dict[str, List[Dict[str, float]]] = {}
+        with self._lock:
+            sources: List[Tuple[str, List[tuple]]] = [
+                (fid, list(d.history)) for fid, d in self.dispatchers.items()]
+            sources += list(self.dispatch_history.items())
+        for fid, hist in sources:
+            if fid.startswith(task_id + "_"):
+                out[fid] = [{"t": t, "sent": s, "dropped": dr,
+                             "accumulated": acc} for t, s, dr, acc in hist]
+        return out
 
     # -- connection info (GetDeviceflowPulsarClient /
     # GetDeviceflowWebsocket, deviceflow_server.py:167-179) --------------

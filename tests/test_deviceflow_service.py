@@ -183,3 +183,22 @@ def test_crash_recovery_rebuilds_flows_from_repo(tmp_path):
         assert fid not in s3.flows
     finally:
         s3.shutdown()
+
+
+def test_dispatch_curve_recorded(svc):
+    """Per-slot dispatch history survives flow release (reference demo
+    operation_amount/accumulated_amount tables)."""
+    svc.register_task("t_curve", ["logical_simulation"])
+    fid = svc.notify_start("t_curve", "train", 0, "logical_simulation",
+                           strategy=rt_strategy(batch=2))
+    for i in range(6):
+        svc.publish(fid, "logical_simulation", {"i": i})
+    assert svc.drain_inbound()
+    svc.notify_complete("t_curve", "train", 0, "logical_simulation")
+    assert wait_until(lambda: svc.check_dispatch_finished("t_curve"))
+    curve = svc.dispatch_curve("t_curve")[fid]
+    assert sum(row["sent"] for row in curve) == 6
+    assert curve[-1]["accumulated"] == 6
+    # release the flow; the curve must still be queryable
+    svc.flow_release_step()
+    assert svc.dispatch_curve("t_curve")[fid][-1]["accumulated"] == 6
