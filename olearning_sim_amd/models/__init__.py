@@ -4,11 +4,13 @@ from .base import ClientBatchedModel, Params
 from .mlp import MLP
 from .lenet import LeNet5
 from .resnet import ResNet18
+from .lstm import CharLSTM
 
 _REGISTRY = {
     "mlp": MLP,
     "lenet": LeNet5,
     "resnet18": ResNet18,
+    "lstm": CharLSTM,
 }
 
 
@@ -33,4 +35,4 @@ def build_model(name: str, **kwargs: Any) -> ClientBatchedModel:
 
 
 __all__ = ["ClientBatchedModel", "Params", "MLP", "LeNet5", "ResNet18",
-           "build_model"]
+           "CharLSTM", "build_model"]
