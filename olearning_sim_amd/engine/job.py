@@ -78,7 +78,11 @@ class EngineJob:
     # client ids are assigned to tiers in order (prefix ranges); empty
     # means a single job.device_tier tier of all clients
     tier_counts: List[Any] = field(default_factory=list)
-    dynamic_nums: List[int] = field(default_factory=list)  # per tier
+    dynamic_nums: List[int] = field(default_factory=list)  # per segment
+    # full (data x tier) segment list for multi-data tasks:
+    # [(data_name, tier, clients), ...] in client-id prefix order;
+    # overrides tier_counts when non-empty
+    data_segments: List[Any] = field(default_factory=list)
 
     def resolved_cohort(self) -> int:
         return self.cohort_size if self.cohort_size > 0 else self.clients

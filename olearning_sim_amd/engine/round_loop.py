@@ -100,13 +100,20 @@ class LogicalEngine:
         self.success_total = 0
         self.failed_total = 0
         self.last_operator = "train"
-        # per-tier client-id prefix ranges (reference device tiers)
-        tiers = job.tier_counts or [(job.device_tier, job.clients)]
-        self.tier_names = [t for t, _ in tiers]
+        # (data x tier) client-id prefix segments (reference: success/
+        # failed vectors per data target and device tier)
+        if job.data_segments:
+            segs = [(d, t, int(n)) for d, t, n in job.data_segments]
+        elif job.tier_counts:
+            segs = [(job.data_name, t, int(n)) for t, n in job.tier_counts]
+        else:
+            segs = [(job.data_name, job.device_tier, job.clients)]
+        self.segments = segs
+        self.tier_names = [t for _, t, _ in segs]
         bounds = [0]
-        for _, n in tiers:
-            bounds.append(bounds[-1] + int(n))
-        self.tier_bounds = bounds          # len = T+1; covers [0, clients)
+        for _, _, n in segs:
+            bounds.append(bounds[-1] + n)
+        self.tier_bounds = bounds          # len = S+1; covers [0, clients)
         self.tier_dynamic = (list(job.dynamic_nums)
                              if job.dynamic_nums else [job.dynamic_num])
 
@@ -384,6 +391,25 @@ class LogicalEngine:
             "records": records,
         }
 
+    def _result_entries(self, rec: Dict[str, Any]) -> List[Dict[str, Any]]:
+        """Group the per-segment counts back into one result entry per
+        data target (run_task.py analyze_results:149-210 shape)."""
+        succ = rec.get("success_per_tier", [rec["success"]])
+        fail = rec.get("failed_per_tier", [rec["failed"]])
+        order: List[str] = []
+        by_data: Dict[str, Dict[str, list]] = {}
+        for i, (dn, tier, _n) in enumerate(self.segments):
+            if dn not in by_data:
+                by_data[dn] = {"devices": [], "success_num": [],
+                               "failed_num": []}
+                order.append(dn)
+            t = by_data[dn]
+            t["devices"].append(tier)
+            t["success_num"].append(succ[i] if i < len(succ) else 0)
+            t["failed_num"].append(fail[i] if i < len(fail) else 0)
+        return [{"name": dn, "simulation_target": by_data[dn]}
+                for dn in order]
+
     def _round_result(self, rec: Dict[str, Any]) -> Dict[str, Any]:
         """Shape a round record like the reference's logical_result row
         (run_task.py analyze_results:149-210)."""
@@ -392,16 +418,7 @@ class LogicalEngine:
             "task_id": job.task_id,
             "logical_round": rec["round"] + 1,
             "logical_operator": self.last_operator,
-            "logical_result": {
-                "logical_result": [{
-                    "name": job.data_name,
-                    "simulation_target": {
-                        "devices": list(self.tier_names),
-                        "success_num": rec.get("success_per_tier",
-                                               [rec["success"]]),
-                        "failed_num": rec.get("failed_per_tier",
-                                              [rec["failed"]]),
-                    }}]},
+            "logical_result": {"logical_result": self._result_entries(rec)},
             "round_failed": rec["round_failed"],
             "loss": rec["loss"],
         }

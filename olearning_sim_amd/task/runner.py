@@ -148,10 +148,31 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
         dynamic_num=dynamic,
         tier_counts=[(t.tier, t.logical) for a in allocations[:1]
                      for t in a.tiers],
-        dynamic_nums=(list(first_data.total_simulation.dynamic_nums)
-                      if first_data else []),
+        data_segments=[(a.data_name, t.tier, t.logical)
+                       for a in allocations for t in a.tiers
+                       if t.logical > 0],
+        dynamic_nums=_segment_dynamic_nums(task, allocations),
     )
     return job
+
+
+def _segment_dynamic_nums(task: TaskConfig,
+                          allocations: List[DataAllocation]) -> List[int]:
+    """Per-(data x tier) failure tolerance aligned with data_segments."""
+    by_data = {d.name: d for d in task.target.data}
+    out: List[int] = []
+    for a in allocations:
+        d = by_data.get(a.data_name)
+        dyn = list(d.total_simulation.dynamic_nums) if d else []
+        devs = list(d.total_simulation.devices) if d else []
+        for t in a.tiers:
+            if t.logical <= 0:
+                continue
+            try:
+                out.append(dyn[devs.index(t.tier)])
+            except (ValueError, IndexError):
+                out.append(0)
+    return out
 
 
 class TaskRunner:

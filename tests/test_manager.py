@@ -286,3 +286,34 @@ def test_three_tasks_contend_for_resources_and_all_finish():
     for tid in done:
         assert mgr.get_task_status(tid) == TaskStatus.SUCCEEDED
         assert not mgr.resources.holding(tid)
+
+
+def test_multi_data_task_reports_each_data():
+    """A task with two data targets gets per-data success/failed
+    vectors in logical_result (reference analyze_results accumulates
+    per data name, run_task.py:149-210)."""
+    import copy
+    raw = json.loads(task_json(task_id="t_multi", rounds=2, clients=6,
+                               dynamic=1))
+    d1 = copy.deepcopy(raw["target"]["data"][0])
+    d1["name"] = "data_1"
+    d1["total_simulation"] = {"devices": ["high"], "nums": [4],
+                              "dynamic_nums": [1]}
+    d1["allocation"] = {"optimization": False, "logical_simulation": [4],
+                        "device_simulation": [0],
+                        "running_response": {"devices": [], "nums": []}}
+    raw["target"]["data"].append(d1)
+    raw["logical_simulation"]["resource_request"].append(
+        {"name": "data_1", "devices": ["high"], "num_request": [1]})
+    mgr = make_manager()
+    ok, msg = mgr.submit_task(json.dumps(raw))
+    assert ok, msg
+    assert mgr.step_schedule() == "t_multi"
+    st = wait_terminal(mgr, "t_multi")
+    assert st == TaskStatus.SUCCEEDED
+    res = json.loads(mgr.table.get_row("t_multi")["logical_result"])
+    entries = {e["name"]: e["simulation_target"]
+               for e in res["logical_result"]}
+    assert entries["data_0"]["success_num"] == [6]
+    assert entries["data_1"]["success_num"] == [4]
+    mgr.shutdown()
