@@ -372,8 +372,15 @@ class TaskManager:
                         ridx = result_names.index(name)
                         succ = result[ridx].get("simulation_target", {}) \
                             .get("success_num", [])
-                        oks.append(all(x >= a - z for x, a, z in
-                                       zip(succ, tgt_nums, dyn)))
+                        # every targeted tier must be reported AND meet
+                        # its tolerance — a shorter vector is a miss
+                        oks.append(len(succ) >= len(tgt_nums)
+                                   and all(x >= a - z for x, a, z in
+                                           zip(succ, tgt_nums, dyn)))
+                    elif tgt_nums:
+                        # a targeted data with no result entry at all
+                        # cannot be a success
+                        oks.append(False)
             return bool(oks) and all(oks)
 
         # logical side
@@ -426,20 +433,25 @@ class TaskManager:
             dyn = data_total.get("simulation_target", {}).get("dynamic_nums", [])
             if not dyn:
                 dyn = [0] * len(nums)
+            def _pad(v, n):
+                # a result vector shorter than the target counts its
+                # missing tiers as 0 successes (never zip-truncates)
+                return list(v) + [0] * max(0, n - len(v))
+
             if name in logical_names:
                 li = logical_names.index(name)
-                lf = logical_result[li].get("simulation_target", {}) \
-                    .get("failed_num", [])
-                ls = logical_result[li].get("simulation_target", {}) \
-                    .get("success_num", [])
+                lf = _pad(logical_result[li].get("simulation_target", {})
+                          .get("failed_num", []), len(dyn))
+                ls = _pad(logical_result[li].get("simulation_target", {})
+                          .get("success_num", []), len(nums))
             else:
                 lf, ls = [0] * len(dyn), [0] * len(nums)
             if name in device_names:
                 di = device_names.index(name)
-                df = device_result[di].get("simulation_target", {}) \
-                    .get("failed_num", [])
-                ds = device_result[di].get("simulation_target", {}) \
-                    .get("success_num", [])
+                df = _pad(device_result[di].get("simulation_target", {})
+                          .get("failed_num", []), len(dyn))
+                ds = _pad(device_result[di].get("simulation_target", {})
+                          .get("success_num", []), len(nums))
             else:
                 df, ds = [0] * len(dyn), [0] * len(nums)
 

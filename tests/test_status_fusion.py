@@ -167,3 +167,52 @@ def test_missing_total_simulation_fails():
     st = mgr.combine_task_status("t", TaskStatus.RUNNING,
                                  {"is_finished": True, "device_result": []})
     assert st == TaskStatus.FAILED
+
+
+def test_missing_data_entry_is_not_success():
+    """A targeted data with no result entry must block SUCCEEDED (the
+    lenient skip would silently pass multi-data tasks that only
+    reported one data)."""
+    mgr, table = mk()
+    table.add_task("t2")
+    total = {"max_round": 1, "operator_name_list": ["train"],
+             "data_name_list": ["d0", "d1"],
+             "total_simulation": [
+                 {"name": "d0", "simulation_target": {
+                     "devices": ["high"], "nums": [10],
+                     "dynamic_nums": [2]}},
+                 {"name": "d1", "simulation_target": {
+                     "devices": ["high"], "nums": [5],
+                     "dynamic_nums": [0]}}]}
+    table.set_items("t2", total_simulation=json.dumps(total))
+    table.set_item_value("t2", "logical_target", json.dumps(
+        {"logical_target": [
+            {"name": "d0", "simulation_target": {"devices": ["high"],
+                                                 "nums": [10]}},
+            {"name": "d1", "simulation_target": {"devices": ["high"],
+                                                 "nums": [5]}}]}))
+    # only d0 reported
+    table.set_items("t2", logical_result=json.dumps({"logical_result": [
+        {"name": "d0", "simulation_target": {
+            "devices": ["high"], "success_num": [10],
+            "failed_num": [0]}}]}),
+        logical_round=1, logical_operator="train")
+    st = mgr.combine_task_status("t2", TaskStatus.SUCCEEDED,
+                                 {"is_finished": True})
+    assert st != TaskStatus.SUCCEEDED
+
+
+def test_short_success_vector_is_not_success():
+    """A result vector missing a targeted tier must not pass via zip
+    truncation."""
+    mgr, table = mk()
+    seed(table, "t3", max_round=1, nums=(10, 5), dynamic=(2, 0))
+    # only one tier reported for a two-tier target
+    table.set_items("t3", logical_result=json.dumps({"logical_result": [
+        {"name": "d0", "simulation_target": {
+            "devices": ["high"], "success_num": [10],
+            "failed_num": [0]}}]}),
+        logical_round=1, logical_operator="train")
+    st = mgr.combine_task_status("t3", TaskStatus.SUCCEEDED,
+                                 {"is_finished": True})
+    assert st != TaskStatus.SUCCEEDED
