@@ -118,7 +118,8 @@ class HybridOptimizer:
         machine-times (the 2024-10-23 constraint in
         utils_runner.py / utils.py:605-621)."""
         if phones <= 0 or n <= rr:
-            dev = rr if phones > 0 else 0
+            # running-response floor is capped by the population itself
+            dev = min(n, rr) if phones > 0 else 0
             return TierAllocation(tier, n, n - dev, dev, rr)
         if actors <= 0:
             return TierAllocation(tier, n, 0, n, rr)

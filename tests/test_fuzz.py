@@ -88,3 +88,61 @@ def test_drop_list_bounds(amounts, p):
     for d, a in zip(drops, amounts):
         assert all(0 <= i < max(1, a) for i in d)
         assert len(d) <= a
+
+
+@settings(max_examples=120, deadline=None)
+@given(n=st.integers(0, 5000), actors=st.integers(0, 64),
+       phones=st.integers(0, 50), rr=st.integers(0, 100))
+def test_auto_allocation_partitions_population(n, actors, phones, rr):
+    """The hybrid optimizer always partitions each tier's population
+    exactly between the two sides and honours the running-response
+    floor (utils_runner.py:939-1022 cost model)."""
+    from olearning_sim_amd.task.allocation import HybridOptimizer
+    t = HybridOptimizer._optimize_tier("high", n, actors, phones, rr)
+    assert t.logical >= 0 and t.device >= 0
+    assert t.logical + t.device == n
+    if phones > 0 and n > rr:
+        assert t.device >= rr
+
+
+@settings(max_examples=60, deadline=None)
+@given(log=st.lists(st.integers(0, 500), min_size=1, max_size=3),
+       dev=st.lists(st.integers(0, 500), min_size=1, max_size=3),
+       dyn=st.lists(st.integers(0, 50), min_size=1, max_size=3))
+def test_submitter_sides_partition_any_split(log, dev, dyn):
+    """assemble_info_* always partitions nums, and the two sides'
+    scaled dynamic_nums never exceed the original tolerance."""
+    import copy
+    from test_schema import EXAMPLE
+    from olearning_sim_amd.task import json2taskconfig as j2t
+    from olearning_sim_amd.task.allocation import HybridOptimizer
+    from olearning_sim_amd.task.submitter import JobSubmitter
+    k = min(len(log), len(dev), len(dyn))
+    log, dev, dyn = log[:k], dev[:k], dyn[:k]
+    nums = [a + b for a, b in zip(log, dev)]
+    if sum(nums) == 0:
+        return
+    raw = copy.deepcopy(EXAMPLE)
+    d = raw["target"]["data"][0]
+    tiers = [f"t{i}" for i in range(k)]
+    d["total_simulation"] = {"devices": tiers, "nums": nums,
+                             "dynamic_nums": dyn}
+    d["allocation"] = {"optimization": False, "logical_simulation": log,
+                       "device_simulation": dev,
+                       "running_response": {"devices": [], "nums": []}}
+    task = j2t(json.dumps(raw))
+    sub = JobSubmitter(task, HybridOptimizer(task).allocate())
+    ls = sub.assemble_info_logical_simulation()
+    ds = sub.assemble_info_device_simulation()
+    got_l = (ls["target"]["data"][0]["total_simulation"]
+             if ls else {"nums": [], "dynamic_nums": []})
+    got_d = (ds["target"]["data"][0]["total_simulation"]
+             if ds else {"nums": [], "dynamic_nums": []})
+    # per-tier partition of nums (absent side contributes 0)
+    for i, n in enumerate(nums):
+        ln = got_l["nums"][i] if i < len(got_l["nums"]) else 0
+        dn = got_d["nums"][i] if i < len(got_d["nums"]) else 0
+        if ls and ds:
+            assert ln + dn == n
+    assert (sum(got_l.get("dynamic_nums", []))
+            + sum(got_d.get("dynamic_nums", []))) <= sum(dyn)
