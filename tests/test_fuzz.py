@@ -146,3 +146,72 @@ def test_submitter_sides_partition_any_split(log, dev, dyn):
             assert ln + dn == n
     assert (sum(got_l.get("dynamic_nums", []))
             + sum(got_d.get("dynamic_nums", []))) <= sum(dyn)
+
+
+@settings(max_examples=80, deadline=None)
+@given(bounds=st.lists(st.integers(1, 50), min_size=1, max_size=5),
+       ranges=st.lists(st.tuples(st.integers(0, 200), st.integers(0, 60)),
+                       max_size=8))
+def test_per_tier_counts_conserve_population(bounds, ranges):
+    """Segment intersection accounting: success+failed == tier size per
+    tier, and totals never go negative, for any failed ranges."""
+    from olearning_sim_amd.engine.script_op import per_tier_counts
+    tier_bounds = [0]
+    for b in bounds:
+        tier_bounds.append(tier_bounds[-1] + b)
+    # clip ranges into [0, clients) and make them well-formed
+    clients = tier_bounds[-1]
+    fr = []
+    seen = set()
+    for lo, ln in ranges:
+        lo = lo % clients
+        hi = min(clients, lo + ln)
+        if lo < hi and all(hi <= a or lo >= b for a, b in seen):
+            fr.append((lo, hi))
+            seen.add((lo, hi))
+    succ, fail = per_tier_counts(fr, tier_bounds)
+    total_failed = sum(hi - lo for lo, hi in fr)
+    assert sum(fail) == total_failed
+    for t in range(len(bounds)):
+        assert succ[t] + fail[t] == bounds[t]
+        assert succ[t] >= 0 and fail[t] >= 0
+
+
+@settings(max_examples=100, deadline=None)
+@given(succ=st.lists(st.integers(0, 20), min_size=1, max_size=3),
+       fail=st.lists(st.integers(0, 20), min_size=1, max_size=3),
+       nums=st.lists(st.integers(1, 20), min_size=1, max_size=3),
+       dyn=st.lists(st.integers(0, 5), min_size=1, max_size=3),
+       rnd=st.integers(0, 3))
+def test_status_fusion_never_crashes_and_is_sound(succ, fail, nums, dyn, rnd):
+    """Fusion accepts any result-vector shape without crashing, and
+    SUCCEEDED implies every tier met its tolerance."""
+    import sys
+    sys.path.insert(0, "tests")
+    from test_status_fusion import mk
+    from olearning_sim_amd.task.status import TaskStatus
+    k = len(nums)
+    dyn = (dyn + [0] * k)[:k]
+    mgr, table = mk()
+    table.add_task("tf")
+    total = {"max_round": 2, "operator_name_list": ["train"],
+             "data_name_list": ["d0"],
+             "total_simulation": [{"name": "d0", "simulation_target": {
+                 "devices": [f"t{i}" for i in range(k)], "nums": nums,
+                 "dynamic_nums": dyn}}]}
+    table.set_items("tf", total_simulation=json.dumps(total))
+    table.set_item_value("tf", "logical_target", json.dumps(
+        {"logical_target": [{"name": "d0", "simulation_target": {
+            "devices": [f"t{i}" for i in range(k)], "nums": nums}}]}))
+    table.set_items("tf", logical_result=json.dumps({"logical_result": [
+        {"name": "d0", "simulation_target": {
+            "devices": [f"t{i}" for i in range(len(succ))],
+            "success_num": succ, "failed_num": fail}}]}),
+        logical_round=rnd, logical_operator="train")
+    st = mgr.combine_task_status("tf", TaskStatus.SUCCEEDED,
+                                 {"is_finished": True})
+    assert isinstance(st, TaskStatus)
+    if st == TaskStatus.SUCCEEDED:
+        assert rnd >= 2
+        padded = list(succ) + [0] * max(0, k - len(succ))
+        assert all(padded[i] >= nums[i] - dyn[i] for i in range(k))
