@@ -29,15 +29,21 @@ def _worker(rank, world, port, conn):
                                   "num_classes": 5},
                     clients=hi - lo, rounds=2, local_steps=1, batch_size=4,
                     lr=0.1, device="cpu", dtype="float32", num_classes=5,
-                    shard_size=8, seed=77)  # same model seed on all ranks
+                    shard_size=8, seed=77,  # same model seed on all ranks
+                    # two tiers per rank: covers the vector stats
+                    # all-reduce (1 + 2*segments doubles)
+                    tier_counts=[("high", (hi - lo) - 1), ("low", 1)],
+                    dynamic_nums=[0, 0])
     eng = LogicalEngine(job, dist_ctx=ctx)
     out = eng.run()
+    last = out["records"][-1]
     # plain bytes, NOT a tensor: torch tensors go through the Pipe by
     # shared-memory FD passing, which races with child exit
     conn.send({
         "rank": rank,
         "master_bytes": eng.master.flat.numpy().tobytes(),
         "success_total": out["success_total"],
+        "success_per_tier": last["success_per_tier"],
         "shard": (lo, hi),
     })
     conn.recv()                     # wait for the parent's ack
@@ -89,5 +95,8 @@ def test_two_rank_gloo_round_aggregates():
     torch.testing.assert_close(m0, m1)
     # success counts were all-reduced: both report the global total
     assert r0["success_total"] == r1["success_total"] == 2 * 8
+    # per-tier vectors all-reduced identically on both ranks:
+    # each rank contributes (3 high, 1 low)
+    assert r0["success_per_tier"] == r1["success_per_tier"] == [6, 2]
     assert r0["shard"] == (0, 4) and r1["shard"] == (4, 8)
     assert m0.isfinite().all()
