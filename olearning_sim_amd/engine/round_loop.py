@@ -300,7 +300,20 @@ class LogicalEngine:
         op = self.script_ops.get(name)
         if op is None:
             raise RuntimeError(f"script operator {name!r} not staged")
+        flow_id = None
+        if self.deviceflow is not None and self.job.behavior_strategy:
+            flow_id = self.deviceflow.notify_start(
+                self.job.task_id, name, round_idx, "logical_simulation",
+                strategy=self.job.behavior_strategy)
         res = op.run_round(round_idx)
+        if flow_id is not None:
+            self.deviceflow.publish(flow_id, "logical_simulation",
+                                    payload={"round": round_idx,
+                                             "success": res["success"],
+                                             "failed": res["failed"]})
+            self.deviceflow.drain_inbound()
+            self.deviceflow.notify_complete(self.job.task_id, name,
+                                            round_idx, "logical_simulation")
         succ_t, fail_t = per_tier_counts(res["failed_ranges"],
                                          self.tier_bounds)
         self.success_total += res["success"]

@@ -125,3 +125,41 @@ def test_script_operator_failure_fails_round(op_dir):
     st = wait_terminal(mgr, "t_sfail")
     assert st.value == "FAILED"
     mgr.shutdown()
+
+
+def test_script_operator_drives_deviceflow(op_dir, tmp_path):
+    """Gradient-house lifecycle wraps script operators too: NotifyStart,
+    a round summary through the message plane, NotifyComplete."""
+    import time as _time
+    from olearning_sim_amd.deviceflow.service import DeviceFlowService
+    from olearning_sim_amd.engine import EngineJob, LogicalEngine
+    from olearning_sim_amd.engine.script_op import ScriptOperator
+    strategy = json.dumps({"real_time_dispatch": {
+        "use_strategy": True, "dispatch_batch_sizes": [1]}})
+    svc = DeviceFlowService(time_scale=0.0, seed=1)
+    svc.register_task("t_sdf", ["logical_simulation"])
+    job = EngineJob(task_id="t_sdf", model_name="mlp",
+                    model_kwargs={"in_features": 16, "hidden": 8,
+                                  "num_classes": 4},
+                    clients=4, rounds=2, local_steps=1, batch_size=2,
+                    lr=0.1, device="cpu", dtype="float32", num_classes=4,
+                    dynamic_num=4, behavior_strategy=strategy,
+                    operators=[("user_train", "script")])
+    sop = ScriptOperator(name="user_train", staged_dir=str(op_dir),
+                         entry_file="train.py", operator_params="{}",
+                         task_id="t_sdf", work_dir=str(tmp_path / "w"),
+                         clients=4, shards=1)
+    eng = LogicalEngine(job, script_ops={"user_train": sop},
+                        deviceflow=svc, behavior=lambda r, c: (
+                            __import__("torch").zeros(c, dtype=bool),
+                            __import__("torch").zeros(c, dtype=bool)))
+    eng.run()
+    t0 = _time.time()
+    while _time.time() - t0 < 10 and not svc.check_dispatch_finished("t_sdf"):
+        _time.sleep(0.01)
+    assert svc.check_dispatch_finished("t_sdf")
+    msgs = svc.outbound.drain()
+    assert len(msgs) == 2                       # one summary per round
+    assert {m.payload["round"] for m in msgs} == {0, 1}
+    assert all(m.payload["success"] in (3, 4) for m in msgs)
+    svc.shutdown()
