@@ -305,7 +305,18 @@ class LogicalEngine:
             flow_id = self.deviceflow.notify_start(
                 self.job.task_id, name, round_idx, "logical_simulation",
                 strategy=self.job.behavior_strategy)
-        res = op.run_round(round_idx)
+        model_path = None
+        if round_idx > 0 and self.job.checkpoint_dir \
+                and self.job.save_every_round:
+            import os
+            from .checkpoint import checkpoint_name
+            cand = os.path.join(
+                self.job.checkpoint_dir,
+                checkpoint_name(self.job.task_id, round_idx - 1,
+                                self.job.model_update_style))
+            if os.path.exists(cand):
+                model_path = cand
+        res = op.run_round(round_idx, model_path=model_path)
         if flow_id is not None:
             self.deviceflow.publish(flow_id, "logical_simulation",
                                     payload={"round": round_idx,

@@ -56,10 +56,17 @@ class ScriptOperator:
         return ranges
 
     def _params_for(self, round_idx: int, shard: int,
-                    lo: int, hi: int) -> Dict[str, Any]:
+                    lo: int, hi: int,
+                    model_path: Optional[str] = None) -> Dict[str, Any]:
         save_dir = os.path.join(self.work_dir, f"round_{round_idx}",
                                 f"shard_{shard}")
         os.makedirs(save_dir, exist_ok=True)
+        model = dict(self.model_info)
+        if model_path:
+            # round r>0 weights under the templated model_update_style
+            # name (reference download_model_files,
+            # utils_run_task.py:327-397)
+            model["current_model_path"] = model_path
         return {
             "task_id": self.task_id,
             "current_round": round_idx,
@@ -67,7 +74,7 @@ class ScriptOperator:
             "operator": {
                 "name": self.name,
                 "use_data": bool(self.data_info),
-                "model": dict(self.model_info),
+                "model": model,
                 "operator_params": self.operator_params,
             },
             "actor_save_dir": save_dir,
@@ -77,10 +84,10 @@ class ScriptOperator:
         }
 
     # -- execution --------------------------------------------------------
-    def _run_shard(self, round_idx: int, shard: int,
-                   lo: int, hi: int) -> Tuple[int, int, int]:
+    def _run_shard(self, round_idx: int, shard: int, lo: int, hi: int,
+                   model_path: Optional[str] = None) -> Tuple[int, int, int]:
         """Returns (lo, hi, success_count)."""
-        params = self._params_for(round_idx, shard, lo, hi)
+        params = self._params_for(round_idx, shard, lo, hi, model_path)
         cmd = [sys.executable, os.path.join(self.staged_dir, self.entry_file),
                "--params", json.dumps(params)]
         try:
@@ -100,14 +107,15 @@ class ScriptOperator:
                 pass
         return lo, hi, (hi - lo) if rc == 0 else 0
 
-    def run_round(self, round_idx: int) -> Dict[str, Any]:
+    def run_round(self, round_idx: int,
+                  model_path: Optional[str] = None) -> Dict[str, Any]:
         """Run every shard; failed devices are attributed to the TAIL of
         their shard's client range so per-tier accounting stays exact."""
         ranges = self.shard_ranges()
         with ThreadPoolExecutor(
                 max_workers=min(self.max_workers, len(ranges))) as pool:
             results = list(pool.map(
-                lambda args: self._run_shard(round_idx, *args),
+                lambda args: self._run_shard(round_idx, *args, model_path),
                 [(s, lo, hi) for s, (lo, hi) in enumerate(ranges)]))
         success = sum(r[2] for r in results)
         failed = self.clients - success

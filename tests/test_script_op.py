@@ -163,3 +163,43 @@ def test_script_operator_drives_deviceflow(op_dir, tmp_path):
     assert {m.payload["round"] for m in msgs} == {0, 1}
     assert all(m.payload["success"] in (3, 4) for m in msgs)
     svc.shutdown()
+
+
+def test_script_operator_receives_model_path(tmp_path):
+    """Round r>0 passes the previous round's templated checkpoint path
+    to the script (reference download_model_files semantics)."""
+    import textwrap as _tw
+    d = tmp_path / "mp_op"
+    d.mkdir()
+    (d / "train.py").write_text(_tw.dedent("""
+        import json, os, sys
+        p = json.loads(sys.argv[sys.argv.index("--params") + 1])
+        out = {"success": p["actor_simulation_num"], "failed": 0,
+               "model_path": p["operator"]["model"].get("current_model_path")}
+        with open(os.path.join(p["actor_save_dir"], "result.json"), "w") as f:
+            json.dump(out, f)
+    """))
+    from olearning_sim_amd.engine import EngineJob, LogicalEngine
+    from olearning_sim_amd.engine.script_op import ScriptOperator
+    ckpt = str(tmp_path / "ckpt")
+    job = EngineJob(task_id="t_mp", model_name="mlp",
+                    model_kwargs={"in_features": 16, "hidden": 8,
+                                  "num_classes": 4},
+                    clients=4, rounds=2, local_steps=1, batch_size=2,
+                    lr=0.1, device="cpu", dtype="float32", num_classes=4,
+                    dynamic_num=4, checkpoint_dir=ckpt,
+                    save_every_round=True,
+                    operators=[("user_train", "script"),
+                               ("save", "checkpoint")])
+    wdir = tmp_path / "w"
+    sop = ScriptOperator(name="user_train", staged_dir=str(d),
+                         entry_file="train.py", operator_params="{}",
+                         task_id="t_mp", work_dir=str(wdir),
+                         clients=4, shards=1)
+    LogicalEngine(job, script_ops={"user_train": sop}).run()
+    r0 = json.loads((wdir / "round_0" / "shard_0" / "result.json").read_text())
+    r1 = json.loads((wdir / "round_1" / "shard_0" / "result.json").read_text())
+    assert r0["model_path"] is None
+    assert r1["model_path"] and r1["model_path"].endswith(
+        "t_mp_0_result_model.safetensors")
+    assert os.path.exists(r1["model_path"])
