@@ -317,3 +317,17 @@ def test_multi_data_task_reports_each_data():
     assert entries["data_0"]["success_num"] == [6]
     assert entries["data_1"]["success_num"] == [4]
     mgr.shutdown()
+
+
+def test_interrupt_overdue_running_task():
+    """The interrupt watchdog also stops tasks running past
+    interrupt_running_time (reference task_manager.py:1150-1200)."""
+    mgr = make_manager()
+    mgr.timers["interrupt_running_time"] = 0.0
+    mgr.submit_task(task_json(task_id="t_long", rounds=50))
+    assert mgr.step_schedule() == "t_long"
+    time.sleep(0.05)
+    assert "t_long" in mgr.step_interrupt()
+    st = wait_terminal(mgr, "t_long")
+    assert st in (TaskStatus.STOPPED, TaskStatus.FAILED)
+    mgr.shutdown()
