@@ -110,6 +110,42 @@ def cmd_queue(args) -> int:
     return 0
 
 
+def cmd_result(args) -> int:
+    if args.server:
+        print(json.dumps(_http(args, "GET",
+                               f"/taskmgr/getTaskResult/{args.task_id}")))
+        return 0
+    sess = _session(args)
+    row = sess.task_mgr.table.get_row(args.task_id)
+    if row is None:
+        print(json.dumps({"error": "task not found"}))
+        sess.shutdown()
+        return 1
+    print(json.dumps({
+        "task_id": args.task_id,
+        "task_status": sess.task_mgr.get_task_status(args.task_id).value,
+        "logical_round": row.get("logical_round"),
+        "logical_operator": row.get("logical_operator"),
+        "logical_result": json.loads(row["logical_result"])
+        if row.get("logical_result") else None,
+        "device_result": json.loads(row["device_result"])
+        if row.get("device_result") else None,
+    }))
+    sess.shutdown()
+    return 0
+
+
+def cmd_perf(args) -> int:
+    if args.server:
+        print(json.dumps(_http(args, "GET",
+                               f"/performancemgr/summary/{args.task_id}")))
+        return 0
+    sess = _session(args)
+    print(json.dumps(sess.performance_mgr.summary(args.task_id)))
+    sess.shutdown()
+    return 0
+
+
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser(prog="olearning_sim_amd")
     ap.add_argument("--server", default="",
@@ -137,6 +173,14 @@ def main(argv=None) -> int:
 
     p = sub.add_parser("queue")
     p.set_defaults(fn=cmd_queue)
+
+    p = sub.add_parser("result")
+    p.add_argument("task_id")
+    p.set_defaults(fn=cmd_result)
+
+    p = sub.add_parser("perf")
+    p.add_argument("task_id")
+    p.set_defaults(fn=cmd_perf)
 
     args = ap.parse_args(argv)
     return args.fn(args)

@@ -331,3 +331,23 @@ def test_interrupt_overdue_running_task():
     st = wait_terminal(mgr, "t_long")
     assert st in (TaskStatus.STOPPED, TaskStatus.FAILED)
     mgr.shutdown()
+
+
+def test_engine_crash_marks_task_failed():
+    """A crash inside the engine (reference: Ray job failure) fails the
+    task instead of wedging it RUNNING."""
+    raw = json.loads(task_json(task_id="t_crash"))
+    op = raw["operatorflow"]["operators"][0]
+    params = json.loads(op["logical_simulation"]["operator_params"])
+    params["model"] = "no_such_model"
+    op["logical_simulation"]["operator_params"] = json.dumps(params)
+    mgr = make_manager()
+    ok, msg = mgr.submit_task(json.dumps(raw))
+    assert ok, msg
+    assert mgr.step_schedule() == "t_crash"
+    st = wait_terminal(mgr, "t_crash")
+    assert st == TaskStatus.FAILED
+    # release step frees the frozen resources
+    mgr.step_release()
+    assert mgr.table.get_item_value("t_crash", "resource_occupied") == 0
+    mgr.shutdown()

@@ -384,3 +384,32 @@ def test_cli_submit_wait_and_status(tmp_path):
         cwd=repo, capture_output=True, text=True, timeout=60)
     assert json.loads(out2.stdout.strip().splitlines()[-1])[
         "task_status"] == "SUCCEEDED"
+
+
+@pytest.mark.timeout(180)
+def test_cli_result_and_perf(tmp_path):
+    import subprocess, sys
+    from test_manager import task_json
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    tf = tmp_path / "task.json"
+    tf.write_text(task_json(task_id="t_cli_r"))
+    data_dir = str(tmp_path / "data")
+    out = subprocess.run(
+        [sys.executable, "-m", "olearning_sim_amd",
+         "--data-dir", data_dir, "submit", str(tf), "--wait"],
+        cwd=repo, capture_output=True, text=True, timeout=150)
+    assert out.returncode == 0, out.stderr[-2000:]
+    res = subprocess.run(
+        [sys.executable, "-m", "olearning_sim_amd",
+         "--data-dir", data_dir, "result", "t_cli_r"],
+        cwd=repo, capture_output=True, text=True, timeout=60)
+    d = json.loads(res.stdout.strip().splitlines()[-1])
+    assert d["task_status"] == "SUCCEEDED"
+    assert d["logical_result"]["logical_result"][0]["simulation_target"][
+        "success_num"]
+    perf = subprocess.run(
+        [sys.executable, "-m", "olearning_sim_amd",
+         "--data-dir", data_dir, "perf", "t_cli_r"],
+        cwd=repo, capture_output=True, text=True, timeout=60)
+    assert perf.returncode == 0, perf.stderr[-2000:]
+    json.loads(perf.stdout.strip().splitlines()[-1])
