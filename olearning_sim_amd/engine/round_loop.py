@@ -95,7 +95,6 @@ class LogicalEngine:
             total_timeout=job.flow_total_timeout,
             work_dir=job.flow_work_dir or job.checkpoint_dir or ".")
         self._delta = self.master.zeros_like_flat()
-        self._scalar = torch.zeros(3, dtype=torch.float64, device=self.device)
         # totals across rounds (reference logical_result accounting)
         self.success_total = 0
         self.failed_total = 0
