@@ -180,3 +180,32 @@ def test_groupnorm_fused_matches_reference(shape, relu, res):
     torch.testing.assert_close(bf.grad.float(), br.grad, atol=2e-1, rtol=5e-2)
     if res:
         torch.testing.assert_close(rf.grad.float(), rr.grad, atol=1e-1, rtol=1e-1)
+
+
+def test_engine_per_tier_accounting_gpu():
+    """Per-(data x tier) segment accounting with GPU behaviour sampling:
+    vectors partition the cohort and feed the result rows."""
+    import json as _json
+    from olearning_sim_amd.engine import EngineJob, LogicalEngine
+    strategy = _json.dumps({"offline_simulation": {
+        "offline_probability": 0.5}})
+    rows = []
+    job = EngineJob(task_id="t_tier", model_name="mlp",
+                    model_kwargs={"in_features": 64, "hidden": 32,
+                                  "num_classes": 10},
+                    clients=32, rounds=2, local_steps=1, batch_size=8,
+                    lr=0.1, device="cuda:0", dtype="bfloat16",
+                    num_classes=10, seed=9, dynamic_num=100,
+                    behavior_strategy=strategy,
+                    tier_counts=[("high", 20), ("low", 12)],
+                    dynamic_nums=[100, 100])
+    eng = LogicalEngine(job, result_sink=rows.append)
+    out = eng.run()
+    assert out["rounds"] == 2
+    for rec in out["records"]:
+        s, f = rec["success_per_tier"], rec["failed_per_tier"]
+        assert s[0] + f[0] == 20 and s[1] + f[1] == 12
+        assert rec["success"] == sum(s) and rec["failed"] == sum(f)
+    tgt = rows[0]["logical_result"]["logical_result"][0]["simulation_target"]
+    assert tgt["devices"] == ["high", "low"]
+    assert eng.master.flat.isfinite().all()
