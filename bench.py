@@ -103,6 +103,9 @@ def main() -> int:
     ap.add_argument("--chunk", type=int, default=0)
     ap.add_argument("--local-steps", type=int, default=0)
     ap.add_argument("--batch", type=int, default=0)
+    ap.add_argument("--trace", default="",
+                    help="write a torch.profiler chrome trace of the timed "
+                         "steps to this path (off by default; adds overhead)")
     args = ap.parse_args()
 
     preset = dict(PRESETS[args.preset])
@@ -143,12 +146,25 @@ def main() -> int:
     pdist.barrier()
     sync()
 
+    profiler = None
+    if args.trace and ctx.rank == 0:
+        from torch.profiler import profile, ProfilerActivity
+        acts = [ProfilerActivity.CPU]
+        if use_gpu:
+            acts.append(ProfilerActivity.CUDA)
+        profiler = profile(activities=acts)
+        profiler.__enter__()
+
     t0 = time.perf_counter()
     for r in range(args.warmup, args.warmup + args.steps):
         eng.run_round(r)
     pdist.barrier()
     sync()
     elapsed = time.perf_counter() - t0
+
+    if profiler is not None:
+        profiler.__exit__(None, None, None)
+        profiler.export_chrome_trace(args.trace)
 
     # MAX over ranks
     t = torch.tensor([elapsed], dtype=torch.float64,

@@ -60,3 +60,18 @@ def test_bench_two_rank_torchrun_contract():
     json_lines = [ln for ln in out.stdout.splitlines()
                   if ln.strip().startswith("{")]
     assert len(json_lines) == 1
+
+
+def test_bench_trace_flag_writes_chrome_trace(tmp_path):
+    import subprocess, sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    trace = tmp_path / "trace.json"
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--preset", "mlp-cpu",
+         "--steps", "2", "--warmup", "1", "--trace", str(trace)],
+        cwd=repo, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    d = json.loads(out.stdout.strip().splitlines()[-1])
+    assert d["steps"] == 2
+    tr = json.loads(trace.read_text())
+    assert len(tr["traceEvents"]) > 0
