@@ -146,3 +146,34 @@ def test_lstm_engine_round_cpu():
     assert eng.master.flat.isfinite().all()
     losses = [r["loss"] for r in rows if r["loss"] is not None]
     assert losses[-1] < losses[0]
+
+
+def test_lstm_chunked_equals_unchunked():
+    from olearning_sim_amd.engine import EngineJob, LogicalEngine
+    def job(chunk):
+        return EngineJob(task_id="lc", model_name="lstm",
+                         model_kwargs={"vocab_size": 20, "embed": 4,
+                                       "hidden": 8, "layers": 1,
+                                       "seq_len": 5},
+                         clients=6, rounds=1, local_steps=1, batch_size=2,
+                         lr=0.2, device="cpu", dtype="float32",
+                         vocab_size=20, seq_len=5, seed=13,
+                         chunk_clients=chunk)
+    e1, e2 = LogicalEngine(job(2)), LogicalEngine(job(6))
+    e1.run_round(0)
+    e2.run_round(0)
+    torch.testing.assert_close(e1.master.flat, e2.master.flat, atol=1e-5,
+                               rtol=1e-5)
+
+
+def test_lstm_evaluate_global_runs():
+    from olearning_sim_amd.engine import EngineJob, LogicalEngine
+    job = EngineJob(task_id="le", model_name="lstm",
+                    model_kwargs={"vocab_size": 20, "embed": 4,
+                                  "hidden": 8, "layers": 1, "seq_len": 5},
+                    clients=4, rounds=1, local_steps=1, batch_size=2,
+                    lr=0.2, device="cpu", dtype="float32",
+                    vocab_size=20, seq_len=5, seed=13)
+    out = LogicalEngine(job).evaluate_global(0)
+    assert 0.0 <= out["eval_acc"] <= 1.0
+    assert out["eval_loss"] > 0
