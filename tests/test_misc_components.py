@@ -413,3 +413,27 @@ def test_cli_result_and_perf(tmp_path):
         cwd=repo, capture_output=True, text=True, timeout=60)
     assert perf.returncode == 0, perf.stderr[-2000:]
     json.loads(perf.stdout.strip().splitlines()[-1])
+
+
+def test_sqlite_repo_crud_cycle(tmp_path):
+    """Direct CRUD exercise of the generic table repo (the reference's
+    SqlDataBase inline self-test, repo_utils.py:346-401)."""
+    from olearning_sim_amd.utils.sqlite_repo import SqlTableRepo
+    repo = SqlTableRepo(str(tmp_path / "t.sqlite"), "things",
+                        {"key": "TEXT", "num": "INTEGER", "blob": "TEXT"},
+                        primary_key="key")
+    assert repo.add_item({"key": "a", "num": 1, "blob": "x"})
+    assert not repo.add_item({"key": "a", "num": 2})   # duplicate pk
+    assert repo.has_item("key", "a")
+    assert repo.get_item_value("key", "a", "num") == 1
+    assert repo.set_item_value("key", "a", "num", 5)
+    assert repo.get_item_value("key", "a", "num") == 5
+    repo.upsert_item("key", {"key": "b", "num": 7})
+    assert repo.count() == 2
+    assert sorted(repo.get_column_not_none("key")) == ["a", "b"]
+    assert repo.get_rows_where({"num": 7})[0]["key"] == "b"
+    assert repo.delete_item("key", "a")
+    assert not repo.has_item("key", "a")
+    repo.clear()
+    assert repo.count() == 0
+    repo.close()
