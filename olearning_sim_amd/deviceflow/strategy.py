@@ -53,6 +53,13 @@ def _eval_rate(func_string: str, t: float) -> float:
     return float(eval(func_string, dict(_SAFE_EVAL_GLOBALS), {"t": t}))
 
 
+
+def _dct(v):
+    """None/garbage-tolerant dict access: user strategy JSON may carry
+    null or mistyped sections (the public analysers never crash —
+    fuzz-tested)."""
+    return v if isinstance(v, dict) else {}
+
 class Strategy:
     """Pure schedule analysis; no transport attached."""
 
@@ -60,14 +67,14 @@ class Strategy:
     @staticmethod
     def check_real_time_dispatch(strategy: str) -> bool:
         s = json.loads(strategy)
-        return bool(s.get("real_time_dispatch", {}).get("use_strategy", False))
+        return bool(_dct(s.get("real_time_dispatch")).get("use_strategy", False))
 
     @staticmethod
     def real_time_strategy_analysis(strategy: str) -> Tuple[List[int], float]:
         s = json.loads(strategy)
-        rt = s.get("real_time_dispatch", {})
+        rt = _dct(s.get("real_time_dispatch"))
         batch_sizes = rt.get("dispatch_batch_sizes", [])
-        drop_p = rt.get("drop_simulation", {}).get("drop_probability", 0)
+        drop_p = _dct(rt.get("drop_simulation")).get("drop_probability", 0)
         return batch_sizes, drop_p
 
     # ----------------------------------------------------------------- flow
@@ -77,21 +84,21 @@ class Strategy:
                                now: Optional[datetime] = None
                                ) -> Tuple[List[float], List[int], List[List[int]]]:
         s = json.loads(strategy)
-        flow = s.get("flow_dispatch", {})
+        flow = _dct(s.get("flow_dispatch"))
         if not flow.get("use_strategy", False):
             return [], [], []
         total = flow.get("total_dispatch_amount", 0)
         if total <= 0:
             return [], [], []
-        use_timing = flow.get("specific_timing", {}).get("use", False)
-        use_interval = flow.get("specific_interval", {}).get("use", False)
+        use_timing = _dct(flow.get("specific_timing")).get("use", False)
+        use_interval = _dct(flow.get("specific_interval")).get("use", False)
         if use_timing == use_interval:          # both or neither -> invalid
             return [], [], []
         inst = cls()
         if use_timing:
-            return inst._specific_timing(flow.get("specific_timing", {}),
+            return inst._specific_timing(_dct(flow.get("specific_timing")),
                                          flow_id, rng, now)
-        return inst._specific_interval(total, flow.get("specific_interval", {}),
+        return inst._specific_interval(total, _dct(flow.get("specific_interval")),
                                        flow_id, rng, now)
 
     # -- explicit send times/amounts ------------------------------------
@@ -114,7 +121,7 @@ class Strategy:
         if len(timings) != len(amounts) or len(timings) == 0:
             return [], [], []
 
-        drop_spec = spec.get("drop_simulation", {})
+        drop_spec = _dct(spec.get("drop_simulation"))
         if drop_spec:
             if len(drop_spec) != 1:
                 return [], [], []
@@ -157,10 +164,10 @@ class Strategy:
             except Exception:
                 return [], [], []
 
-        rules = spec.get("dispatch_rules", {})
+        rules = _dct(spec.get("dispatch_rules"))
         domains = rules.get("domains", [])
         functions = rules.get("functions", [])
-        drop_spec = spec.get("drop_simulation", {})
+        drop_spec = _dct(spec.get("drop_simulation"))
         if len(intervals) != len(domains) or len(domains) != len(functions):
             return [], [], []
         if len(intervals) == 0:

@@ -53,9 +53,12 @@ class SqlTableRepo:
         ph = ", ".join("?" for _ in names)
         collist = ", ".join(f'"{n}"' for n in names)
         with self._lock:
-            self._conn.execute(
-                f'INSERT INTO "{self._table}" ({collist}) VALUES ({ph})',
-                [item[n] for n in names])
+            try:
+                self._conn.execute(
+                    f'INSERT INTO "{self._table}" ({collist}) VALUES ({ph})',
+                    [item[n] for n in names])
+            except sqlite3.IntegrityError:
+                return False    # duplicate primary key
             self._conn.commit()
         return True
 
