@@ -215,3 +215,25 @@ def test_status_fusion_never_crashes_and_is_sound(succ, fail, nums, dyn, rnd):
         assert rnd >= 2
         padded = list(succ) + [0] * max(0, k - len(succ))
         assert all(padded[i] >= nums[i] - dyn[i] for i in range(k))
+
+
+@settings(max_examples=40, deadline=None)
+@given(clients=st.integers(2, 20), batch=st.integers(1, 6),
+       rnd=st.integers(0, 3), step=st.integers(0, 2),
+       vocab=st.sampled_from([0, 17]))
+def test_synthetic_data_chunk_invariance(clients, batch, rnd, step, vocab):
+    """A client's batch is identical whether fetched alone or inside
+    any chunk — the invariant that makes chunked == unchunked training
+    exact (engine/data.py per-(seed,round,step) generators)."""
+    import torch
+    from olearning_sim_amd.engine.data import SyntheticFederatedData
+    data = SyntheticFederatedData(
+        clients=clients, num_classes=5, input_shape=(12,),
+        dirichlet_alpha=0.5, shard_size=8, seed=3, device="cpu",
+        vocab_size=vocab, seq_len=4 if vocab else 0)
+    ids = torch.arange(clients)
+    x_all, y_all = data.batch(ids, rnd, step, batch, torch.float32)
+    c = clients // 2
+    x_one, y_one = data.batch(ids[c:c + 1], rnd, step, batch, torch.float32)
+    torch.testing.assert_close(x_all[c:c + 1], x_one)
+    torch.testing.assert_close(y_all[c:c + 1], y_one)
