@@ -237,3 +237,37 @@ def test_synthetic_data_chunk_invariance(clients, batch, rnd, step, vocab):
     x_one, y_one = data.batch(ids[c:c + 1], rnd, step, batch, torch.float32)
     torch.testing.assert_close(x_all[c:c + 1], x_one)
     torch.testing.assert_close(y_all[c:c + 1], y_one)
+
+
+@settings(max_examples=40, deadline=None)
+@given(rounds=st.lists(st.integers(0, 40), min_size=1, max_size=6,
+                       unique=True),
+       prefix=st.text(alphabet=st.characters(
+           whitelist_categories=("Ll", "Nd")), min_size=1, max_size=8))
+def test_checkpoint_latest_round_roundtrip(tmp_path_factory, rounds, prefix):
+    """save_checkpoint -> latest_round finds the max round for any
+    templated style and task id."""
+    import torch
+    from olearning_sim_amd.engine.checkpoint import (latest_round,
+                                                     save_checkpoint)
+    d = str(tmp_path_factory.mktemp("ck"))
+    style = prefix + "_{task_id}_{current_round}_result_model.safetensors"
+    for r in rounds:
+        save_checkpoint(d, "tk", r, {"w": torch.zeros(2)}, style)
+    assert latest_round(d, "tk", style) == max(rounds)
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.data())
+def test_schema_roundtrip_idempotent(data):
+    """taskconfig2json(json2taskconfig(x)) is a fixed point: a second
+    round-trip yields byte-identical JSON."""
+    import copy
+    from test_schema import EXAMPLE
+    raw = copy.deepcopy(EXAMPLE)
+    raw["target"]["priority"] = data.draw(st.integers(0, 10))
+    raw["task_id"] = "t_" + data.draw(st.text(
+        alphabet="abcdefghij0123456789", min_size=1, max_size=10))
+    once = taskconfig2json(json2taskconfig(json.dumps(raw)))
+    twice = taskconfig2json(json2taskconfig(once))
+    assert once == twice
