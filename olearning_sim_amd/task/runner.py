@@ -136,6 +136,10 @@ def engine_job_from_task(task: TaskConfig, allocations: List[DataAllocation],
         vocab_size=params.get("vocab_size", 0),
         seq_len=params.get("seq_len", 0),
         behavior_strategy=behavior,
+        # crash recovery of an interrupted task is opt-in via the
+        # operator_params JSON (resume: true) — a fresh submission with
+        # the same task_id must NOT silently load stale artifacts
+        resume=bool(params.get("resume", False)),
         checkpoint_dir=checkpoint_dir,
         save_every_round=bool(train_op and train_op.model.use_model
                               and checkpoint_dir),

@@ -351,3 +351,53 @@ def test_engine_crash_marks_task_failed():
     mgr.step_release()
     assert mgr.table.get_item_value("t_crash", "resource_occupied") == 0
     mgr.shutdown()
+
+
+def test_resubmit_with_resume_continues_from_checkpoint(tmp_path):
+    """operator_params resume:true picks up the newest templated
+    checkpoint of the same task id (reference: actors download round
+    r-1 weights; task state persists across manager restarts)."""
+    from olearning_sim_amd.resource.manager import ResourceManager
+    from olearning_sim_amd.task.runner import TaskRunner
+    from olearning_sim_amd.task.table import TaskTableRepo
+    from olearning_sim_amd.task.manager import TaskManager
+    ckpt = str(tmp_path / "ck")
+    table = TaskTableRepo(":memory:")
+    res = ResourceManager(":memory:", totals={"cpu": 8.0, "mem": 64.0,
+                                              "gpu": 0, "hbm_gb": 0})
+    mgr = TaskManager(table=table, resource_mgr=res,
+                      runner=TaskRunner(table, device="cpu",
+                                        checkpoint_dir=ckpt))
+    raw = json.loads(task_json(task_id="t_res2", rounds=2))
+    op = raw["operatorflow"]["operators"][0]
+    op["model"] = {"use_model": True, "model_for_train": True,
+                   "model_transfer_type": "FILE", "model_path": "m",
+                   "model_update_style":
+                       "{task_id}_{current_round}_result_model.safetensors"}
+    ok, msg = mgr.submit_task(json.dumps(raw))
+    assert ok, msg
+    assert mgr.step_schedule() == "t_res2"
+    assert wait_terminal(mgr, "t_res2") == TaskStatus.SUCCEEDED
+    mgr.step_release()
+    import os
+    assert os.path.exists(os.path.join(
+        ckpt, "t_res2_1_result_model.safetensors"))
+
+    # resubmit for 4 rounds with resume: engine continues at round 2
+    round1 = os.path.join(ckpt, "t_res2_1_result_model.safetensors")
+    mtime_before = os.path.getmtime(round1)
+    raw["operatorflow"]["flow_setting"]["round"] = 4
+    p = json.loads(op["logical_simulation"]["operator_params"])
+    p["resume"] = True
+    op["logical_simulation"]["operator_params"] = json.dumps(p)
+    ok, msg = mgr.submit_task(json.dumps(raw))
+    assert ok, msg
+    assert mgr.step_schedule() == "t_res2"
+    assert wait_terminal(mgr, "t_res2") == TaskStatus.SUCCEEDED
+    # final cursor reaches max_round and rounds 2..3 were saved
+    assert table.get_item_value("t_res2", "logical_round") == 4
+    assert os.path.exists(os.path.join(
+        ckpt, "t_res2_3_result_model.safetensors"))
+    # proof of resume: round 1's artifact was NOT rewritten
+    assert os.path.getmtime(round1) == mtime_before
+    mgr.shutdown()
