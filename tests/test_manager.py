@@ -401,3 +401,36 @@ def test_resubmit_with_resume_continues_from_checkpoint(tmp_path):
     # proof of resume: round 1's artifact was NOT rewritten
     assert os.path.getmtime(round1) == mtime_before
     mgr.shutdown()
+
+
+@pytest.mark.timeout(120)
+def test_threaded_manager_runs_concurrent_tasks():
+    """With the real daemon loops on (fast timers), several tasks
+    schedule, run, and release concurrently without manual stepping —
+    the reference's 3-thread operation."""
+    table = TaskTableRepo(":memory:")
+    res = ResourceManager(":memory:", totals={"cpu": 4.0, "mem": 64.0,
+                                              "gpu": 0, "hbm_gb": 0})
+    mgr = TaskManager(table=table, resource_mgr=res,
+                      runner=TaskRunner(table, device="cpu"),
+                      timers={"scheduler_sleep_time": 0.05,
+                              "release_sleep_time": 0.05,
+                              "interrupt_sleep_time": 60.0},
+                      auto_start=True)
+    try:
+        for tid in ("th1", "th2", "th3"):
+            ok, msg = mgr.submit_task(task_json(task_id=tid, rounds=1))
+            assert ok, msg
+        t0 = time.time()
+        while time.time() - t0 < 60:
+            done = {tid for tid in ("th1", "th2", "th3")
+                    if mgr.get_task_status(tid) == TaskStatus.SUCCEEDED
+                    and table.get_item_value(tid, "resource_occupied") == 0}
+            if len(done) == 3:
+                break
+            time.sleep(0.05)
+        assert len(done) == 3, {
+            tid: mgr.get_task_status(tid).value for tid in
+            ("th1", "th2", "th3")}
+    finally:
+        mgr.shutdown()
