@@ -47,6 +47,17 @@ def main(argv=None) -> int:
     lo, hi = shard_clients(total_clients, ctx.rank, max(1, ctx.world_size))
     spec = dict(spec)
     spec["clients"] = max(1, hi - lo)
+    # tier/data segments in the spec describe the TOTAL population; after
+    # sharding they no longer sum to this rank's clients, so fall back to
+    # single-segment accounting rather than miscount
+    dropped = False
+    for key in ("tier_counts", "data_segments"):
+        seg = spec.get(key)
+        if seg and sum(int(s[-1]) for s in seg) != spec["clients"]:
+            spec.pop(key, None)
+            dropped = True
+    if dropped:
+        spec.pop("dynamic_nums", None)
     known = {f.name for f in EngineJob.__dataclass_fields__.values()}
     job = EngineJob(**{k: v for k, v in spec.items() if k in known})
     job.device = ctx.device
