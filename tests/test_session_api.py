@@ -262,3 +262,12 @@ def test_deviceflow_data_plane_over_http(client):
         got += client.get("/deviceflow/outbound").json()["messages"]
         _time.sleep(0.01)
     assert sorted(m["payload"]["i"] for m in got) == [0, 1, 2]
+
+
+def test_svc_codes_1_and_3(tmp_path):
+    s = SimulatorSession(svc=1, data_dir=str(tmp_path / "c"))
+    assert s.task_mgr is not None and s.performance_mgr is None
+    s.shutdown()
+    s = SimulatorSession(svc=3, data_dir=str(tmp_path / "d"))
+    assert s.deviceflow is not None and s.task_mgr is None
+    s.shutdown()
