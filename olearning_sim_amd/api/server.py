@@ -192,6 +192,18 @@ def build_app(session) -> FastAPI:
                  "payload": m.payload}
                 for m in df.outbound.drain(max_items)]}
 
+    # -- phone-side surface (phoneMgr.proto analogue; the proprietary
+    # real-phone farm is simulated by the runner's device jobs) ---------
+    if session.task_mgr is not None:
+        ptm = session.task_mgr
+
+        @app.get("/phonemgr/getDeviceTaskStatus/{task_id}")
+        def get_device_task_status(task_id: str):
+            row = ptm.table.get_row(task_id)
+            if row is None:
+                return {"error": "task not found"}
+            return ptm._device_task_result(task_id, row)
+
     # -- PerformanceMgr service (performanceService.proto) ---------------
     if session.performance_mgr is not None:
         pm = session.performance_mgr

@@ -271,3 +271,15 @@ def test_svc_codes_1_and_3(tmp_path):
     s = SimulatorSession(svc=3, data_dir=str(tmp_path / "d"))
     assert s.deviceflow is not None and s.task_mgr is None
     s.shutdown()
+
+
+def test_phonemgr_device_task_status_route(client, session):
+    """The phone-side RPC surface (phoneMgr.proto analogue) reports the
+    simulated farm's DeviceTaskResult."""
+    body = {"task": json.loads(task_json(task_id="t_phone"))}
+    assert client.post("/taskmgr/submitTask", json=body).json()["is_success"]
+    # logical-only task: no device side, so the farm reports finished
+    r = client.get("/phonemgr/getDeviceTaskStatus/t_phone").json()
+    assert r["is_finished"] and r["device_result"] == []
+    assert "error" in client.get(
+        "/phonemgr/getDeviceTaskStatus/nope").json()
