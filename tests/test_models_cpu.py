@@ -177,3 +177,21 @@ def test_lstm_evaluate_global_runs():
     out = LogicalEngine(job).evaluate_global(0)
     assert 0.0 <= out["eval_acc"] <= 1.0
     assert out["eval_loss"] > 0
+
+
+def test_resnet_custom_conv_path_matches_default_cpu():
+    """forward_cbf (the hand-written-kernel path; CPU fallbacks here)
+    must equal the grouped-conv default path."""
+    m = build_model("resnet18", num_classes=10, width_mult=0.25)
+    gen = torch.Generator().manual_seed(4)
+    gp = m.init_global(generator=gen)
+    master = FlatParams(gp)
+    C, B = 2, 2
+    buf = replicate_flat(master.cast(torch.float32), C)
+    with torch.no_grad():
+        buf += 0.01 * torch.randn(buf.shape, generator=gen)
+    params = batched_views(buf.detach(), master.shapes, C)
+    x = torch.randn((C, B) + m.input_shape, generator=gen)
+    default = m.forward(params, x)
+    custom = m.forward_cbf(params, x)
+    torch.testing.assert_close(custom, default, atol=2e-4, rtol=1e-3)
