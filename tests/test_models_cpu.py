@@ -195,3 +195,26 @@ def test_resnet_custom_conv_path_matches_default_cpu():
     default = m.forward(params, x)
     custom = m.forward_cbf(params, x)
     torch.testing.assert_close(custom, default, atol=2e-4, rtol=1e-3)
+
+
+def test_resnet_custom_conv_path_backward_matches_default_cpu():
+    m = build_model("resnet18", num_classes=10, width_mult=0.25)
+    gen = torch.Generator().manual_seed(5)
+    gp = m.init_global(generator=gen)
+    master = FlatParams(gp)
+    C, B = 2, 2
+    x = torch.randn((C, B) + m.input_shape, generator=gen)
+    y = torch.randint(0, 10, (C, B), generator=gen)
+
+    def grad_of(fwd):
+        buf = replicate_flat(master.cast(torch.float32), C)
+        params = batched_views(buf, master.shapes, C)
+        logits = fwd(params, x)
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(C * B, -1), y.reshape(-1))
+        g, = torch.autograd.grad(loss, [buf])
+        return g
+
+    g_default = grad_of(m.forward)
+    g_custom = grad_of(m.forward_cbf)
+    torch.testing.assert_close(g_custom, g_default, atol=5e-4, rtol=5e-3)
