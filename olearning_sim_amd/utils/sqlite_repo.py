@@ -45,6 +45,15 @@ class SqlTableRepo:
         )
         with self._lock:
             self._conn.execute(f'CREATE TABLE IF NOT EXISTS "{table}" ({cols})')
+            # migrate pre-existing databases: a schema that gained
+            # columns since the table was created must ALTER, or every
+            # write to a new column fails ("no such column")
+            have = {r[1] for r in self._conn.execute(
+                f'PRAGMA table_info("{table}")')}
+            for name, ctype in columns.items():
+                if name not in have:
+                    self._conn.execute(
+                        f'ALTER TABLE "{table}" ADD COLUMN "{name}" {ctype}')
             self._conn.commit()
 
     # -- row ops ---------------------------------------------------------

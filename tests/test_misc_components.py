@@ -437,3 +437,24 @@ def test_sqlite_repo_crud_cycle(tmp_path):
     repo.clear()
     assert repo.count() == 0
     repo.close()
+
+
+def test_sqlite_repo_migrates_new_columns(tmp_path):
+    """Reopening a database with a schema that gained columns must
+    ALTER the table, not fail on the first write to a new column
+    (regression: pre-existing data dirs broke when the task table
+    gained the per-side params columns)."""
+    from olearning_sim_amd.utils.sqlite_repo import SqlTableRepo
+    p = str(tmp_path / "m.sqlite")
+    old = SqlTableRepo(p, "t", {"k": "TEXT", "a": "INTEGER"},
+                       primary_key="k")
+    old.add_item({"k": "x", "a": 1})
+    old.close()
+    new = SqlTableRepo(p, "t", {"k": "TEXT", "a": "INTEGER",
+                                "b": "TEXT", "c": "REAL"},
+                       primary_key="k")
+    assert new.set_item_value("k", "x", "b", "hello")
+    assert new.get_item_value("k", "x", "b") == "hello"
+    assert new.get_item_value("k", "x", "a") == 1
+    assert new.add_item({"k": "y", "a": 2, "b": "z", "c": 1.5})
+    new.close()
