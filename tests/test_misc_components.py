@@ -458,3 +458,21 @@ def test_sqlite_repo_migrates_new_columns(tmp_path):
     assert new.get_item_value("k", "x", "a") == 1
     assert new.add_item({"k": "y", "a": 2, "b": "z", "c": 1.5})
     new.close()
+
+
+@pytest.mark.timeout(300)
+def test_examples_run_hermetically(tmp_path):
+    """Both examples run green against a fresh data dir (OLSIM_CONFIG
+    points the session at tmp) — examples double as integration
+    regressions; a stale-schema data dir broke demo_task once."""
+    import subprocess, sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cfg = tmp_path / "cfg.yaml"
+    cfg.write_text(f"data_dir: {tmp_path / 'data'}\ndevice: cpu\n")
+    env = dict(os.environ, OLSIM_CONFIG=str(cfg))
+    for script in ("examples/demo_task.py",
+                   "examples/submit_script_operator.py"):
+        out = subprocess.run([sys.executable, script], cwd=repo, env=env,
+                             capture_output=True, text=True, timeout=240)
+        assert out.returncode == 0, (script, out.stderr[-2000:])
+        assert "SUCCEEDED" in out.stdout, (script, out.stdout[-2000:])
