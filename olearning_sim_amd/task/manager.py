@@ -83,6 +83,21 @@ class TaskManager:
                 except Exception:
                     self.table.set_item_value(task_id, "task_status",
                                               TaskStatus.FAILED.value)
+        # tasks RUNNING when the previous process died are orphans: the
+        # engine threads died with it and this runner knows nothing of
+        # their job ids, so they would poll as RUNNING forever.  The
+        # reference survives restarts because Ray tracks its jobs
+        # (task_manager.py:521-524); here the job IS the process, so a
+        # crashed run is a failed run.
+        for task_id in self.table.tasks_with_status(TaskStatus.RUNNING.value):
+            if task_id not in self.runner.task_jobs:
+                self.table.set_items(task_id,
+                                     task_status=TaskStatus.FAILED.value,
+                                     resource_occupied=0,
+                                     finish_task_time=time.time())
+                self.resources.release_resource(task_id)
+                self.log.error(task_id, "TaskMgr", "manager",
+                               "orphaned RUNNING task failed on restart")
         # free resources frozen by a previous process with nothing running
         for task_id in self.resources.orphaned_tasks():
             occupied = self.table.get_item_value(task_id, "resource_occupied")

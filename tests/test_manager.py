@@ -434,3 +434,35 @@ def test_threaded_manager_runs_concurrent_tasks():
             ("th1", "th2", "th3")}
     finally:
         mgr.shutdown()
+
+
+def test_orphaned_running_task_fails_on_restart(tmp_path):
+    """A task RUNNING when the process died must not poll as RUNNING
+    forever in the next process — it fails and its resources free."""
+    db = str(tmp_path / "t.sqlite")
+    rdb = str(tmp_path / "r.sqlite")
+    table = TaskTableRepo(db)
+    res = ResourceManager(rdb, totals={"cpu": 8, "mem": 64, "gpu": 0,
+                                       "hbm_gb": 0})
+    mgr1 = TaskManager(table=table, resource_mgr=res,
+                       runner=TaskRunner(table))
+    ok, _ = mgr1.submit_task(task_json(task_id="t_orph", rounds=1))
+    assert ok
+    # simulate crash mid-run: row says RUNNING + resources frozen, but
+    # the next process's runner has no such job
+    res.request_resource("t_orph", "u1", cpu=2.0)
+    table.set_items("t_orph", task_status=TaskStatus.RUNNING.value,
+                    resource_occupied=1, job_id="olsjob_dead")
+    mgr1.shutdown()
+
+    table2 = TaskTableRepo(db)
+    res2 = ResourceManager(rdb, totals={"cpu": 8, "mem": 64, "gpu": 0,
+                                        "hbm_gb": 0})
+    mgr2 = TaskManager(table=table2, resource_mgr=res2,
+                       runner=TaskRunner(table2))
+    assert mgr2.get_task_status("t_orph") == TaskStatus.FAILED
+    assert not res2.holding("t_orph")
+    # and the task can be resubmitted fresh
+    ok, msg = mgr2.submit_task(task_json(task_id="t_orph", rounds=1))
+    assert ok, msg
+    mgr2.shutdown()
