@@ -26,7 +26,7 @@ from typing import Dict, Tuple
 
 import torch
 
-from .strategy import AREA_CALCULATION_NUM
+from .strategy import AREA_CALCULATION_NUM, _dct
 
 
 def _torch_eval_rate(func_string: str, t: torch.Tensor) -> torch.Tensor:
@@ -59,16 +59,16 @@ class BehaviorSampler:
         """(forwarded_fraction, dropped_fraction_of_forwarded) for flow mode."""
         if round_idx in self._cache:
             return self._cache[round_idx]
-        flow = self.spec.get("flow_dispatch", {})
+        flow = _dct(self.spec.get("flow_dispatch"))
         total = flow.get("total_dispatch_amount", 0)
-        spec = flow.get("specific_interval", {})
+        spec = _dct(flow.get("specific_interval"))
         intervals = spec.get("intervals", [])
         if spec.get("time_type", "relative") != "relative":
             try:
                 intervals = intervals[round_idx]
             except Exception:
                 intervals = []
-        rules = spec.get("dispatch_rules", {})
+        rules = _dct(spec.get("dispatch_rules"))
         domains = rules.get("domains", [])
         functions = rules.get("functions", [])
         if not intervals or len(intervals) != len(domains) != len(functions):
@@ -96,7 +96,7 @@ class BehaviorSampler:
             return self._cache[round_idx]
         # amounts are exactly total (they always apportion fully); the
         # engine-visible knob is the drop settings
-        drop_spec = spec.get("drop_simulation", {})
+        drop_spec = _dct(spec.get("drop_simulation"))
         drop_frac = 0.0
         if "drop_probability" in drop_spec:
             probs = drop_spec["drop_probability"]
@@ -118,7 +118,7 @@ class BehaviorSampler:
         # offline_simulation: {"offline_probability": p} or
         #   {"spike_period": k, "spike_offline_fraction": f} —
         # periodic offline spikes (BASELINE config 4).
-        off = self.spec.get("offline_simulation", {})
+        off = _dct(self.spec.get("offline_simulation"))
         if off:
             p = float(off.get("offline_probability", 0.0))
             period = int(off.get("spike_period", 0))
@@ -128,17 +128,17 @@ class BehaviorSampler:
                 offline |= (torch.rand(cohort, generator=self.gen,
                                        device=self.device) < p)
 
-        rt = self.spec.get("real_time_dispatch", {})
+        rt = _dct(self.spec.get("real_time_dispatch"))
         if rt.get("use_strategy", False):
-            p = float(rt.get("drop_simulation", {}).get("drop_probability", 0))
+            p = float(_dct(rt.get("drop_simulation")).get("drop_probability", 0) or 0)
             if p > 0:
                 dropped |= (torch.rand(cohort, generator=self.gen,
                                        device=self.device) < p)
             return offline, dropped
 
-        flow = self.spec.get("flow_dispatch", {})
+        flow = _dct(self.spec.get("flow_dispatch"))
         if flow.get("use_strategy", False) and \
-                flow.get("specific_interval", {}).get("use", False):
+                _dct(flow.get("specific_interval")).get("use", False):
             fwd_frac, drop_frac = self._flow_fractions(round_idx)
             if fwd_frac < 1.0:
                 n_off = int(round((1.0 - fwd_frac) * cohort))

@@ -271,3 +271,44 @@ def test_schema_roundtrip_idempotent(data):
     once = taskconfig2json(json2taskconfig(json.dumps(raw)))
     twice = taskconfig2json(json2taskconfig(once))
     assert once == twice
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.dictionaries(
+    st.sampled_from(["flow_dispatch", "real_time_dispatch",
+                     "offline_simulation", "specific_interval",
+                     "drop_simulation", "dispatch_rules", "use_strategy",
+                     "total_dispatch_amount", "intervals", "domains",
+                     "functions", "drop_probability", "offline_probability"]),
+    json_values, max_size=5),
+    st.integers(0, 3), st.integers(1, 16))
+def test_sampler_tolerates_real_key_garbage(raw, rnd, cohort):
+    """BehaviorSampler with REAL schema keys holding garbage values
+    (null sections, mistyped lists) never crashes."""
+    from olearning_sim_amd.deviceflow.sampler import BehaviorSampler
+    try:
+        s = BehaviorSampler(json.dumps(raw), seed=1, device="cpu")
+    except (ValueError, TypeError, KeyError):
+        return
+    off, drop = s(rnd, cohort)
+    assert off.shape[0] == cohort and drop.shape[0] == cohort
+
+
+@settings(max_examples=250, deadline=None)
+@given(st.dictionaries(
+    st.sampled_from(["flow_dispatch", "real_time_dispatch", "use_strategy",
+                     "specific_timing", "specific_interval", "timings",
+                     "amounts", "intervals", "domains", "functions",
+                     "dispatch_rules", "drop_simulation", "drop_probability",
+                     "drop_amounts", "total_dispatch_amount", "time_type",
+                     "dispatch_batch_sizes", "use"]),
+    json_values, max_size=6))
+def test_strategy_tolerates_real_key_garbage(raw):
+    """Strategy analysers and the validator with REAL schema keys
+    holding garbage values never crash."""
+    text = json.dumps(raw)
+    assert isinstance(ValidateStrategy().check(text), bool)
+    out = Strategy.flow_strategy_analysis(text, "t_op_0")
+    assert isinstance(out, tuple) and len(out) == 3
+    Strategy.check_real_time_dispatch(text)
+    Strategy.real_time_strategy_analysis(text)
