@@ -312,3 +312,39 @@ def test_strategy_tolerates_real_key_garbage(raw):
     assert isinstance(out, tuple) and len(out) == 3
     Strategy.check_real_time_dispatch(text)
     Strategy.real_time_strategy_analysis(text)
+
+
+_TASK_KEYS = [
+    "user_id", "task_id", "target", "priority", "data", "name",
+    "data_path", "total_simulation", "devices", "nums", "dynamic_nums",
+    "allocation", "optimization", "logical_simulation",
+    "device_simulation", "running_response", "operatorflow",
+    "flow_setting", "round", "start", "stop", "operators", "model",
+    "use_model", "computation_unit", "setting", "num_cpus",
+    "resource_request", "num_request", "operator_params",
+    "operator_code_path", "operator_entry_file"]
+_task_vals = st.recursive(
+    st.one_of(st.none(), st.booleans(), st.integers(-100, 1000),
+              st.sampled_from(["", "x", "FILE", "high", "{}", "not json"])),
+    lambda ch: st.one_of(
+        st.lists(ch, max_size=3),
+        st.dictionaries(st.sampled_from(_TASK_KEYS), ch, max_size=5)),
+    max_leaves=20)
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.dictionaries(st.sampled_from(_TASK_KEYS), _task_vals, max_size=6))
+def test_task_parsing_tolerates_real_key_garbage(raw):
+    """Real schema keys holding null/mistyped values: parse either
+    raises cleanly (submit wraps it) or validates to a bool, and
+    anything accepted round-trips."""
+    text = json.dumps(raw)
+    try:
+        cfg = json2taskconfig(text)
+    except (TypeError, AttributeError, ValueError, KeyError):
+        return
+    v = ValidateParameters()
+    ok = v.validate_task_parameters(raw, cfg)
+    assert isinstance(ok, bool)
+    if ok:
+        json.loads(taskconfig2json(cfg))
