@@ -106,6 +106,25 @@ class BehaviorSampler:
         self._cache[round_idx] = (1.0, drop_frac)
         return self._cache[round_idx]
 
+    def _timing_drop_fraction(self, round_idx: int) -> float:
+        """Dropped / dispatched for a specific_timing schedule, from the
+        SAME analysis the dispatcher runs (strategy.py
+        _specific_timing) — flow-id round suffix selects per-round
+        timing lists."""
+        key = -1000 - round_idx
+        if key in self._cache:
+            return self._cache[key][1]
+        import random as _random
+        from .strategy import Strategy
+        _, amounts, drops = Strategy.flow_strategy_analysis(
+            json.dumps(self.spec),
+            f"{self.task_id}_{self.operator}_{round_idx}",
+            rng=_random.Random(round_idx * 9176 + 13))
+        total = sum(amounts)
+        frac = (sum(len(d) for d in drops) / total) if total else 0.0
+        self._cache[key] = (1.0, frac)
+        return frac
+
     # ------------------------------------------------------------------
     def __call__(self, round_idx: int, cohort: int
                  ) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -137,6 +156,16 @@ class BehaviorSampler:
             return offline, dropped
 
         flow = _dct(self.spec.get("flow_dispatch"))
+        if flow.get("use_strategy", False) and \
+                _dct(flow.get("specific_timing")).get("use", False):
+            # timing mode: the exact (amounts, drop-index) schedule the
+            # dispatcher would run — dropped fraction of the round's
+            # dispatch total becomes the aggregation drop mask
+            drop_frac = self._timing_drop_fraction(round_idx)
+            if drop_frac > 0:
+                dropped |= (torch.rand(cohort, generator=self.gen,
+                                       device=self.device) < drop_frac)
+            return offline, dropped
         if flow.get("use_strategy", False) and \
                 _dct(flow.get("specific_interval")).get("use", False):
             fwd_frac, drop_frac = self._flow_fractions(round_idx)

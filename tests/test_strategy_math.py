@@ -192,3 +192,36 @@ def test_sampler_torch_integration_matches_scalar_area():
     # analytic: integral of sin+1 over [0, 6.28]
     area_ref = (-m.cos(6.28) + m.cos(0.0)) + 6.28
     assert abs(area_t - area_ref) < 1e-3
+
+
+def test_sampler_timing_mode_drop_fraction():
+    """specific_timing strategies map to drop masks with the dispatch
+    schedule's exact dropped/dispatched ratio."""
+    import torch
+    from olearning_sim_amd.deviceflow.sampler import BehaviorSampler
+    spec = {"flow_dispatch": {
+        "use_strategy": True, "total_dispatch_amount": 100,
+        "specific_timing": {
+            "use": True,
+            "timings": [0, 1, 2, 3],
+            "amounts": [25, 25, 25, 25],
+            "drop_simulation": {"drop_amounts": [5, 5, 5, 5]}}}}
+    s = BehaviorSampler(json.dumps(spec), seed=3, device="cpu")
+    assert abs(s._timing_drop_fraction(0) - 0.2) < 1e-9
+    off, drop = s(0, 1000)
+    assert not off.any()
+    # ~20% dropped (binomial around 200 of 1000)
+    n = int(drop.sum())
+    assert 120 <= n <= 280
+
+
+def test_sampler_timing_mode_no_drops_is_noop():
+    import torch
+    from olearning_sim_amd.deviceflow.sampler import BehaviorSampler
+    spec = {"flow_dispatch": {
+        "use_strategy": True, "total_dispatch_amount": 10,
+        "specific_timing": {"use": True, "timings": [0, 1],
+                            "amounts": [5, 5]}}}
+    s = BehaviorSampler(json.dumps(spec), seed=3, device="cpu")
+    off, drop = s(1, 64)
+    assert not off.any() and not drop.any()
