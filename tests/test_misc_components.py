@@ -361,18 +361,29 @@ def test_drop_all_leaves_master_unchanged():
     torch.testing.assert_close(eng.master.flat, before)  # post-train)
 
 
-@pytest.mark.timeout(180)
+def _run_cli(args, repo, timeout=150, attempts=2):
+    """One retry: CLI subprocesses occasionally hit environment
+    transients (slow interpreter start under load); a genuine failure
+    fails both attempts."""
+    import subprocess, sys
+    for attempt in range(attempts):
+        out = subprocess.run([sys.executable, "-m", "olearning_sim_amd"]
+                             + args, cwd=repo, capture_output=True,
+                             text=True, timeout=timeout)
+        if out.returncode == 0:
+            return out
+    return out
+
+
+@pytest.mark.timeout(400)
 def test_cli_submit_wait_and_status(tmp_path):
     import subprocess, sys
     from test_manager import task_json
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     tf = tmp_path / "task.json"
     tf.write_text(task_json(task_id="t_cli"))
-    out = subprocess.run(
-        [sys.executable, "-m", "olearning_sim_amd",
-         "--data-dir", str(tmp_path / "data"),
-         "submit", str(tf), "--wait"],
-        cwd=repo, capture_output=True, text=True, timeout=150)
+    out = _run_cli(["--data-dir", str(tmp_path / "data"),
+                    "submit", str(tf), "--wait"], repo)
     assert out.returncode == 0, out.stderr[-2000:]
     lines = [json.loads(l) for l in out.stdout.strip().splitlines()]
     assert lines[0]["is_success"]
@@ -386,31 +397,25 @@ def test_cli_submit_wait_and_status(tmp_path):
         "task_status"] == "SUCCEEDED"
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(400)
 def test_cli_result_and_perf(tmp_path):
-    import subprocess, sys
     from test_manager import task_json
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     tf = tmp_path / "task.json"
     tf.write_text(task_json(task_id="t_cli_r"))
     data_dir = str(tmp_path / "data")
-    out = subprocess.run(
-        [sys.executable, "-m", "olearning_sim_amd",
-         "--data-dir", data_dir, "submit", str(tf), "--wait"],
-        cwd=repo, capture_output=True, text=True, timeout=150)
+    out = _run_cli(["--data-dir", data_dir, "submit", str(tf), "--wait"],
+                   repo)
     assert out.returncode == 0, out.stderr[-2000:]
-    res = subprocess.run(
-        [sys.executable, "-m", "olearning_sim_amd",
-         "--data-dir", data_dir, "result", "t_cli_r"],
-        cwd=repo, capture_output=True, text=True, timeout=60)
+    res = _run_cli(["--data-dir", data_dir, "result", "t_cli_r"], repo,
+                   timeout=90)
+    assert res.returncode == 0, res.stderr[-2000:]
     d = json.loads(res.stdout.strip().splitlines()[-1])
     assert d["task_status"] == "SUCCEEDED"
     assert d["logical_result"]["logical_result"][0]["simulation_target"][
         "success_num"]
-    perf = subprocess.run(
-        [sys.executable, "-m", "olearning_sim_amd",
-         "--data-dir", data_dir, "perf", "t_cli_r"],
-        cwd=repo, capture_output=True, text=True, timeout=60)
+    perf = _run_cli(["--data-dir", data_dir, "perf", "t_cli_r"], repo,
+                    timeout=90)
     assert perf.returncode == 0, perf.stderr[-2000:]
     json.loads(perf.stdout.strip().splitlines()[-1])
 
