@@ -123,10 +123,11 @@ class ResNet18(ClientBatchedModel):
     def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
         from ..ops.fused import groupnorm_act
         # OLSIM_CONV=custom routes 3x3 convs through the hand-written
-        # implicit-GEMM MFMA kernels (ops/csrc/client_conv.hip).  The
-        # default is MIOpen grouped conv, which currently wins at these
-        # shapes once its Find phase has run (see profiles/): custom
-        # ~70 TF/s vs CK ~110-150 TF/s steady state.
+        # implicit-GEMM MFMA kernels (ops/csrc/client_conv.hip; v5
+        # pipeline fwd 99-137 TF/s, conv fwd+dgrad+wgrad at CK parity).
+        # The default stays MIOpen grouped conv while the full custom
+        # round is 406 vs 387 ms at C=250 (see profiles/ and
+        # docs/ROADMAP.md for the flip plan).
         import os
         if x.is_cuda and x.dtype == torch.bfloat16 and \
                 os.environ.get("OLSIM_CONV", "") == "custom":
