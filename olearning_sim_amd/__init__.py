@@ -11,4 +11,22 @@ __version__ = "0.1.0"
 
 from . import utils  # noqa: F401
 
-__all__ = ["utils", "__version__"]
+__all__ = ["utils", "__version__", "SimulatorSession", "build_model",
+           "EngineJob", "LogicalEngine"]
+
+
+def __getattr__(name):
+    # lazy top-level conveniences (importing torch/fastapi only on use)
+    if name == "SimulatorSession":
+        from .session import SimulatorSession
+        return SimulatorSession
+    if name == "build_model":
+        from .models import build_model
+        return build_model
+    if name == "EngineJob":
+        from .engine import EngineJob
+        return EngineJob
+    if name == "LogicalEngine":
+        from .engine import LogicalEngine
+        return LogicalEngine
+    raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
