@@ -70,8 +70,12 @@ class PerformanceManager:
             by_metric.setdefault(r["metric"], []).append(r["value"])
         out: Dict[str, Any] = {"task_id": task_id, "metrics": {}}
         for m, vals in by_metric.items():
+            s = sorted(vals)
+            n = len(s)
             out["metrics"][m] = {
-                "count": len(vals), "last": vals[-1],
-                "mean": sum(vals) / len(vals),
-                "min": min(vals), "max": max(vals)}
+                "count": n, "last": vals[-1],
+                "mean": sum(vals) / n,
+                "min": s[0], "max": s[-1],
+                "p50": s[n // 2],
+                "p95": s[min(n - 1, (n * 95) // 100)]}
         return out

@@ -481,3 +481,15 @@ def test_examples_run_hermetically(tmp_path):
                              capture_output=True, text=True, timeout=240)
         assert out.returncode == 0, (script, out.stderr[-2000:])
         assert "SUCCEEDED" in out.stdout, (script, out.stdout[-2000:])
+
+
+def test_perf_summary_percentiles():
+    from olearning_sim_amd.perf.manager import PerformanceManager
+    pm = PerformanceManager(":memory:")
+    for i, v in enumerate([10.0, 20.0, 30.0, 40.0, 100.0]):
+        pm.record("tp", "round_time_s", v, round_idx=i)
+    m = pm.summary("tp")["metrics"]["round_time_s"]
+    assert m["count"] == 5 and m["last"] == 100.0
+    assert m["p50"] == 30.0
+    assert m["p95"] == 100.0
+    assert m["min"] == 10.0 and m["max"] == 100.0
