@@ -202,3 +202,20 @@ def test_dispatch_curve_recorded(svc):
     # release the flow; the curve must still be queryable
     svc.flow_release_step()
     assert svc.dispatch_curve("t_curve")[fid][-1]["accumulated"] == 6
+
+
+def test_outbound_subscriber_receives_directly(svc):
+    """A subscribed consumer (the in-process aggregation service role)
+    gets messages pushed instead of queued."""
+    got = []
+    svc.outbound.subscribe(got.append)
+    svc.register_task("t_sub", ["logical_simulation"])
+    fid = svc.notify_start("t_sub", "train", 0, "logical_simulation",
+                           strategy=rt_strategy(batch=1))
+    for i in range(3):
+        svc.publish(fid, "logical_simulation", {"i": i})
+    assert svc.drain_inbound()
+    svc.notify_complete("t_sub", "train", 0, "logical_simulation")
+    assert wait_until(lambda: len(got) == 3)
+    assert svc.outbound.qsize() == 0            # pushed, not queued
+    assert sorted(m.payload["i"] for m in got) == [0, 1, 2]
