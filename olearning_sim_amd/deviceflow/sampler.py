@@ -56,7 +56,14 @@ class BehaviorSampler:
 
     # ------------------------------------------------------------------
     def _flow_fractions(self, round_idx: int) -> Tuple[float, float]:
-        """(forwarded_fraction, dropped_fraction_of_forwarded) for flow mode."""
+        """(schedule_total, dropped_fraction_of_forwarded) for flow mode.
+
+        schedule_total is the flow's dispatch budget
+        (total_dispatch_amount): a cohort larger than it sees the
+        surplus clients as offline shortfall (the dispatcher forwards
+        only the scheduled amounts before round release; reference
+        dispatcher.py:174-242), <= 0 meaning "no schedule" (forward
+        everything)."""
         if round_idx in self._cache:
             return self._cache[round_idx]
         flow = _dct(self.spec.get("flow_dispatch"))
@@ -71,10 +78,11 @@ class BehaviorSampler:
         rules = _dct(spec.get("dispatch_rules"))
         domains = rules.get("domains", [])
         functions = rules.get("functions", [])
-        if not intervals or len(intervals) != len(domains) != len(functions):
-            if not (intervals and len(intervals) == len(domains) == len(functions)):
-                self._cache[round_idx] = (1.0, 0.0)
-                return self._cache[round_idx]
+        if not (intervals and len(intervals) == len(domains) == len(functions)):
+            # invalid config: canonical Strategy returns [],[],[] — no
+            # schedule, everything forwards
+            self._cache[round_idx] = (0.0, 0.0)
+            return self._cache[round_idx]
         # integrate every interval's curve on device in one grid pass
         per_area = []
         for interval, domain, func in zip(intervals, domains, functions):
@@ -92,10 +100,12 @@ class BehaviorSampler:
             per_area.append(float(pieces.clamp_min(0).sum()) * scale)
         total_area = sum(per_area)
         if total_area <= 0 or total <= 0:
-            self._cache[round_idx] = (1.0, 0.0)
+            # zero-area curves / zero budget: Strategy yields no
+            # schedule -> everything forwards, nothing drops
+            self._cache[round_idx] = (0.0, 0.0)
             return self._cache[round_idx]
-        # amounts are exactly total (they always apportion fully); the
-        # engine-visible knob is the drop settings
+        # the schedule apportions exactly `total` messages over the
+        # slots; drops are a fraction of THOSE
         drop_spec = _dct(spec.get("drop_simulation"))
         drop_frac = 0.0
         if "drop_probability" in drop_spec:
@@ -103,7 +113,7 @@ class BehaviorSampler:
             drop_frac = sum(p * a for p, a in zip(probs, per_area)) / total_area
         elif "drop_amounts" in drop_spec:
             drop_frac = min(1.0, sum(drop_spec["drop_amounts"]) / max(1, total))
-        self._cache[round_idx] = (1.0, drop_frac)
+        self._cache[round_idx] = (float(total), drop_frac)
         return self._cache[round_idx]
 
     def _timing_drop_fraction(self, round_idx: int) -> float:
@@ -168,7 +178,12 @@ class BehaviorSampler:
             return offline, dropped
         if flow.get("use_strategy", False) and \
                 _dct(flow.get("specific_interval")).get("use", False):
-            fwd_frac, drop_frac = self._flow_fractions(round_idx)
+            sched_total, drop_frac = self._flow_fractions(round_idx)
+            # forwarded fraction = schedule budget vs this round's
+            # cohort: a budget below the cohort is offline shortfall
+            # (the dispatcher never forwards more than the schedule)
+            fwd_frac = 1.0 if sched_total <= 0 else \
+                min(1.0, sched_total / max(1, cohort))
             if fwd_frac < 1.0:
                 n_off = int(round((1.0 - fwd_frac) * cohort))
                 if n_off > 0:

@@ -47,6 +47,22 @@ AREA_CALCULATION_NUM = 100  # integration sub-steps per 1-second slot
 _SAFE_EVAL_GLOBALS = {"math": math, "np": np, "abs": abs, "min": min,
                       "max": max, "pow": pow, "__builtins__": {}}
 
+DEFAULT_TIME_ZONE = "Asia/Shanghai"   # reference strategy.py:116-121
+
+
+def _zone_now(spec: Dict[str, Any]) -> datetime:
+    """Wall-clock 'now' in the strategy's configured time zone (naive,
+    for comparison against the spec's absolute timestamps).  The
+    reference converts via pytz with Asia/Shanghai as the default; a
+    host whose local TZ differs would otherwise shift every absolute
+    schedule by the TZ offset."""
+    tz_name = spec.get("time_zone") or DEFAULT_TIME_ZONE
+    try:
+        from zoneinfo import ZoneInfo
+        return datetime.now(ZoneInfo(tz_name)).replace(tzinfo=None)
+    except Exception:
+        return datetime.now()
+
 
 def _eval_rate(func_string: str, t: float) -> float:
     """Evaluate an arrival-rate expression f(t) (e.g. 'math.sin(t)+1')."""
@@ -131,7 +147,7 @@ class Strategy:
 
         if time_type == "absolute":
             fmt = "%Y-%m-%d %H:%M:%S"
-            current = now or datetime.now()
+            current = now or _zone_now(spec)
             frac = current.microsecond / 1e6
             cur = datetime.strptime(current.strftime(fmt), fmt)
             abs_secs = [(datetime.strptime(t, fmt) - cur).total_seconds()
@@ -192,7 +208,7 @@ class Strategy:
                 total, rel, domains, functions, dict(drop_spec), rng)
             if not timing:
                 return [], [], []
-            current = now or datetime.now()
+            current = now or _zone_now(spec)
             frac = current.microsecond / 1e6
             cur = datetime.strptime(current.strftime(fmt), fmt)
             start = datetime.strptime(abs_intervals[0][0], fmt)

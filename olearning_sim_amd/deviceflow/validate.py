@@ -85,12 +85,23 @@ class ValidateStrategy:
                 "relative timings must be non-decreasing"
         else:
             assert time_type == "absolute", f"unknown time_type {time_type!r}"
+            self._check_time_zone(st)
             for round_list in timings:
                 assert len(round_list) == len(amounts), \
                     "each round's timings must match amounts length"
                 for t in round_list:
                     datetime.strptime(t, _FMT)
         self._check_drop(st.get("drop_simulation", {}), len(amounts))
+
+    @staticmethod
+    def _check_time_zone(spec: Dict[str, Any]) -> None:
+        tz = spec.get("time_zone")
+        if tz:
+            from zoneinfo import ZoneInfo
+            try:
+                ZoneInfo(tz)
+            except Exception:
+                raise AssertionError(f"unknown time_zone {tz!r}")
 
     def _check_interval(self, si: Dict[str, Any]) -> None:
         time_type = si.get("time_type", "relative")
@@ -117,6 +128,7 @@ class ValidateStrategy:
             check_rel(intervals)
         else:
             assert time_type == "absolute", f"unknown time_type {time_type!r}"
+            self._check_time_zone(si)
             for round_iv in intervals:
                 prev_end = None
                 assert len(round_iv) == len(domains)

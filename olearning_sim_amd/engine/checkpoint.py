@@ -22,7 +22,14 @@ DEFAULT_STYLE = "{task_id}_{current_round}_result_model.safetensors"
 def checkpoint_name(task_id: str, current_round: int,
                     model_update_style: str = "") -> str:
     style = model_update_style or DEFAULT_STYLE
-    return style.format(task_id=task_id, current_round=current_round)
+    name = style.format(task_id=task_id, current_round=current_round)
+    # the rendered name is joined into the checkpoint directory: a
+    # template (or task_id) carrying separators or '..' must not be
+    # able to address files outside it (validate.py rejects these at
+    # submit time; this guards direct engine use)
+    if os.sep in name or (os.altsep and os.altsep in name) or ".." in name:
+        raise ValueError(f"unsafe checkpoint name {name!r}")
+    return name
 
 
 def save_checkpoint(directory: str, task_id: str, current_round: int,

@@ -142,8 +142,10 @@ class LogicalEngine:
         bytes_per_client += (self.job.batch_size
                              * self.model.act_elems_per_sample
                              * self.dtype.itemsize)
-        free = torch.cuda.get_device_properties(self.device).total_memory
-        budget = int(free * 0.5)
+        # budget against memory actually free NOW (total_memory would
+        # over-commit the moment anything else holds HBM on this device)
+        free, total = torch.cuda.mem_get_info(self.device)
+        budget = int(min(free * 0.8, total * 0.5))
         return max(1, min(cohort, budget // max(1, bytes_per_client)))
 
     def select_cohort(self, round_idx: int) -> torch.Tensor:
@@ -278,8 +280,10 @@ class LogicalEngine:
         dyn = self.tier_dynamic
         if len(dyn) != T:
             dyn = [job.dynamic_num] * T
-        round_failed = (failed > job.dynamic_num if T == 1
-                        else any(f > d for f, d in zip(fail_t, dyn)))
+        # per-segment tolerance uniformly (reference checks each tier's
+        # failed > dynamic_nums; job.dynamic_num sums across all
+        # data/tiers and would inflate the single-segment tolerance)
+        round_failed = any(f > d for f, d in zip(fail_t, dyn))
         return {
             "success": success,
             "failed": failed,

@@ -35,8 +35,18 @@ def stage_operator_code(code_path: str, entry_file: str, operator_name: str,
     """
     if not code_path or code_path.startswith("builtin:"):
         return None
+    # operator_name comes from submitted task JSON: it must resolve to a
+    # direct child of work_dir (never '..', absolute paths or separators
+    # — the resolved dst is rmtree'd below)
+    if (not operator_name or "/" in operator_name or "\\" in operator_name
+            or operator_name in (".", "..")):
+        raise OperatorStagingError(
+            f"operator name {operator_name!r} must be a plain directory name")
     os.makedirs(work_dir, exist_ok=True)
     dst = os.path.join(work_dir, operator_name)
+    if os.path.realpath(os.path.dirname(dst)) != os.path.realpath(work_dir):
+        raise OperatorStagingError(
+            f"operator name {operator_name!r} escapes the task work dir")
     if os.path.isdir(dst):
         shutil.rmtree(dst)
 
@@ -54,6 +64,13 @@ def stage_operator_code(code_path: str, entry_file: str, operator_name: str,
         shutil.copytree(src, dst)
     elif src.endswith(".zip"):
         with zipfile.ZipFile(src) as z:
+            base = os.path.realpath(dst)
+            for m in z.namelist():
+                tgt = os.path.realpath(os.path.join(dst, m))
+                if tgt != base and not tgt.startswith(base + os.sep):
+                    raise OperatorStagingError(
+                        f"operator {operator_name}: zip member {m!r} "
+                        f"escapes the staging dir")
             z.extractall(dst)
         # a zip that wraps everything in one top-level dir is flattened
         entries = os.listdir(dst)

@@ -63,6 +63,13 @@ def _has_ext(s: str, ext: str) -> bool:
     return s.endswith(ext) and len(s) > len(ext)
 
 
+def _path_safe(s: str) -> bool:
+    """True when s can be joined into a work/checkpoint directory
+    without escaping it (staging.py rmtree's work_dir/<name>;
+    checkpoint.py joins the rendered model_update_style)."""
+    return "/" not in s and "\\" not in s and ".." not in s
+
+
 class ValidationError(Exception):
     pass
 
@@ -197,6 +204,8 @@ class ValidateParameters:
         need(_ascii_ok(cfg.user_id), "user_id contains non-ASCII characters")
         need(cfg.task_id != "", "task_id must not be empty")
         need(_ascii_ok(cfg.task_id), "task_id contains non-ASCII characters")
+        need(_path_safe(cfg.task_id) and cfg.task_id not in (".", ".."),
+             "task_id must not contain path separators or '..'")
 
         for i, d in enumerate(cfg.target.data):
             where = f"target.data[{i}] ({d.name!r})"
@@ -246,6 +255,8 @@ class ValidateParameters:
             need(op.name != "", f"operators[{i}].name must not be empty")
             need(_ascii_ok(op.name), f"{where}: name has non-ASCII characters")
             need(" " not in op.name, f"{where}: name must not contain spaces")
+            need(_path_safe(op.name) and op.name not in (".", ".."),
+                 f"{where}: name must not contain path separators or '..'")
             bc = op.operation_behavior_controller
             need(_ascii_ok(bc.strategy_gradient_house),
                  f"{where}: strategy_gradient_house has non-ASCII characters")
@@ -259,6 +270,9 @@ class ValidateParameters:
                  f"{where}: model_path has non-ASCII characters")
             need(_ascii_ok(op.model.model_update_style),
                  f"{where}: model_update_style has non-ASCII characters")
+            need(_path_safe(op.model.model_update_style),
+                 f"{where}: model_update_style must not contain path "
+                 f"separators or '..'")
             lsim = op.logical_simulation
             need(lsim.operator_transfer_type in FILE_TRANSFER_TYPES,
                  f"{where}: unknown logical operator_transfer_type")

@@ -266,6 +266,67 @@ def test_operator_code_staging(tmp_path):
     assert os.path.exists(os.path.join(dst, "train.py"))
 
 
+def test_operator_staging_rejects_path_traversal(tmp_path):
+    """Operator names from task JSON must never address paths outside
+    the task work dir (the resolved target is rmtree'd)."""
+    from olearning_sim_amd.task.staging import (stage_operator_code,
+                                                OperatorStagingError)
+    src = tmp_path / "opsrc"
+    src.mkdir()
+    (src / "train.py").write_text("print('hi')")
+    victim = tmp_path / "victim"
+    victim.mkdir()
+    (victim / "keep.txt").write_text("data")
+    work = tmp_path / "w"
+    for bad in ("../victim", "/abs/path", "..", ".", "a/b", "", "a\\b"):
+        with pytest.raises(OperatorStagingError):
+            stage_operator_code(str(src), "train.py", bad, str(work))
+    assert (victim / "keep.txt").exists()
+    # zip members that escape the staging dir are rejected
+    import zipfile
+    z = tmp_path / "evil.zip"
+    with zipfile.ZipFile(z, "w") as zf:
+        zf.writestr("../escape.py", "print('hi')")
+        zf.writestr("train.py", "print('hi')")
+    with pytest.raises(OperatorStagingError):
+        stage_operator_code(str(z), "train.py", "train", str(tmp_path / "w5"))
+    assert not (tmp_path / "escape.py").exists()
+
+
+def test_validate_rejects_unsafe_names():
+    """Submit-time validation rejects operator names / update styles /
+    task ids carrying path separators or '..'."""
+    import copy
+    import json as _json
+    from olearning_sim_amd.task import json2taskconfig
+    from olearning_sim_amd.task.validate import ValidateParameters
+    from test_schema import EXAMPLE
+
+    def check(raw):
+        v = ValidateParameters()
+        cfg = json2taskconfig(_json.dumps(raw))
+        return v.validate_task_parameters(raw, cfg), v.last_error
+
+    assert check(EXAMPLE)[0]
+    bad = copy.deepcopy(EXAMPLE)
+    bad["task_id"] = "../t1"
+    ok, err = check(bad)
+    assert not ok and "task_id" in err
+    bad = copy.deepcopy(EXAMPLE)
+    bad["operatorflow"]["operators"][0]["name"] = "../train"
+    ok, err = check(bad)
+    assert not ok
+    bad = copy.deepcopy(EXAMPLE)
+    bad["operatorflow"]["operators"][0]["model"]["model_update_style"] = \
+        "../{task_id}_{current_round}.safetensors"
+    ok, err = check(bad)
+    assert not ok and "model_update_style" in err
+    # engine-level guard on the rendered checkpoint name
+    from olearning_sim_amd.engine.checkpoint import checkpoint_name
+    with pytest.raises(ValueError):
+        checkpoint_name("t1", 0, "../{task_id}_{current_round}.bin")
+
+
 def test_custom_model_plugin(tmp_path, monkeypatch):
     import sys
     (tmp_path / "user_models.py").write_text(

@@ -171,14 +171,36 @@ def test_sampler_drop_fraction_matches_strategy_schedule():
     assert sum(amounts) == 1000
     canonical_drop = sum(len(d) for d in drops) / 1000
     sampler = BehaviorSampler(s, seed=0, device="cpu")
-    fwd_frac, drop_frac = sampler._flow_fractions(0)
-    assert fwd_frac == 1.0
+    sched_total, drop_frac = sampler._flow_fractions(0)
+    # the sampler sees the dispatcher's schedule budget (sum(amounts))
+    assert sched_total == sum(amounts) == 1000
     assert abs(drop_frac - 0.3) < 0.02          # configured probability
     assert abs(canonical_drop - drop_frac) < 0.06  # sampled realisation
-    # masks have the right statistics over a large cohort
-    offline, dropped = sampler(0, 20000)
+    # cohort within the schedule budget: nobody offline, drops ~ p
+    offline, dropped = sampler(0, 1000)
     assert int(offline.sum()) == 0
-    assert abs(float(dropped.float().mean()) - 0.3) < 0.02
+    assert abs(float(dropped.float().mean()) - 0.3) < 0.04
+
+
+def test_sampler_interval_shortfall_is_offline():
+    """A flow schedule whose total_dispatch_amount is below the cohort
+    forwards only the scheduled amount: the surplus clients are offline
+    shortfall (dispatcher forwards exactly the schedule's amounts;
+    reference dispatcher.py:174-242), and drops apply to forwarded
+    clients only."""
+    from olearning_sim_amd.deviceflow.sampler import BehaviorSampler
+    s = flow({"total_dispatch_amount": 1000,
+              "specific_interval": {
+                  "use": True, "intervals": [[0, 10]],
+                  "dispatch_rules": {"domains": [[0.0, 6.28]],
+                                     "functions": ["math.sin(t)+1"]},
+                  "drop_simulation": {"drop_probability": [0.3]}}})
+    sampler = BehaviorSampler(s, seed=0, device="cpu")
+    offline, dropped = sampler(0, 4000)
+    assert int(offline.sum()) == 3000           # 4000 cohort - 1000 budget
+    assert not bool((dropped & offline).any())  # drops only among online
+    online_drop = float(dropped.sum()) / 1000
+    assert abs(online_drop - 0.3) < 0.06
 
 
 def test_sampler_torch_integration_matches_scalar_area():
