@@ -35,9 +35,10 @@ def _torch_eval_rate(func_string: str, t: torch.Tensor) -> torch.Tensor:
     'math.*'/'np.*' resolve to torch equivalents; plain arithmetic works
     natively on the tensor.
     """
+    from .strategy import compile_rate_expr
     ns = {"math": torch, "np": torch, "t": t, "abs": torch.abs,
           "min": torch.minimum, "max": torch.maximum, "__builtins__": {}}
-    out = eval(func_string, ns)
+    out = eval(compile_rate_expr(func_string), ns)
     if not torch.is_tensor(out):
         out = torch.full_like(t, float(out))
     return out

@@ -64,9 +64,67 @@ def _zone_now(spec: Dict[str, Any]) -> datetime:
         return datetime.now()
 
 
+# -- arrival-rate expression compiler ----------------------------------
+# Rate functions are user task input ("math.sin(t)+1").  Instead of a
+# scrubbed-globals eval (escapable via attribute chains), parse the
+# expression and allow only arithmetic, numeric literals, the variable
+# t, calls on math.*/np.* and abs/min/max/pow — then compile the vetted
+# AST once and cache the code object (the integrator calls f(t) 100x
+# per schedule slot).
+
+_ALLOWED_NODES = (
+    "Expression", "BinOp", "UnaryOp", "Call", "Constant", "Name", "Load",
+    "Attribute", "Add", "Sub", "Mult", "Div", "FloorDiv", "Mod", "Pow",
+    "USub", "UAdd", "IfExp", "Compare", "Lt", "LtE", "Gt", "GtE", "Eq",
+    "NotEq", "BoolOp", "And", "Or", "Tuple",
+)
+_ALLOWED_NAMES = {"t", "abs", "min", "max", "pow", "math", "np"}
+_EXPR_CACHE: Dict[str, Any] = {}
+
+
+class RateExprError(ValueError):
+    pass
+
+
+def compile_rate_expr(func_string: str):
+    """Compile an arrival-rate expression to a code object, allowing
+    only whitelisted AST nodes/names.  Raises RateExprError otherwise."""
+    code = _EXPR_CACHE.get(func_string)
+    if code is not None:
+        return code
+    import ast
+    try:
+        tree = ast.parse(func_string, mode="eval")
+    except SyntaxError as e:
+        raise RateExprError(f"invalid rate expression: {e}") from None
+    for node in ast.walk(tree):
+        kind = type(node).__name__
+        if kind not in _ALLOWED_NODES:
+            raise RateExprError(
+                f"rate expression uses disallowed syntax {kind!r}")
+        if isinstance(node, ast.Name) and node.id not in _ALLOWED_NAMES:
+            raise RateExprError(
+                f"rate expression uses unknown name {node.id!r}")
+        if isinstance(node, ast.Attribute):
+            # only one attribute level: math.X / np.X, no dunders
+            if not (isinstance(node.value, ast.Name)
+                    and node.value.id in ("math", "np")
+                    and not node.attr.startswith("_")):
+                raise RateExprError(
+                    "rate expressions may only access math.* / np.*")
+        if isinstance(node, ast.Constant) and \
+                not isinstance(node.value, (int, float, complex)):
+            raise RateExprError("rate expression literals must be numeric")
+    code = compile(tree, "<rate-expr>", "eval")
+    if len(_EXPR_CACHE) < 4096:
+        _EXPR_CACHE[func_string] = code
+    return code
+
+
 def _eval_rate(func_string: str, t: float) -> float:
     """Evaluate an arrival-rate expression f(t) (e.g. 'math.sin(t)+1')."""
-    return float(eval(func_string, dict(_SAFE_EVAL_GLOBALS), {"t": t}))
+    code = compile_rate_expr(func_string)
+    return float(eval(code, dict(_SAFE_EVAL_GLOBALS), {"t": t}))
 
 
 

@@ -247,3 +247,26 @@ def test_sampler_timing_mode_no_drops_is_noop():
     s = BehaviorSampler(json.dumps(spec), seed=3, device="cpu")
     off, drop = s(1, 64)
     assert not off.any() and not drop.any()
+
+
+def test_rate_expression_whitelist():
+    """Arrival-rate expressions are AST-whitelisted: arithmetic and
+    math.*/np.* only — attribute-chain escapes and arbitrary names are
+    rejected at parse time (the reference bare-evals task input)."""
+    from olearning_sim_amd.deviceflow.strategy import (_eval_rate,
+                                                       RateExprError)
+    import math as m
+    assert _eval_rate("math.sin(t)+1", 0.3) == pytest.approx(m.sin(0.3) + 1)
+    assert _eval_rate("min(2.0, max(t, 0.5)) * 2", 0.1) == pytest.approx(1.0)
+    assert _eval_rate("np.exp(-t)", 1.0) == pytest.approx(m.exp(-1.0))
+    for bad in (
+        "abs.__self__",                       # builtins escape
+        "().__class__",                       # object traversal
+        "__import__('os')",                   # unknown name
+        "math.__dict__",                      # dunder attribute
+        "[x for x in (1,)]",                  # comprehension
+        "'a'*9999999",                        # non-numeric literal
+        "lambda: 1",
+    ):
+        with pytest.raises(RateExprError):
+            _eval_rate(bad, 0.0)
