@@ -20,7 +20,8 @@ def build(verbose: bool = True) -> str:
     os.makedirs(build_dir, exist_ok=True)
     sources = [os.path.join(csrc, f) for f in
                ("bindings.cpp", "fused_update.hip", "aggregate.hip",
-                "cross_entropy.hip", "groupnorm.hip", "client_conv.hip")]
+                "cross_entropy.hip", "groupnorm.hip", "client_conv.hip",
+                "client_conv2.hip")]
     from torch.utils.cpp_extension import load
     mod_path = load(
         name="olsim_hip_ops",

@@ -50,10 +50,50 @@ class _Conv3x3Fn(torch.autograd.Function):
         return dx, dw, None
 
 
+class _Conv3x3PadFn(torch.autograd.Function):
+    """v6 padded path: the kernels gather from a 1-element zero halo so
+    the implicit-im2col reads carry no bounds masks (client_conv2.hip).
+    The padded activation is saved for backward (wgrad reuses it)."""
+
+    @staticmethod
+    def forward(ctx, x, w, stride):
+        ops = load_hip_ops(required=True)
+        x_pad = F.pad(x, (1, 1, 1, 1))
+        y = ops.conv3x3_fwd_p(x_pad, w, stride)
+        ctx.save_for_backward(x_pad, w)
+        ctx.stride = stride
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x_pad, w = ctx.saved_tensors
+        ops = load_hip_ops(required=True)
+        dy = dy.contiguous()
+        dx = dw = None
+        H, W = x_pad.shape[3] - 2, x_pad.shape[4] - 2
+        if ctx.needs_input_grad[0]:
+            dy_pad = F.pad(dy, (1, 1, 1, 1))
+            dx = ops.conv3x3_dgrad_p(dy_pad, w, H, W, ctx.stride)
+        if ctx.needs_input_grad[1]:
+            dw = ops.conv3x3_wgrad_p(x_pad, dy, ctx.stride).to(w.dtype)
+        return dx, dw, None
+
+
+def _v6_ok(ops, x: torch.Tensor, w: torch.Tensor, stride: int) -> bool:
+    import os
+    if os.environ.get("OLSIM_CONV_V", "") == "5":
+        return False
+    C, IC, B, H, W = x.shape
+    return bool(ops.conv3x3_v6_ok(IC, w.shape[1], B, H, W, stride))
+
+
 def client_conv3x3(x: torch.Tensor, w: torch.Tensor,
                    stride: int = 1) -> torch.Tensor:
     """y[C,OC,B,OH,OW] = conv3x3(x[C,IC,B,H,W], w[C,OC,IC,3,3]), pad 1."""
     if x.is_cuda and x.dtype == torch.bfloat16:
+        ops = load_hip_ops(required=True)
+        if _v6_ok(ops, x, w, stride):
+            return _Conv3x3PadFn.apply(x.contiguous(), w.contiguous(), stride)
         return _Conv3x3Fn.apply(x.contiguous(), w.contiguous(), stride)
     return _cpu_conv3x3(x, w, stride)
 

@@ -39,6 +39,18 @@ extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
                                   int C, int IC, int OC, int B, int H, int W,
                                   int stride, hipStream_t stream);
 
+extern "C" int ols_conv3x3_v6_ok(int IC, int OC, int B, int H, int W,
+                                 int stride);
+extern "C" void ols_conv3x3_fwd_p(const void* xp, const void* w, void* y,
+                                  int C, int IC, int OC, int B, int H, int W,
+                                  int stride, hipStream_t stream);
+extern "C" void ols_conv3x3_dgrad_p(const void* dyp, const void* w, void* dx,
+                                    int C, int IC, int OC, int B, int H,
+                                    int W, int stride, hipStream_t stream);
+extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, float* dw,
+                                    int C, int IC, int OC, int B, int H,
+                                    int W, int stride, hipStream_t stream);
+
 extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const void* dy, void* dx, void* dres,
                                   const float* mean, const float* rstd,
@@ -231,6 +243,62 @@ at::Tensor conv3x3_wgrad(at::Tensor x, at::Tensor dy, int64_t stride) {
   return dw;
 }
 
+// ---- v6 padded family (client_conv2.hip) --------------------------------
+// x_pad: [C, IC, B, H+2, W+2] bf16 (1-element zero halo per plane)
+
+bool conv3x3_v6_ok(int64_t IC, int64_t OC, int64_t B, int64_t H, int64_t W,
+                   int64_t stride) {
+  return ols_conv3x3_v6_ok((int)IC, (int)OC, (int)B, (int)H, (int)W,
+                           (int)stride) != 0;
+}
+
+at::Tensor conv3x3_fwd_p(at::Tensor xp, at::Tensor w, int64_t stride) {
+  TORCH_CHECK(xp.is_cuda() && xp.is_contiguous() && xp.dim() == 5);
+  TORCH_CHECK(w.is_contiguous() && w.dim() == 5 && w.size(3) == 3);
+  TORCH_CHECK(xp.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16);
+  int C = xp.size(0), IC = xp.size(1), B = xp.size(2),
+      H = xp.size(3) - 2, W = xp.size(4) - 2, OC = w.size(1);
+  TORCH_CHECK(w.size(0) == C && w.size(2) == IC);
+  TORCH_CHECK(ols_conv3x3_v6_ok(IC, OC, B, H, W, (int)stride),
+              "shape unsupported by the v6 conv path");
+  int OH = (H + stride - 1) / stride, OW = (W + stride - 1) / stride;
+  auto y = at::empty({C, OC, B, OH, OW}, xp.options());
+  ols_conv3x3_fwd_p(xp.data_ptr(), w.data_ptr(), y.data_ptr(), C, IC, OC, B,
+                    H, W, (int)stride,
+                    at::cuda::getCurrentCUDAStream().stream());
+  return y;
+}
+
+// dy_pad: [C, OC, B, OH+2, OW+2]; returns dx [C, IC, B, H, W]
+at::Tensor conv3x3_dgrad_p(at::Tensor dyp, at::Tensor w, int64_t H, int64_t W,
+                           int64_t stride) {
+  TORCH_CHECK(dyp.is_cuda() && dyp.is_contiguous() && dyp.dim() == 5);
+  TORCH_CHECK(w.is_contiguous() && w.dim() == 5);
+  int C = dyp.size(0), OC = dyp.size(1), B = dyp.size(2);
+  int IC = w.size(2);
+  TORCH_CHECK(ols_conv3x3_v6_ok(IC, OC, B, H, W, (int)stride),
+              "shape unsupported by the v6 conv path");
+  auto dx = at::empty({C, IC, B, H, W}, dyp.options());
+  ols_conv3x3_dgrad_p(dyp.data_ptr(), w.data_ptr(), dx.data_ptr(), C, IC, OC,
+                      B, (int)H, (int)W, (int)stride,
+                      at::cuda::getCurrentCUDAStream().stream());
+  return dx;
+}
+
+at::Tensor conv3x3_wgrad_p(at::Tensor xp, at::Tensor dy, int64_t stride) {
+  TORCH_CHECK(xp.is_cuda() && xp.is_contiguous() && dy.is_contiguous());
+  int C = xp.size(0), IC = xp.size(1), B = xp.size(2),
+      H = xp.size(3) - 2, W = xp.size(4) - 2, OC = dy.size(1);
+  TORCH_CHECK(ols_conv3x3_v6_ok(IC, OC, B, H, W, (int)stride),
+              "shape unsupported by the v6 conv path");
+  auto dw = at::empty({C, OC, IC, 3, 3}, xp.options().dtype(at::kFloat));
+  ols_conv3x3_wgrad_p(xp.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), C,
+                      IC, OC, B, H, W, (int)stride,
+                      at::cuda::getCurrentCUDAStream().stream());
+  return dw;
+}
+
 }  // namespace
 
 TORCH_LIBRARY(olsim_hip, m) {
@@ -248,6 +316,11 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("conv3x3_fwd(Tensor x, Tensor w, int stride) -> Tensor");
   m.def("conv3x3_dgrad(Tensor dy, Tensor w, int H, int W, int stride) -> Tensor");
   m.def("conv3x3_wgrad(Tensor x, Tensor dy, int stride) -> Tensor");
+  m.def("conv3x3_v6_ok(int IC, int OC, int B, int H, int W, int stride) -> bool",
+        &conv3x3_v6_ok);
+  m.def("conv3x3_fwd_p(Tensor xp, Tensor w, int stride) -> Tensor");
+  m.def("conv3x3_dgrad_p(Tensor dyp, Tensor w, int H, int W, int stride) -> Tensor");
+  m.def("conv3x3_wgrad_p(Tensor xp, Tensor dy, int stride) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
@@ -260,4 +333,7 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("conv3x3_fwd", &conv3x3_fwd);
   m.impl("conv3x3_dgrad", &conv3x3_dgrad);
   m.impl("conv3x3_wgrad", &conv3x3_wgrad);
+  m.impl("conv3x3_fwd_p", &conv3x3_fwd_p);
+  m.impl("conv3x3_dgrad_p", &conv3x3_dgrad_p);
+  m.impl("conv3x3_wgrad_p", &conv3x3_wgrad_p);
 }

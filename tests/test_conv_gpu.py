@@ -33,7 +33,43 @@ SHAPES = [
     (2, 3, 64, 4, 32, 1),      # stem: IC=3 (K=27, sub-tile)
     (2, 512, 512, 2, 4, 1),    # deepest stage
     (1, 128, 256, 3, 8, 2),
+    # v6 template variants (client_conv2.hip): stride-2 parity classes
+    # at LG_OW 4/3/2 and small-plane fwd/dgrad sub-row hoisting
+    (2, 64, 128, 4, 32, 2),    # s2 classes, dy plane 16x16 (LG4)
+    (2, 128, 256, 4, 16, 2),   # s2 classes, dy plane 8x8 (LG3)
+    (2, 256, 512, 2, 8, 2),    # s2 classes, dy plane 4x4 (LG2)
+    (2, 64, 64, 4, 8, 1),      # fwd/dgrad LG3 sub-rows
+    (2, 128, 128, 2, 4, 1),    # fwd/dgrad LG2 sub-rows
+    (9, 64, 64, 2, 16, 1),     # C not a multiple of 8: XCD remap tail
 ]
+
+
+def test_v6_matches_v5_paths():
+    """The padded v6 kernels must agree with the v5 family on the same
+    inputs (both bf16; tolerance covers bf16 accumulation-order drift)."""
+    import os
+    from olearning_sim_amd.ops.conv import client_conv3x3
+    g = torch.Generator().manual_seed(7)
+    for (C, IC, OC, B, H, st) in [(3, 64, 64, 4, 16, 1),
+                                  (2, 64, 128, 4, 16, 2),
+                                  (2, 128, 128, 2, 8, 1)]:
+        x0 = torch.randn(C, IC, B, H, H, generator=g) * 0.5
+        w0 = torch.randn(C, OC, IC, 3, 3, generator=g) * 0.1
+        dy0 = torch.randn(C, OC, B, H // st, H // st, generator=g) * 0.1
+        outs = {}
+        for ver in ("5", "6"):
+            os.environ["OLSIM_CONV_V"] = ver if ver == "5" else ""
+            try:
+                xg = x0.to(torch.bfloat16).cuda().requires_grad_(True)
+                wg = w0.to(torch.bfloat16).cuda().requires_grad_(True)
+                y = client_conv3x3(xg, wg, st)
+                y.backward(dy0.to(torch.bfloat16).cuda())
+                outs[ver] = (y.detach(), xg.grad.clone(), wg.grad.clone())
+            finally:
+                os.environ.pop("OLSIM_CONV_V", None)
+        for a, b in zip(outs["5"], outs["6"]):
+            torch.testing.assert_close(a.float(), b.float(),
+                                       atol=5e-2, rtol=5e-2)
 
 
 @pytest.mark.parametrize("C,IC,OC,B,H,stride", SHAPES)
