@@ -40,6 +40,18 @@ class LeNet5(ClientBatchedModel):
         }
 
     def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        # Fast path: client-channel-first layout + the hand-written
+        # valid-conv5x5 MFMA kernels with fused bias+ReLU
+        # (ops/csrc/client_conv5.hip) — replaces MIOpen's naive_conv_*
+        # fallbacks at thousands of groups (~50% of round GPU time in
+        # profiles/fedprox_churn_r01.md).  OLSIM_CONV=miopen forces the
+        # grouped reference path.
+        import os
+        if x.is_cuda and x.dtype == torch.bfloat16 and \
+                os.environ.get("OLSIM_CONV", "custom") == "custom":
+            from ..ops.fused import hip_ops_available
+            if hip_ops_available():
+                return self.forward_cbf(params, x)
         # x: [C, B, 3, 32, 32] -> channel-grouped [B, C*3, 32, 32]
         C, B = x.shape[0], x.shape[1]
         xg = x.permute(1, 0, 2, 3, 4).reshape(B, C * self.in_ch, 32, 32)
@@ -48,6 +60,21 @@ class LeNet5(ClientBatchedModel):
         h = F.relu(bconv2d(h, params["conv2.w"], C, params["conv2.b"]))    # 10
         h = F.max_pool2d(h, 2)                                             # 5
         h = h.reshape(B, C, 16 * 5 * 5).permute(1, 0, 2)                   # [C,B,400]
+        h = F.relu(blinear(h, params["fc1.w"], params["fc1.b"]))
+        h = F.relu(blinear(h, params["fc2.w"], params["fc2.b"]))
+        return blinear(h, params["fc3.w"], params["fc3.b"])
+
+    def forward_cbf(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.conv import client_conv5x5
+        C, B = x.shape[0], x.shape[1]
+        h = x.permute(0, 2, 1, 3, 4).contiguous()          # [C, 3, B, 32, 32]
+        h = client_conv5x5(h, params["conv1.w"], params["conv1.b"],
+                           relu=True)                      # [C, 6, B, 28, 28]
+        h = F.max_pool2d(h.view(C * 6, B, 28, 28), 2).view(C, 6, B, 14, 14)
+        h = client_conv5x5(h, params["conv2.w"], params["conv2.b"],
+                           relu=True)                      # [C, 16, B, 10, 10]
+        h = F.max_pool2d(h.view(C * 16, B, 10, 10), 2)     # [C*16, B, 5, 5]
+        h = h.view(C, 16, B, 25).permute(0, 2, 1, 3).reshape(C, B, 400)
         h = F.relu(blinear(h, params["fc1.w"], params["fc1.b"]))
         h = F.relu(blinear(h, params["fc2.w"], params["fc2.b"]))
         return blinear(h, params["fc3.w"], params["fc3.b"])

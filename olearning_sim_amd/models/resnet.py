@@ -122,15 +122,16 @@ class ResNet18(ClientBatchedModel):
 
     def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
         from ..ops.fused import groupnorm_act
-        # OLSIM_CONV=custom routes 3x3 convs through the hand-written
-        # implicit-GEMM MFMA kernels (ops/csrc/client_conv.hip; v5
-        # pipeline fwd 99-137 TF/s, conv fwd+dgrad+wgrad at CK parity).
-        # The default stays MIOpen grouped conv while the full custom
-        # round is 406 vs 387 ms at C=250 (see profiles/ and
-        # docs/ROADMAP.md for the flip plan).
+        # Default: the hand-written client-batched implicit-GEMM MFMA
+        # conv kernels (ops/csrc/client_conv2.hip v6 padded family —
+        # mask-free gathers, XCD-aware tiling, parity-decomposed
+        # stride-2 dgrad).  Measured vs MIOpen grouped conv at C=250
+        # (tools/convbench3.py, profiles/): all-shape conv total
+        # 30.9 ms vs 57.1 ms (0.54x).  OLSIM_CONV=miopen forces the
+        # grouped-conv reference path.
         import os
         if x.is_cuda and x.dtype == torch.bfloat16 and \
-                os.environ.get("OLSIM_CONV", "") == "custom":
+                os.environ.get("OLSIM_CONV", "custom") == "custom":
             from ..ops.fused import hip_ops_available
             if hip_ops_available():
                 return self.forward_cbf(params, x)

@@ -98,6 +98,93 @@ def client_conv3x3(x: torch.Tensor, w: torch.Tensor,
     return _cpu_conv3x3(x, w, stride)
 
 
+# ---------------------------------------------------------------------------
+# 5x5 VALID convolution (LeNet family, client_conv5.hip)
+
+_NTAB_CACHE: dict = {}
+
+
+def _ntab(B: int, OH: int, OW: int, H: int, W: int,
+          device: torch.device) -> torch.Tensor:
+    """int32 [B*OH*OW] plane offsets b*H*W + oh*W + ow (gather base per
+    output position; the planes are not powers of two, so the kernels
+    index a table instead of dividing per element)."""
+    key = (B, OH, OW, H, W, str(device))
+    t = _NTAB_CACHE.get(key)
+    if t is None:
+        n = torch.arange(B * OH * OW, dtype=torch.int64)
+        b, q = n // (OH * OW), n % (OH * OW)
+        t = (b * H * W + (q // OW) * W + (q % OW)).to(torch.int32).to(device)
+        if len(_NTAB_CACHE) < 64:
+            _NTAB_CACHE[key] = t
+    return t
+
+
+def _cpu_conv5x5(x, w, b, relu):
+    # x [C, IC, B, H, W], w [C, OC, IC, 5, 5] valid conv
+    C, IC, B, H, W = x.shape
+    OC = w.shape[1]
+    xg = x.permute(2, 0, 1, 3, 4).reshape(B, C * IC, H, W)
+    y = F.conv2d(xg, w.reshape(C * OC, IC, 5, 5), b.reshape(C * OC),
+                 groups=C)
+    OH, OW = y.shape[-2:]
+    y = y.reshape(B, C, OC, OH, OW).permute(1, 2, 0, 3, 4).contiguous()
+    return F.relu(y) if relu else y
+
+
+class _Conv5x5Fn(torch.autograd.Function):
+    """Valid 5x5 conv + bias (+ReLU) on the MFMA kernels.  ReLU backward
+    uses the saved post-activation output (y > 0 mask)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, relu):
+        ops = load_hip_ops(required=True)
+        C, IC, B, H, W = x.shape
+        nt = _ntab(B, H - 4, W - 4, H, W, x.device)
+        y = ops.conv5x5_fwd(x, w, b, nt, relu)
+        ctx.save_for_backward(x, w, y)
+        ctx.relu = relu
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, y = ctx.saved_tensors
+        ops = load_hip_ops(required=True)
+        C, IC, B, H, W = x.shape
+        dy = dy.contiguous()
+        if ctx.relu:
+            dy = dy * (y > 0).to(dy.dtype)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dy_pad = F.pad(dy, (4, 4, 4, 4))
+            nt = _ntab(B, H, W, H + 4, W + 4, x.device)
+            dx = ops.conv5x5_dgrad(dy_pad, w, nt, H, W)
+        if ctx.needs_input_grad[1]:
+            nt = _ntab(B, H - 4, W - 4, H, W, x.device)
+            dw = ops.conv5x5_wgrad(x, dy, nt).to(w.dtype)
+        db = dy.sum(dim=(2, 3, 4)) if ctx.needs_input_grad[2] else None
+        return dx, dw, db, None
+
+
+def conv5x5_supported(x: torch.Tensor, w: torch.Tensor) -> bool:
+    if not (x.is_cuda and x.dtype == torch.bfloat16):
+        return False
+    C, IC, B, H, W = x.shape
+    if w.shape[1] > 16 or IC > 16:
+        return False
+    return (B * (H - 4) * (W - 4)) % 32 == 0
+
+
+def client_conv5x5(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
+                   relu: bool = False) -> torch.Tensor:
+    """y[C,OC,B,H-4,W-4] = valid conv5x5(x[C,IC,B,H,W], w[C,OC,IC,5,5])
+    + bias[C,OC], optional fused ReLU."""
+    if conv5x5_supported(x, w):
+        return _Conv5x5Fn.apply(x.contiguous(), w.contiguous(),
+                                b.contiguous(), relu)
+    return _cpu_conv5x5(x, w, b, relu)
+
+
 def client_conv1x1(x: torch.Tensor, w: torch.Tensor,
                    stride: int = 1) -> torch.Tensor:
     """1x1 conv = one batched GEMM: y[C,OC,n] = w[C,OC,IC] @ x[C,IC,n]."""

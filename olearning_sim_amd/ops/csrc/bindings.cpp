@@ -51,6 +51,17 @@ extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, float* dw,
                                     int C, int IC, int OC, int B, int H,
                                     int W, int stride, hipStream_t stream);
 
+extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
+                                void* y, const int* ntab, int C, int IC,
+                                int OC, int B, int H, int W, int relu,
+                                hipStream_t stream);
+extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
+                                  const int* ntab, int C, int IC, int OC,
+                                  int B, int H, int W, hipStream_t stream);
+extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, float* dw,
+                                  const int* ntab, int C, int IC, int OC,
+                                  int B, int H, int W, hipStream_t stream);
+
 extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const void* dy, void* dx, void* dres,
                                   const float* mean, const float* rstd,
@@ -286,6 +297,55 @@ at::Tensor conv3x3_dgrad_p(at::Tensor dyp, at::Tensor w, int64_t H, int64_t W,
   return dx;
 }
 
+// ---- 5x5 VALID family (client_conv5.hip, LeNet) -------------------------
+// ntab: int32 [B*OH*OW] plane offsets (b*H*W + oh*W + ow) built host-side
+
+at::Tensor conv5x5_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                       at::Tensor ntab, bool relu) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 5);
+  TORCH_CHECK(w.is_contiguous() && w.dim() == 5 && w.size(3) == 5);
+  TORCH_CHECK(ntab.is_cuda() && ntab.scalar_type() == at::kInt &&
+              ntab.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  int C = x.size(0), IC = x.size(1), B = x.size(2), H = x.size(3),
+      W = x.size(4), OC = w.size(1);
+  TORCH_CHECK(w.size(0) == C && w.size(2) == IC && OC <= 16);
+  TORCH_CHECK(ntab.numel() == (int64_t)B * (H - 4) * (W - 4));
+  auto y = at::empty({C, OC, B, H - 4, W - 4}, x.options());
+  ols_conv5x5_fwd(x.data_ptr(), w.data_ptr(), bias.contiguous().data_ptr(),
+                  y.data_ptr(), ntab.data_ptr<int>(), C, IC, OC, B, H, W,
+                  relu ? 1 : 0, at::cuda::getCurrentCUDAStream().stream());
+  return y;
+}
+
+// dyp: [C,OC,B,OH+8,OW+8] (pad 4); returns dx [C,IC,B,H,W]
+at::Tensor conv5x5_dgrad(at::Tensor dyp, at::Tensor w, at::Tensor ntab,
+                         int64_t H, int64_t W) {
+  TORCH_CHECK(dyp.is_cuda() && dyp.is_contiguous() && dyp.dim() == 5);
+  TORCH_CHECK(ntab.is_cuda() && ntab.scalar_type() == at::kInt);
+  int C = dyp.size(0), OC = dyp.size(1), B = dyp.size(2);
+  int IC = w.size(2);
+  TORCH_CHECK(IC <= 16 && ntab.numel() == (int64_t)B * H * W);
+  auto dx = at::empty({C, IC, B, H, W}, dyp.options());
+  ols_conv5x5_dgrad(dyp.data_ptr(), w.contiguous().data_ptr(), dx.data_ptr(),
+                    ntab.data_ptr<int>(), C, IC, OC, B, (int)H, (int)W,
+                    at::cuda::getCurrentCUDAStream().stream());
+  return dx;
+}
+
+at::Tensor conv5x5_wgrad(at::Tensor x, at::Tensor dy, at::Tensor ntab) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  TORCH_CHECK(ntab.is_cuda() && ntab.scalar_type() == at::kInt);
+  int C = x.size(0), IC = x.size(1), B = x.size(2), H = x.size(3),
+      W = x.size(4), OC = dy.size(1);
+  TORCH_CHECK(OC <= 16 && ((int64_t)B * (H - 4) * (W - 4)) % 32 == 0);
+  auto dw = at::empty({C, OC, IC, 5, 5}, x.options().dtype(at::kFloat));
+  ols_conv5x5_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
+                    ntab.data_ptr<int>(), C, IC, OC, B, H, W,
+                    at::cuda::getCurrentCUDAStream().stream());
+  return dw;
+}
+
 at::Tensor conv3x3_wgrad_p(at::Tensor xp, at::Tensor dy, int64_t stride) {
   TORCH_CHECK(xp.is_cuda() && xp.is_contiguous() && dy.is_contiguous());
   int C = xp.size(0), IC = xp.size(1), B = xp.size(2),
@@ -321,6 +381,9 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("conv3x3_fwd_p(Tensor xp, Tensor w, int stride) -> Tensor");
   m.def("conv3x3_dgrad_p(Tensor dyp, Tensor w, int H, int W, int stride) -> Tensor");
   m.def("conv3x3_wgrad_p(Tensor xp, Tensor dy, int stride) -> Tensor");
+  m.def("conv5x5_fwd(Tensor x, Tensor w, Tensor bias, Tensor ntab, bool relu) -> Tensor");
+  m.def("conv5x5_dgrad(Tensor dyp, Tensor w, Tensor ntab, int H, int W) -> Tensor");
+  m.def("conv5x5_wgrad(Tensor x, Tensor dy, Tensor ntab) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
@@ -336,4 +399,7 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("conv3x3_fwd_p", &conv3x3_fwd_p);
   m.impl("conv3x3_dgrad_p", &conv3x3_dgrad_p);
   m.impl("conv3x3_wgrad_p", &conv3x3_wgrad_p);
+  m.impl("conv5x5_fwd", &conv5x5_fwd);
+  m.impl("conv5x5_dgrad", &conv5x5_dgrad);
+  m.impl("conv5x5_wgrad", &conv5x5_wgrad);
 }

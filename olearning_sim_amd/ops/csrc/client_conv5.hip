@@ -1,0 +1,389 @@
+// Client-batched 5x5 VALID convolution (LeNet-5 family) as implicit
+// GEMM on MFMA — BASELINE configs 2 and 4 (LeNet CIFAR-10 / FedProx
+// churn) spend ~50% of GPU time in MIOpen naive_conv_* fallbacks at
+// thousands of groups (profiles/fedprox_churn_r01.md); these kernels
+// replace them.
+//
+// Channel counts are tiny (3->6, 6->16), so the GEMM M dimension is a
+// single 16-row MFMA fragment (BM=16) and the 4 waves of a workgroup
+// split the N (=B*OH*OW) dimension.  VALID convolution means every
+// gather address is in bounds — no masks at all; the n -> (b,oh,ow)
+// plane offset is a precomputed int32 table (ntab, built host-side
+// once per geometry) because the output planes are not powers of two
+// (28x28, 10x10).
+//
+// Views (per client c):
+//   fwd   : y[oc][n] = relu?(bias[oc] + sum_k W[oc][k] P[k][n]),
+//           k=(ic,dh,dw) over IC*25, P[k][n] = x[ic][b][oh+dh][ow+dw]
+//   dgrad : dX[ic][n=(b,ih,iw)] = sum_{k=(oc,dh,dw)} W[oc][ic][24-(5dh+dw)]
+//           * dy_pad4[oc][b][ih+dh][iw+dw]      (full correlation)
+//   wgrad : dW[oc][k] = sum_q dY[oc][q] P[k][q]  (fp32 out)
+//
+// K is padded to a BK=32 multiple with zero A-columns (B-side clamps
+// the channel index; the zero A rows annihilate the products).
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define CV5_BM 16
+#define CV5_BN 128
+#define CV5_BK 32
+#define CV5_PAD 8
+#define CV5_THREADS 256
+
+struct ConvGeom5 {
+  int B, H, W;          // input plane (or padded dy plane for dgrad)
+  int OH, OW;           // output plane
+  int IC, OC;           // contraction channels / output channels
+  int C, tiles_n;
+  int K, KP;            // true K and padded-to-32 K
+};
+
+__device__ __forceinline__ bool xcd_remap5(int lid, int C, int T,
+                                           int& c, int& tile) {
+  const int xcd = lid & 7;
+  const int s = lid >> 3;
+  const int grp = s / T;
+  tile = s - grp * T;
+  c = xcd + 8 * grp;
+  return c < C;
+}
+
+// ---------------------------------------------------------------------------
+// fwd (RELU templated): grid = xcd_blocks(C, tiles_n)
+template <bool RELU>
+__global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_fwd(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    const __hip_bfloat16* __restrict__ bias, __hip_bfloat16* __restrict__ y,
+    const int* __restrict__ ntab, ConvGeom5 g) {
+  int c, tile;
+  if (!xcd_remap5(blockIdx.x, g.C, g.tiles_n, c, tile)) return;
+  const int n0 = tile * CV5_BN;
+  extern __shared__ __attribute__((aligned(16))) short smem[];
+  short* a_lds = smem;                        // [16][KP]
+  short* bT_lds = smem + 16 * g.KP;           // [2][BN][BK+PAD]
+  const int N = g.B * g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const int64_t planeB = (int64_t)g.B * HW;
+  const ushort* xc =
+      reinterpret_cast<const ushort*>(x) + (int64_t)c * g.IC * planeB;
+  const ushort* wc =
+      reinterpret_cast<const ushort*>(w) + (int64_t)c * g.OC * g.K;
+  const __hip_bfloat16* bc = bias + (int64_t)c * g.OC;
+  __hip_bfloat16* yc = y + (int64_t)c * g.OC * N;
+
+  // stage A = W[0:16][0:KP] once (zero padding rows/cols)
+  for (int i = threadIdx.x; i < 16 * g.KP; i += CV5_THREADS) {
+    int m = i / g.KP, k = i - m * g.KP;
+    a_lds[i] = (m < g.OC && k < g.K) ? (short)wc[(int64_t)m * g.K + k]
+                                     : (short)0;
+  }
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;          // wave w: n-subtiles {2w,2w+1}
+  f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+
+  const int kk = threadIdx.x % CV5_BK;
+  const int nn0 = (threadIdx.x / CV5_BK) * (CV5_BN / 8);
+  ushort breg[CV5_BN / 8];
+  const int LDB = CV5_BK + CV5_PAD;
+
+  auto gather = [&](int k0) {
+    const int k = k0 + kk;
+    const int kc = min(k, g.K - 1);           // K-tail: A is zero there
+    const int ic = kc / 25, r = kc - ic * 25;
+    const int dh = r / 5, dw = r - dh * 5;
+    const ushort* plane = xc + (int64_t)ic * planeB + dh * g.W + dw;
+#pragma unroll
+    for (int j = 0; j < CV5_BN / 8; ++j) {
+      int n = min(n0 + nn0 + j, N - 1);
+      breg[j] = plane[ntab[n]];
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < CV5_BN / 8; ++j)
+      bT_lds[buf * CV5_BN * LDB + (nn0 + j) * LDB + kk] = (short)breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int k0 = 0; k0 < g.KP; k0 += CV5_BK) {
+    __syncthreads();
+    if (k0 + CV5_BK < g.KP) gather(k0 + CV5_BK);
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(
+        &a_lds[(lane & 15) * g.KP + k0 + 8 * (lane >> 4)]);
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      int sub = wave * 2 + t;
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur * CV5_BN * LDB + (sub * 16 + (lane & 15)) * LDB
+                  + 8 * (lane >> 4)]);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+    }
+    if (k0 + CV5_BK < g.KP) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    int n = n0 + (wave * 2 + t) * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = (lane >> 4) * 4 + r;
+      if (m < g.OC) {
+        float v = acc[t][r] + to_f32(bc[m]);
+        if (RELU) v = fmaxf(v, 0.f);
+        yc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dgrad: dy_pad4 [C][OC][B][OH+8][OW+8]; output plane = (H,W) of x.
+// g: B; H,W := padded dy plane dims; OH,OW := dX plane; IC := dX
+// channels (M dim), OC := contracted channels; K = OC*25.
+__global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_dgrad(
+    const __hip_bfloat16* __restrict__ dyp, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ dx, const int* __restrict__ ntab,
+    ConvGeom5 g) {
+  int c, tile;
+  if (!xcd_remap5(blockIdx.x, g.C, g.tiles_n, c, tile)) return;
+  const int n0 = tile * CV5_BN;
+  extern __shared__ __attribute__((aligned(16))) short smem[];
+  short* a_lds = smem;                        // [16][KP]
+  short* bT_lds = smem + 16 * g.KP;
+  const int N = g.B * g.OH * g.OW;            // dX positions
+  const int HW = g.H * g.W;                   // padded dy plane
+  const int64_t planeB = (int64_t)g.B * HW;
+  const ushort* dyc =
+      reinterpret_cast<const ushort*>(dyp) + (int64_t)c * g.OC * planeB;
+  const ushort* wc =
+      reinterpret_cast<const ushort*>(w) + (int64_t)c * g.OC * g.IC * 25;
+  __hip_bfloat16* dxc = dx + (int64_t)c * g.IC * N;
+
+  // stage A once: A[ic][k=(oc,dh,dw)] = W[oc][ic][24 - (5dh+dw)]
+  for (int i = threadIdx.x; i < 16 * g.KP; i += CV5_THREADS) {
+    int m = i / g.KP, k = i - m * g.KP;
+    short v = 0;
+    if (m < g.IC && k < g.K) {
+      int oc = k / 25, r = k - oc * 25;
+      v = (short)wc[((int64_t)oc * g.IC + m) * 25 + (24 - r)];
+    }
+    a_lds[i] = v;
+  }
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  const int kk = threadIdx.x % CV5_BK;
+  const int nn0 = (threadIdx.x / CV5_BK) * (CV5_BN / 8);
+  ushort breg[CV5_BN / 8];
+  const int LDB = CV5_BK + CV5_PAD;
+
+  auto gather = [&](int k0) {
+    const int k = k0 + kk;
+    const int kc = min(k, g.K - 1);
+    const int oc = kc / 25, r = kc - oc * 25;
+    const int dh = r / 5, dw = r - dh * 5;
+    const ushort* plane = dyc + (int64_t)oc * planeB + dh * g.W + dw;
+#pragma unroll
+    for (int j = 0; j < CV5_BN / 8; ++j) {
+      int n = min(n0 + nn0 + j, N - 1);
+      breg[j] = plane[ntab[n]];
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < CV5_BN / 8; ++j)
+      bT_lds[buf * CV5_BN * LDB + (nn0 + j) * LDB + kk] = (short)breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int k0 = 0; k0 < g.KP; k0 += CV5_BK) {
+    __syncthreads();
+    if (k0 + CV5_BK < g.KP) gather(k0 + CV5_BK);
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(
+        &a_lds[(lane & 15) * g.KP + k0 + 8 * (lane >> 4)]);
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      int sub = wave * 2 + t;
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur * CV5_BN * LDB + (sub * 16 + (lane & 15)) * LDB
+                  + 8 * (lane >> 4)]);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+    }
+    if (k0 + CV5_BK < g.KP) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    int n = n0 + (wave * 2 + t) * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = (lane >> 4) * 4 + r;
+      if (m < g.IC)
+        dxc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[t][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad: dW[oc][k=(ic,dh,dw)] (fp32) = sum_q dY[oc][q] x[ic][..q..];
+// A = dY rows direct from global (contiguous, NN%32==0), B = x gather
+// with per-column offsets hoisted; per q-step only ntab[q] changes.
+// g: H,W := x plane; OH,OW := y plane; K := NN (reduction), KP unused.
+__global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
+    float* __restrict__ dw, const int* __restrict__ ntab, ConvGeom5 g) {
+  int c, tile;
+  if (!xcd_remap5(blockIdx.x, g.C, g.tiles_n, c, tile)) return;
+  const int n0 = tile * CV5_BN;               // over IC*25
+  __shared__ short bT_lds[2][CV5_BN * (CV5_BK + CV5_PAD)];
+  const int K25 = g.IC * 25;
+  const int NN = g.B * g.OH * g.OW;
+  const int HW = g.H * g.W;
+  const int64_t planeB = (int64_t)g.B * HW;
+  const ushort* xc =
+      reinterpret_cast<const ushort*>(x) + (int64_t)c * g.IC * planeB;
+  const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
+  float* dwc = dw + (int64_t)c * g.OC * K25;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int arow = lane & 15;                 // dY row (oc), BM=16
+  const __hip_bfloat16* dyrow = dyc + (int64_t)min(arow, g.OC - 1) * NN;
+  const bool arow_ok = arow < g.OC;
+
+  f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  const int qq = threadIdx.x % CV5_BK;
+  const int nn0 = (threadIdx.x / CV5_BK) * (CV5_BN / 8);
+  ushort breg[CV5_BN / 8];
+  const int LDB = CV5_BK + CV5_PAD;
+
+  int64_t off[CV5_BN / 8];
+#pragma unroll
+  for (int j = 0; j < CV5_BN / 8; ++j) {
+    int k = min(n0 + nn0 + j, K25 - 1);
+    int ic = k / 25, r = k - ic * 25;
+    int dh = r / 5, dw2 = r - dh * 5;
+    off[j] = (int64_t)ic * planeB + dh * g.W + dw2;
+  }
+
+  auto gather = [&](int q0) {
+    int q = min(q0 + qq, NN - 1);
+    const ushort* base = xc + ntab[q];
+#pragma unroll
+    for (int j = 0; j < CV5_BN / 8; ++j) breg[j] = base[off[j]];
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < CV5_BN / 8; ++j)
+      bT_lds[buf][(nn0 + j) * LDB + qq] = (short)breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int q0 = 0; q0 < NN; q0 += CV5_BK) {
+    __syncthreads();
+    if (q0 + CV5_BK < NN) gather(q0 + CV5_BK);
+    bf16x8 a;
+    {
+      uint4 av = *reinterpret_cast<const uint4*>(dyrow + q0 + 8 * (lane >> 4));
+      a = *reinterpret_cast<const bf16x8*>(&av);
+      if (!arow_ok) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) a[e] = 0;
+      }
+    }
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      int sub = wave * 2 + t;
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur][(sub * 16 + (lane & 15)) * LDB + 8 * (lane >> 4)]);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+    }
+    if (q0 + CV5_BK < NN) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+    int k = n0 + (wave * 2 + t) * 16 + (lane & 15);
+    if (k >= K25) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        dwc[(int64_t)m * K25 + k] = acc[t][r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+static inline int cdiv5(int a, int b) { return (a + b - 1) / b; }
+static inline int xcd_blocks5(int C, int T) { return ((C + 7) / 8 * 8) * T; }
+
+extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
+                                void* y, const int* ntab, int C, int IC,
+                                int OC, int B, int H, int W, int relu,
+                                hipStream_t stream) {
+  ConvGeom5 g;
+  g.B = B; g.H = H; g.W = W; g.OH = H - 4; g.OW = W - 4;
+  g.IC = IC; g.OC = OC; g.C = C;
+  g.K = IC * 25; g.KP = cdiv5(g.K, CV5_BK) * CV5_BK;
+  g.tiles_n = cdiv5(B * g.OH * g.OW, CV5_BN);
+  size_t lds = (16 * g.KP + 2 * CV5_BN * (CV5_BK + CV5_PAD)) * sizeof(short);
+  dim3 grid(xcd_blocks5(C, g.tiles_n));
+  if (relu)
+    hipLaunchKernelGGL((k_conv5x5_fwd<true>), grid, dim3(CV5_THREADS), lds,
+                       stream, (const __hip_bfloat16*)x,
+                       (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,
+                       (__hip_bfloat16*)y, ntab, g);
+  else
+    hipLaunchKernelGGL((k_conv5x5_fwd<false>), grid, dim3(CV5_THREADS), lds,
+                       stream, (const __hip_bfloat16*)x,
+                       (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,
+                       (__hip_bfloat16*)y, ntab, g);
+}
+
+// H, W: the dX plane; dyp is [C][OC][B][OH+8][OW+8] with OH=H-4
+extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
+                                  const int* ntab, int C, int IC, int OC,
+                                  int B, int H, int W, hipStream_t stream) {
+  ConvGeom5 g;
+  g.B = B; g.H = (H - 4) + 8; g.W = (W - 4) + 8;   // padded dy plane
+  g.OH = H; g.OW = W;                              // dX plane
+  g.IC = IC; g.OC = OC; g.C = C;
+  g.K = OC * 25; g.KP = cdiv5(g.K, CV5_BK) * CV5_BK;
+  g.tiles_n = cdiv5(B * H * W, CV5_BN);
+  size_t lds = (16 * g.KP + 2 * CV5_BN * (CV5_BK + CV5_PAD)) * sizeof(short);
+  dim3 grid(xcd_blocks5(C, g.tiles_n));
+  hipLaunchKernelGGL(k_conv5x5_dgrad, grid, dim3(CV5_THREADS), lds, stream,
+                     (const __hip_bfloat16*)dyp, (const __hip_bfloat16*)w,
+                     (__hip_bfloat16*)dx, ntab, g);
+}
+
+extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, float* dw,
+                                  const int* ntab, int C, int IC, int OC,
+                                  int B, int H, int W, hipStream_t stream) {
+  ConvGeom5 g;
+  g.B = B; g.H = H; g.W = W; g.OH = H - 4; g.OW = W - 4;
+  g.IC = IC; g.OC = OC; g.C = C;
+  g.K = 0; g.KP = 0;
+  g.tiles_n = cdiv5(IC * 25, CV5_BN);
+  dim3 grid(xcd_blocks5(C, g.tiles_n));
+  hipLaunchKernelGGL(k_conv5x5_wgrad, grid, dim3(CV5_THREADS), 0, stream,
+                     (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
+                     dw, ntab, g);
+}

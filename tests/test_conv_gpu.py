@@ -177,3 +177,63 @@ def test_resnet_cbf_bench_speed():
     dt = (time.perf_counter() - t0) / 3
     print(f"\ncbf fwd+bwd C={C} B={B}: {dt*1000:.1f} ms")
     assert dt < 10.0
+
+
+CONV5_SHAPES = [
+    # (C, IC, OC, B, H)
+    (3, 3, 6, 8, 32),     # LeNet conv1
+    (2, 6, 16, 8, 14),    # LeNet conv2
+    (9, 3, 6, 16, 32),    # C not a multiple of 8 (XCD remap tail)
+]
+
+
+@pytest.mark.parametrize("C,IC,OC,B,H", CONV5_SHAPES)
+@pytest.mark.parametrize("relu", [False, True])
+def test_conv5x5_matches_reference(C, IC, OC, B, H, relu):
+    """Valid 5x5 conv fwd+bwd (fused bias/ReLU) vs fp32 torch."""
+    from olearning_sim_amd.ops.conv import client_conv5x5, _cpu_conv5x5
+    g = torch.Generator().manual_seed(11)
+    x0 = torch.randn(C, IC, B, H, H, generator=g) * 0.5
+    w0 = torch.randn(C, OC, IC, 5, 5, generator=g) * 0.1
+    b0 = torch.randn(C, OC, generator=g) * 0.1
+
+    xg = x0.to(torch.bfloat16).cuda().requires_grad_(True)
+    wg = w0.to(torch.bfloat16).cuda().requires_grad_(True)
+    bg = b0.to(torch.bfloat16).cuda().requires_grad_(True)
+    y = client_conv5x5(xg, wg, bg, relu=relu)
+    dy = torch.randn(y.shape, generator=g) * 0.1
+    y.backward(dy.to(torch.bfloat16).cuda())
+
+    xr = x0.clone().requires_grad_(True)
+    wr = w0.clone().requires_grad_(True)
+    br = b0.clone().requires_grad_(True)
+    yr = _cpu_conv5x5(xr, wr, br, relu)
+    yr.backward(dy)
+
+    torch.testing.assert_close(y.float().cpu(), yr,
+                               atol=0.05 * (IC ** 0.5) + 0.02, rtol=5e-2)
+    torch.testing.assert_close(xg.grad.float().cpu(), xr.grad,
+                               atol=0.05 * (OC ** 0.5) + 0.02, rtol=8e-2)
+    torch.testing.assert_close(wg.grad.float().cpu(), wr.grad,
+                               atol=2e-2 * (B * H * H) ** 0.5 * 0.1 + 2e-2,
+                               rtol=8e-2)
+    torch.testing.assert_close(bg.grad.float().cpu(), br.grad,
+                               atol=0.5, rtol=5e-2)
+
+
+def test_lenet_cbf_matches_grouped_path():
+    """LeNet's conv5x5 MFMA fast path == grouped reference path."""
+    from olearning_sim_amd.models import build_model
+    from olearning_sim_amd.engine.client_manager import (FlatParams,
+                                                         replicate_params)
+    m = build_model("lenet")
+    gen = torch.Generator().manual_seed(0)
+    gp = {k: v.cuda() for k, v in m.init_global(generator=gen).items()}
+    master = FlatParams(gp)
+    C, B = 5, 8
+    params = replicate_params(master.cast(torch.bfloat16), C)
+    x = torch.randn(C, B, 3, 32, 32, generator=gen).to(torch.bfloat16).cuda()
+    fast = m.forward_cbf(params, x)
+    params32 = {k: v.detach().float().cpu() for k, v in params.items()}
+    ref = m.forward(params32, x.float().cpu())
+    torch.testing.assert_close(fast.float().cpu(), ref, atol=0.15, rtol=0.08)
