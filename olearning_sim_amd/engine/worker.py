@@ -62,11 +62,19 @@ def main(argv=None) -> int:
     job = EngineJob(**{k: v for k, v in spec.items() if k in known})
     job.device = ctx.device
 
-    eng = LogicalEngine(job, dist_ctx=ctx if ctx.enabled else None)
+    # rank 0 collects the shaped per-round result rows (the
+    # logical_round/logical_result rows the in-process runner writes to
+    # the task table; the group monitor replays them on completion)
+    round_rows = []
+    sink = round_rows.append if ctx.rank == 0 else None
+    eng = LogicalEngine(job, dist_ctx=ctx if ctx.enabled else None,
+                        result_sink=sink)
     out = eng.run()
     out["total_clients"] = total_clients
     if ctx.rank == 0:
-        line = json.dumps({k: v for k, v in out.items() if k != "records"})
+        out["round_rows"] = round_rows
+        line = json.dumps({k: v for k, v in out.items()
+                           if k not in ("records", "round_rows")})
         print(line)
         if args.result_json:
             with open(args.result_json, "w") as f:

@@ -61,14 +61,15 @@ class SimulatorSession:
                            if svc in (0, 3) else None)
         self.performance_mgr = (PerformanceManager(db("perf.sqlite"))
                                 if svc in (0, 4) else None)
-        self.cluster_mgr = NodeClusterManager() if svc == 0 else None
+        self.cluster_mgr = NodeClusterManager() if svc in (0, 1) else None
         self.task_mgr = None
         if svc in (0, 1):
             table = TaskTableRepo(db("taskmgr.sqlite"))
             runner = TaskRunner(
                 table, device=device,
                 checkpoint_dir=os.path.join(self.data_dir, "checkpoints"),
-                deviceflow=self.deviceflow, perf=self.performance_mgr)
+                deviceflow=self.deviceflow, perf=self.performance_mgr,
+                cluster=self.cluster_mgr)
             self.task_mgr = TaskManager(
                 table=table, resource_mgr=self.resource_mgr, runner=runner,
                 deviceflow=self.deviceflow, timers=self.config.timers(),

@@ -181,12 +181,18 @@ def _segment_dynamic_nums(task: TaskConfig,
 
 class TaskRunner:
     def __init__(self, table: TaskTableRepo, device: str = "cpu",
-                 checkpoint_dir: str = "", deviceflow=None, perf=None):
+                 checkpoint_dir: str = "", deviceflow=None, perf=None,
+                 cluster=None):
         self.table = table
         self.device = device
         self.checkpoint_dir = checkpoint_dir
         self.deviceflow = deviceflow   # deviceflow service facade (optional)
         self.perf = perf               # PerformanceManager (optional)
+        # cluster: NodeClusterManager — tasks whose train operator asks
+        # for num_gpus > 1 run as a one-process-per-GPU worker group
+        # (reference TaskRunner submits to the Ray cluster fabric,
+        # task_runner.py:41-87)
+        self.cluster = cluster
         self.jobs: Dict[str, JobHandle] = {}
         self.task_jobs: Dict[str, List[str]] = {}   # task_id -> job ids
         self.log = Logger.shared()
@@ -300,11 +306,30 @@ class TaskRunner:
             sides.append("device_simulation")
         return sides
 
+    @staticmethod
+    def _group_world_size(task: TaskConfig) -> int:
+        """Worker-group size from the train operator's params JSON
+        (operator-facing knob ``num_gpus``; 0/1 = in-process)."""
+        for op in task.operatorflow.operators:
+            try:
+                params = json.loads(op.logical_simulation.operator_params
+                                    or "{}")
+            except Exception:
+                continue
+            n = params.get("num_gpus", 0)
+            if isinstance(n, int) and n > 1:
+                return n
+        return 1
+
     # -- logical side ----------------------------------------------------
     def _submit_logical(self, task: TaskConfig,
                         allocations: List[DataAllocation]) -> str:
         job = engine_job_from_task(task, allocations, self.device,
                                    self.checkpoint_dir)
+        world = self._group_world_size(task)
+        has_script = any(kind == "script" for _, kind in job.operators)
+        if world > 1 and self.cluster is not None and not has_script:
+            return self._submit_logical_group(task, job, world)
         script_ops = self._stage_script_operators(task, job)
         job_id = f"olsjob_{uuid.uuid4().hex[:12]}"
         handle = JobHandle(job_id, "logical")
@@ -351,6 +376,103 @@ class TaskRunner:
             self.jobs[job_id] = handle
             self.task_jobs.setdefault(task.task_id, []).append(job_id)
         handle.thread.start()
+        return job_id
+
+    def _submit_logical_group(self, task: TaskConfig, job: EngineJob,
+                              world: int) -> str:
+        """Launch the logical simulation as a one-process-per-GPU worker
+        group through the NodeClusterManager (the execution fabric; one
+        rank per GPU over RCCL, gloo on CPU).  A monitor thread polls
+        the group and replays rank 0's per-round result rows into the
+        task table on completion, so status fusion sees the same
+        logical_round/logical_result rows the in-process path writes."""
+        import dataclasses
+        import tempfile
+
+        if self.device.startswith("cuda"):
+            import torch
+            avail = torch.cuda.device_count()
+            if avail > 0:
+                world = min(world, avail)
+
+        job_id = f"olsgrp_{uuid.uuid4().hex[:12]}"
+        handle = JobHandle(job_id, "logical")
+        work_dir = os.path.join(self.checkpoint_dir or tempfile.gettempdir(),
+                                f"group_{task.task_id}_{job_id}")
+        os.makedirs(work_dir, exist_ok=True)
+        spec_path = os.path.join(work_dir, "job.json")
+        result_path = os.path.join(work_dir, "result.json")
+        spec = dataclasses.asdict(job)
+        spec["device"] = ""           # each rank picks cuda:LOCAL_RANK / cpu
+        with open(spec_path, "w") as f:
+            json.dump(spec, f)
+
+        from ..cluster.node_manager import WorkerGroupSpec
+        cname = f"grp_{job_id}"
+        created = self.cluster.create_cluster(WorkerGroupSpec(
+            name=cname, replicas=world,
+            entry_module="olearning_sim_amd.engine.worker",
+            args=["--job-json", spec_path, "--result-json", result_path]))
+        if not created:
+            handle.status = JobStatus.FAILED
+            handle.error = f"worker group {cname} already exists"
+            with self._lock:
+                self.jobs[job_id] = handle
+                self.task_jobs.setdefault(task.task_id, []).append(job_id)
+            return job_id
+
+        def monitor():
+            handle.status = JobStatus.RUNNING
+            try:
+                while True:
+                    info = self.cluster.get_cluster(cname)
+                    state = info["status"] if info else "failed"
+                    if handle.stop_requested:
+                        self.cluster.delete_cluster(cname)
+                        handle.status = JobStatus.STOPPED
+                        return
+                    if state in ("succeeded", "failed"):
+                        break
+                    time.sleep(0.3)
+                out = None
+                if os.path.exists(result_path):
+                    try:
+                        with open(result_path) as f:
+                            out = json.load(f)
+                    except Exception:
+                        out = None
+                if out is not None:
+                    for row in out.get("round_rows", []):
+                        self.table.set_items(
+                            task.task_id,
+                            logical_round=row["logical_round"],
+                            logical_operator=row["logical_operator"],
+                            logical_result=json.dumps(row["logical_result"]))
+                    failed = any(r.get("round_failed")
+                                 for r in out.get("round_rows", []))
+                else:
+                    failed = True
+                if state == "succeeded" and not failed:
+                    handle.status = JobStatus.SUCCEEDED
+                else:
+                    handle.status = JobStatus.FAILED
+                    if out is None:
+                        handle.error = "worker group produced no result"
+                self.cluster.delete_cluster(cname)
+            except Exception as e:
+                handle.error = str(e)
+                handle.status = JobStatus.FAILED
+                self.log.error(task.task_id, "TaskMgr", "runner",
+                               f"worker group failed: {e}")
+
+        handle.thread = threading.Thread(target=monitor, daemon=True)
+        with self._lock:
+            self.jobs[job_id] = handle
+            self.task_jobs.setdefault(task.task_id, []).append(job_id)
+        handle.thread.start()
+        self.log.info(task.task_id, "TaskMgr", "runner",
+                      f"logical simulation on worker group {cname} "
+                      f"(world={world})")
         return job_id
 
     # -- device side (simulated phone farm) -----------------------------

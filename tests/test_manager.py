@@ -466,3 +466,40 @@ def test_orphaned_running_task_fails_on_restart(tmp_path):
     ok, msg = mgr2.submit_task(task_json(task_id="t_orph", rounds=1))
     assert ok, msg
     mgr2.shutdown()
+
+
+def test_submit_routes_multi_gpu_task_to_worker_group(tmp_path):
+    """A task whose train operator asks num_gpus > 1 runs as a worker
+    group (one process per device, torchrun; gloo on CPU) launched via
+    the NodeClusterManager, and its result rows fuse to SUCCEEDED
+    (reference TaskRunner submits to the Ray fabric,
+    task_runner.py:41-87)."""
+    from olearning_sim_amd.cluster.node_manager import NodeClusterManager
+    table = TaskTableRepo(":memory:")
+    res = ResourceManager(":memory:", totals={"cpu": 8.0, "mem": 64.0,
+                                              "gpu": 0, "hbm_gb": 0})
+    cluster = NodeClusterManager()
+    runner = TaskRunner(table, device="cpu",
+                        checkpoint_dir=str(tmp_path), cluster=cluster)
+    mgr = TaskManager(table=table, resource_mgr=res, runner=runner)
+    try:
+        ok, msg = mgr.submit_task(task_json(
+            task_id="t_grp", rounds=2, clients=6,
+            params={"num_gpus": 2}))
+        assert ok, msg
+        assert mgr.step_schedule() == "t_grp"
+        st = wait_terminal(mgr, "t_grp", timeout=180.0)
+        assert st == TaskStatus.SUCCEEDED, \
+            f"status={st} err={[ (h.job_id, h.error) for h in runner.jobs.values() ]}"
+        jids = runner.task_jobs["t_grp"]
+        assert any(j.startswith("olsgrp_") for j in jids)
+        # rank 0's round rows were replayed into the table
+        assert int(table.get_item_value("t_grp", "logical_round")) == 2
+        result = json.loads(table.get_item_value("t_grp", "logical_result"))
+        entry = result["logical_result"][0]["simulation_target"]
+        assert sum(entry["success_num"]) == 6      # both ranks' shards
+        # the group was reaped
+        assert cluster.list_clusters() == []
+    finally:
+        mgr.shutdown()
+        cluster.shutdown()
