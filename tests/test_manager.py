@@ -324,7 +324,9 @@ def test_interrupt_overdue_running_task():
     interrupt_running_time (reference task_manager.py:1150-1200)."""
     mgr = make_manager()
     mgr.timers["interrupt_running_time"] = 0.0
-    mgr.submit_task(task_json(task_id="t_long", rounds=50))
+    # enough rounds that the engine cannot finish before the watchdog
+    # fires even on a loaded host (the stop lands within a round or two)
+    mgr.submit_task(task_json(task_id="t_long", rounds=5000))
     assert mgr.step_schedule() == "t_long"
     time.sleep(0.05)
     assert "t_long" in mgr.step_interrupt()
