@@ -132,21 +132,34 @@ class LogicalEngine:
 
     # ------------------------------------------------------------------
     def _chunk_size(self, cohort: int) -> int:
+        """Clients per chunk.  Computed ONCE per engine (cached): the
+        free-memory probe must not move between rounds or kernel shapes
+        churn (shape-specialised kernels re-tune, fixed-shape padding
+        re-pads a different tail).  Chunks are equalised so a split
+        cohort never pads a tiny tail to a full chunk."""
+        cached = getattr(self, "_chunk_cache", None)
+        if cached is not None and cached[0] == cohort:
+            return cached[1]
         if self.job.chunk_clients > 0:
-            return min(self.job.chunk_clients, cohort)
-        if self.device.type != "cuda":
-            return min(cohort, 64)
-        # auto: bound replica+grad+activation memory to a fraction of HBM
-        p = self.master.numel()
-        bytes_per_client = p * self.dtype.itemsize * 2  # weights + grad
-        bytes_per_client += (self.job.batch_size
-                             * self.model.act_elems_per_sample
-                             * self.dtype.itemsize)
-        # budget against memory actually free NOW (total_memory would
-        # over-commit the moment anything else holds HBM on this device)
-        free, total = torch.cuda.mem_get_info(self.device)
-        budget = int(min(free * 0.8, total * 0.5))
-        return max(1, min(cohort, budget // max(1, bytes_per_client)))
+            chunk = min(self.job.chunk_clients, cohort)
+        elif self.device.type != "cuda":
+            chunk = min(cohort, 64)
+        else:
+            # bound replica+grad+activation memory by HBM actually free
+            # at engine start (torch's cache may already hold the rest)
+            p = self.master.numel()
+            bytes_per_client = p * self.dtype.itemsize * 2  # weights + grad
+            bytes_per_client += (self.job.batch_size
+                                 * self.model.act_elems_per_sample
+                                 * self.dtype.itemsize)
+            free, total = torch.cuda.mem_get_info(self.device)
+            free += torch.cuda.memory_reserved(self.device)
+            budget = int(min(free * 0.8, total * 0.5))
+            max_chunk = max(1, min(cohort, budget // max(1, bytes_per_client)))
+            n_chunks = (cohort + max_chunk - 1) // max_chunk
+            chunk = (cohort + n_chunks - 1) // n_chunks
+        self._chunk_cache = (cohort, chunk)
+        return chunk
 
     def select_cohort(self, round_idx: int) -> torch.Tensor:
         """Deterministic rotating window over this rank's population

@@ -51,11 +51,31 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_fwd(
 
   __shared__ float red[2][OLS_THREADS / WAVE];
   float s1 = 0.f, s2 = 0.f;
-  for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)(i / HW) * pstride + i % HW);
-    float v = to_f32(xg[idx]);
-    s1 += v;
-    s2 += v * v;
+  // vector path: 8 elements (16 B) per lane per iteration — scalar bf16
+  // loads measure 2-2.5x slower on gfx950 (cdna_hip_programming.md G13);
+  // HW is a multiple of 8 for every model plane, so a vector never
+  // crosses a plane (LAYOUT 1) or channel boundary
+  const bool vec = (HW % 8) == 0;
+  if (vec) {
+    const int nv = n / 8;
+    for (int v8 = threadIdx.x; v8 < nv; v8 += blockDim.x) {
+      int i = v8 * 8;
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)(i / HW) * pstride + i % HW);
+      Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xg[idx]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float v = to_f32(px.v[e]);
+        s1 += v;
+        s2 += v * v;
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)(i / HW) * pstride + i % HW);
+      float v = to_f32(xg[idx]);
+      s1 += v;
+      s2 += v * v;
+    }
   }
   s1 = wave_sum(s1);
   s2 = wave_sum(s2);
@@ -71,14 +91,37 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_fwd(
 
   const T* gam = gamma + (int64_t)c * ch + g * cg;
   const T* bet = beta + (int64_t)c * ch + g * cg;
-  for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    int chan = i / HW;                       // channel within the group
-    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
-    float v = (to_f32(xg[idx]) - mean) * rstd;
-    v = v * to_f32(gam[chan]) + to_f32(bet[chan]);
-    if (HAS_RES) v += to_f32(rg[idx]);
-    if (RELU) v = fmaxf(v, 0.f);
-    yg[idx] = from_f32<T>(v);
+  if (vec) {
+    const int nv = n / 8;
+    for (int v8 = threadIdx.x; v8 < nv; v8 += blockDim.x) {
+      int i = v8 * 8;
+      int chan = i / HW;
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+      Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xg[idx]);
+      Pack<T, 8> pr;
+      if (HAS_RES) pr = *reinterpret_cast<const Pack<T, 8>*>(&rg[idx]);
+      Pack<T, 8> py;
+      const float ga = to_f32(gam[chan]), be = to_f32(bet[chan]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float v = (to_f32(px.v[e]) - mean) * rstd;
+        v = v * ga + be;
+        if (HAS_RES) v += to_f32(pr.v[e]);
+        if (RELU) v = fmaxf(v, 0.f);
+        py.v[e] = from_f32<T>(v);
+      }
+      *reinterpret_cast<Pack<T, 8>*>(&yg[idx]) = py;
+    }
+  } else {
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+      int chan = i / HW;                     // channel within the group
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+      float v = (to_f32(xg[idx]) - mean) * rstd;
+      v = v * to_f32(gam[chan]) + to_f32(bet[chan]);
+      if (HAS_RES) v += to_f32(rg[idx]);
+      if (RELU) v = fmaxf(v, 0.f);
+      yg[idx] = from_f32<T>(v);
+    }
   }
 }
 
@@ -123,18 +166,71 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_bwd(
   }
   __syncthreads();
 
+  const int lane = threadIdx.x & (WAVE - 1);
   float s1 = 0.f, s2 = 0.f;
-  for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    int chan = i / HW;
-    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
-    float grad = to_f32(dyg_in[idx]);
-    if (RELU) grad = to_f32(yg[idx]) > 0.f ? grad : 0.f;
-    float xhat = (to_f32(xg[idx]) - mean) * rstd;
-    float dxhat = grad * to_f32(gam[chan]);
-    s1 += dxhat;
-    s2 += dxhat * xhat;
-    atomicAdd(&ch_dg[chan], grad * xhat);
-    atomicAdd(&ch_db[chan], grad);
+  // Vector path (HW % 8 == 0, every model plane): 16 B loads per lane,
+  // and the per-channel dgamma/dbeta partials are reduced across the
+  // lanes that share a channel BEFORE one LDS atomic — the previous
+  // per-element atomics serialised up to 256 ways on one address when
+  // HW >= the block span.  lpc = lanes per channel (pow2).
+  const bool vec = (HW % 8) == 0;
+  if (vec) {
+    const int nv = n / 8;
+    const int hv = HW / 8;                     // vectors per channel-plane
+    const int lpc = min(WAVE, hv);             // pow2 (HW, 8 are pow2)
+    // whole waves iterate together (the shuffle reduce needs every lane
+    // of a channel group present; groups of lpc consecutive vectors are
+    // always fully active or fully inactive since lpc divides nv)
+    for (int v8 = threadIdx.x; v8 - lane < nv; v8 += blockDim.x) {
+      const bool active = v8 < nv;
+      int i = active ? v8 * 8 : 0;
+      int chan = i / HW;
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+      float dg = 0.f, db = 0.f;
+      if (active) {
+        Pack<T, 8> pdy = *reinterpret_cast<const Pack<T, 8>*>(&dyg_in[idx]);
+        Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xg[idx]);
+        Pack<T, 8> py;
+        if (RELU) py = *reinterpret_cast<const Pack<T, 8>*>(&yg[idx]);
+        const float ga = to_f32(gam[chan]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float grad = to_f32(pdy.v[e]);
+          if (RELU) grad = to_f32(py.v[e]) > 0.f ? grad : 0.f;
+          float xhat = (to_f32(px.v[e]) - mean) * rstd;
+          float dxhat = grad * ga;
+          s1 += dxhat;
+          s2 += dxhat * xhat;
+          dg += grad * xhat;
+          db += grad;
+        }
+      }
+      // reduce over the lpc lanes sharing this channel, one atomic per
+      // group (distinct channels land on distinct LDS addresses)
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1)
+        if (off < lpc) {
+          dg += __shfl_xor(dg, off, WAVE);
+          db += __shfl_xor(db, off, WAVE);
+        }
+      if (active && (lane & (lpc - 1)) == 0) {
+        atomicAdd(&ch_dg[chan], dg);
+        atomicAdd(&ch_db[chan], db);
+      }
+    }
+  } else {
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+      int chan = i / HW;
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+      float grad = to_f32(dyg_in[idx]);
+      if (RELU) grad = to_f32(yg[idx]) > 0.f ? grad : 0.f;
+      float xhat = (to_f32(xg[idx]) - mean) * rstd;
+      float dxhat = grad * to_f32(gam[chan]);
+      s1 += dxhat;
+      s2 += dxhat * xhat;
+      atomicAdd(&ch_dg[chan], grad * xhat);
+      atomicAdd(&ch_db[chan], grad);
+    }
   }
   s1 = wave_sum(s1);
   s2 = wave_sum(s2);
@@ -145,15 +241,41 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_bwd(
   for (int w = 0; w < nw; ++w) { s1 += red[0][w]; s2 += red[1][w]; }
   const float m1 = s1 / n, m2 = s2 / n;
 
-  for (int i = threadIdx.x; i < n; i += blockDim.x) {
-    int chan = i / HW;
-    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
-    float grad = to_f32(dyg_in[idx]);
-    if (RELU) grad = to_f32(yg[idx]) > 0.f ? grad : 0.f;
-    float xhat = (to_f32(xg[idx]) - mean) * rstd;
-    float dxhat = grad * to_f32(gam[chan]);
-    dxg[idx] = from_f32<T>(rstd * (dxhat - m1 - xhat * m2));
-    if (HAS_RES) drg[idx] = from_f32<T>(grad);
+  if (vec) {
+    const int nv = n / 8;
+    for (int v8 = threadIdx.x; v8 < nv; v8 += blockDim.x) {
+      int i = v8 * 8;
+      int chan = i / HW;
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+      Pack<T, 8> pdy = *reinterpret_cast<const Pack<T, 8>*>(&dyg_in[idx]);
+      Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xg[idx]);
+      Pack<T, 8> py;
+      if (RELU) py = *reinterpret_cast<const Pack<T, 8>*>(&yg[idx]);
+      const float ga = to_f32(gam[chan]);
+      Pack<T, 8> pdx, pdr;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float grad = to_f32(pdy.v[e]);
+        if (RELU) grad = to_f32(py.v[e]) > 0.f ? grad : 0.f;
+        float xhat = (to_f32(px.v[e]) - mean) * rstd;
+        float dxhat = grad * ga;
+        pdx.v[e] = from_f32<T>(rstd * (dxhat - m1 - xhat * m2));
+        if (HAS_RES) pdr.v[e] = from_f32<T>(grad);
+      }
+      *reinterpret_cast<Pack<T, 8>*>(&dxg[idx]) = pdx;
+      if (HAS_RES) *reinterpret_cast<Pack<T, 8>*>(&drg[idx]) = pdr;
+    }
+  } else {
+    for (int i = threadIdx.x; i < n; i += blockDim.x) {
+      int chan = i / HW;
+      int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+      float grad = to_f32(dyg_in[idx]);
+      if (RELU) grad = to_f32(yg[idx]) > 0.f ? grad : 0.f;
+      float xhat = (to_f32(xg[idx]) - mean) * rstd;
+      float dxhat = grad * to_f32(gam[chan]);
+      dxg[idx] = from_f32<T>(rstd * (dxhat - m1 - xhat * m2));
+      if (HAS_RES) drg[idx] = from_f32<T>(grad);
+    }
   }
   __syncthreads();
   for (int i = threadIdx.x; i < cg; i += blockDim.x) {
