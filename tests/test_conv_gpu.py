@@ -208,7 +208,15 @@ def test_conv5x5_matches_reference(C, IC, OC, B, H, relu):
     wr = w0.clone().requires_grad_(True)
     br = b0.clone().requires_grad_(True)
     yr = _cpu_conv5x5(xr, wr, br, relu)
-    yr.backward(dy)
+    if relu:
+        # bf16 y and fp32 y disagree on sign for pre-activations near 0,
+        # flipping the ReLU mask at a few positions; compare gradients
+        # under the KERNEL's mask (fwd output itself is checked below)
+        dy_eff = dy * (y.float().cpu() > 0)
+        yr_lin = _cpu_conv5x5(xr, wr, br, False)
+        yr_lin.backward(dy_eff)
+    else:
+        yr.backward(dy)
 
     torch.testing.assert_close(y.float().cpu(), yr,
                                atol=0.05 * (IC ** 0.5) + 0.02, rtol=5e-2)
