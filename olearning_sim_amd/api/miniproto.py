@@ -181,16 +181,16 @@ class ProtoFile:
                 f.type_name = ".google.protobuf.Empty"
             else:
                 # enum vs message resolved after parsing (enums may be
-                # declared later in the file) — mark and fix in a second
-                # pass via type_name; pool resolves either kind
+                # declared later in the file) — leave type unset and fix
+                # in the second pass; the pool resolves either kind
                 f.type_name = ftype          # patched to full name below
-                f.type = 0                   # fixed in _finalize
             i = self._skip_to(toks, i, ";") + 1
         return i + 1
 
     def _pending_fixup(self, md) -> None:
         for f in md.field:
-            if f.type == 0:
+            if not f.HasField("type") and f.type_name and \
+                    not f.type_name.startswith("."):
                 local = f.type_name
                 pkg = f".{self.package}." if self.package else "."
                 if local in self._enum_names:

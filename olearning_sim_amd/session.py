@@ -91,6 +91,16 @@ class SimulatorSession:
         t.start()
         return server
 
+    def serve_grpc(self, host: str = "127.0.0.1", port: int = 0):
+        """Serve the six services over real protobuf wire format (the
+        reference's gRPC surface; message classes compiled at runtime
+        from api/protos/*.proto by api/miniproto.py).  Returns the
+        started grpc.Server; the bound port is ``server._ols_port``."""
+        from .api.grpc_server import build_grpc_server
+        server = build_grpc_server(self, port=port, host=host)
+        server.start()
+        return server
+
     def shutdown(self) -> None:
         if self.task_mgr is not None:
             self.task_mgr.shutdown()

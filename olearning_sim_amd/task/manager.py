@@ -283,6 +283,16 @@ class TaskManager:
         return False, "task not queued or running"
 
     # -- status fusion -----------------------------------------------------
+    def change_scheduler(self, name: str) -> bool:
+        """Swap the scheduling strategy by name (reference changeScheduler
+        RPC, taskService.proto; StrategyFactory holds the registry)."""
+        from .scheduler import StrategyFactory
+        try:
+            self.scheduler.strategy = StrategyFactory.create(name or "default")
+            return True
+        except KeyError:
+            return False
+
     def get_task_queue(self) -> List[str]:
         return self.queue.get_task_ids()
 
