@@ -166,13 +166,24 @@ def main() -> int:
         profiler.__exit__(None, None, None)
         profiler.export_chrome_trace(args.trace)
 
-    # MAX over ranks
-    t = torch.tensor([elapsed], dtype=torch.float64,
-                     device=eng.device if ctx.enabled and use_gpu else "cpu")
+    # MAX over ranks; also gather per-rank timings so a straggler or a
+    # rank that failed to initialise is visible in the output
+    per_rank = [round(elapsed, 4)]
     if ctx.enabled:
         import torch.distributed as dist
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-    elapsed = float(t[0])
+        dev = eng.device if use_gpu else "cpu"
+        all_t = torch.zeros(world, dtype=torch.float64, device=dev)
+        all_t[ctx.rank] = elapsed
+        dist.all_reduce(all_t)
+        per_rank = [round(float(x), 4) for x in all_t.tolist()]
+        elapsed = max(per_rank)
+        # N ranks really initialised (a silent single-rank fallback must
+        # not masquerade as an N-GPU number)
+        assert dist.get_world_size() == world, \
+            f"world_size {dist.get_world_size()} != requested {world}"
+    if args.gpus > 1 and world != args.gpus and ctx.rank == 0:
+        print(f"WARNING: --gpus {args.gpus} requested but world={world} "
+              f"(launch under torch.distributed.run)", file=sys.stderr)
 
     clients_per_round = clients_per_gpu * world
     value = clients_per_round * args.steps / elapsed
@@ -205,6 +216,7 @@ def main() -> int:
                 "parallelism": f"dp{world}",
                 "global_batch": clients_per_round * job.batch_size,
                 "seq_len": job.seq_len or None,
+                "per_rank_s": per_rank,
             },
         }
         print(json.dumps(out))
