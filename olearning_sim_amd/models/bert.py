@@ -18,6 +18,82 @@ import torch.nn.functional as F
 from .base import (ClientBatchedModel, Params, binit, blinear, blayernorm)
 
 
+class _TiedHeadCE(torch.autograd.Function):
+    """Fused tied-LM-head + cross-entropy over vocab slices.
+
+    loss = mean_n CE(hs @ tok^T + bias, labels) computed slice-by-slice
+    with online softmax statistics — the [C·B·L, V] logits tensor
+    (3.9 GB at C=125 for BERT-base) is never materialised, forward or
+    backward (docs/ROUND2_DESIGN.md §5).  All GEMMs are client-batched
+    bmm on natural-layout operands (the hipBLASLt strided-transpose
+    fault, models/base.py, is avoided by materialising per-slice
+    transposes of the SMALL dlogits tile only)."""
+
+    SLICE_V = 8192
+
+    @staticmethod
+    def forward(ctx, hs, tok, bias, labels):
+        # hs [C, N, H] (bf16), tok [C, V, H], bias [C, V], labels [C, N]
+        C, N, H = hs.shape
+        V = tok.shape[1]
+        sv = _TiedHeadCE.SLICE_V
+        tok_t = tok.transpose(1, 2).contiguous()        # [C, H, V] (once)
+        m = torch.full((C, N), float("-inf"), device=hs.device,
+                       dtype=torch.float32)
+        l = torch.zeros(C, N, device=hs.device, dtype=torch.float32)
+        zy = torch.zeros(C, N, device=hs.device, dtype=torch.float32)
+        lab = labels.view(C, N)
+        for s0 in range(0, V, sv):
+            s1 = min(s0 + sv, V)
+            logits = torch.bmm(hs, tok_t[:, :, s0:s1]).float() \
+                + bias[:, s0:s1].float().unsqueeze(1)
+            m_new = torch.maximum(m, logits.max(dim=2).values)
+            l = l * torch.exp(m - m_new) + \
+                torch.exp(logits - m_new.unsqueeze(2)).sum(dim=2)
+            m = m_new
+            sel = (lab >= s0) & (lab < s1)
+            idx = (lab - s0).clamp(0, s1 - s0 - 1)
+            zy = torch.where(sel, logits.gather(2, idx.unsqueeze(2))
+                             .squeeze(2), zy)
+        loss = (torch.log(l) + m - zy).mean()
+        ctx.save_for_backward(hs, tok, tok_t, bias, lab, m, l)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        hs, tok, tok_t, bias, lab, m, l = ctx.saved_tensors
+        C, N, H = hs.shape
+        V = tok.shape[1]
+        sv = _TiedHeadCE.SLICE_V
+        g = grad_out / (C * N)
+        lse = (m + torch.log(l)).unsqueeze(2)           # [C, N, 1]
+        dhs = torch.zeros_like(hs)
+        dtok = torch.empty_like(tok)
+        dbias = torch.empty_like(bias)
+        flat_rows = torch.arange(C * N, device=hs.device)
+        for s0 in range(0, V, sv):
+            s1 = min(s0 + sv, V)
+            logits = torch.bmm(hs, tok_t[:, :, s0:s1]).float() \
+                + bias[:, s0:s1].float().unsqueeze(1)
+            p = torch.exp(logits - lse)                  # softmax slice
+            sel = (lab >= s0) & (lab < s1)
+            idx = (lab - s0).clamp(0, s1 - s0 - 1)
+            pf = p.view(C * N, s1 - s0)
+            self_ = sel.view(-1)
+            pf[flat_rows[self_], idx.view(-1)[self_]] -= 1.0
+            dlogits = (p * g).to(hs.dtype)               # [C, N, s]
+            dbias[:, s0:s1] = dlogits.sum(dim=1).to(bias.dtype)
+            dhs += torch.bmm(dlogits, tok[:, s0:s1].to(hs.dtype))
+            dlt = dlogits.transpose(1, 2).contiguous()   # [C, s, N] small
+            dtok[:, s0:s1] = torch.bmm(dlt, hs).to(tok.dtype)
+        return dhs, dtok, dbias, None
+
+
+def tied_head_ce(hs: torch.Tensor, tok: torch.Tensor, bias: torch.Tensor,
+                 labels: torch.Tensor) -> torch.Tensor:
+    return _TiedHeadCE.apply(hs, tok, bias, labels)
+
+
 class BertLM(ClientBatchedModel):
     name = "bert"
     sequence_model = True
@@ -71,7 +147,8 @@ class BertLM(ClientBatchedModel):
         return p
 
     # ------------------------------------------------------------------
-    def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+    def encode(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        """Transformer body -> hidden states [C, B*L, H]."""
         # x: [C, B, L] token ids
         C, B, L = x.shape
         h, nh = self.hidden, self.heads
@@ -104,21 +181,34 @@ class BertLM(ClientBatchedModel):
             hs = hs + blinear(ff, params[f"{pre}.ffn_out.w"],
                               params[f"{pre}.ffn_out.b"]).view(C, B, L, h)
             hs = blayernorm(hs, params[f"{pre}.ln2.g"], params[f"{pre}.ln2.b"])
+        return hs.view(C, B * L, h).contiguous()
 
+    def forward(self, params: Params, x: torch.Tensor) -> torch.Tensor:
+        C, B, L = x.shape
+        hs = self.encode(params, x)
         # tied LM head: logits = hs @ emb^T + bias  -> [C, B, L, V].
         # The transpose is materialised and the GEMM goes through
         # blinear so fwd AND bwd operands are contiguous (hipBLASLt
         # strided-view fault, models/base.py _BLinearFn).
-        tok_t = tok.transpose(1, 2).contiguous()
-        logits = blinear(hs.view(C, B * L, h), tok_t,
-                         params["head.bias"])
+        tok_t = params["emb.tok"].transpose(1, 2).contiguous()
+        logits = blinear(hs, tok_t, params["head.bias"])
         return logits.view(C, B, L, self.vocab_size)
 
     def loss(self, params: Params, x: torch.Tensor,
              y: torch.Tensor) -> torch.Tensor:
         # y: [C, B, L] next-token targets
+        C, B, L = x.shape
+        if x.is_cuda:
+            # fused tied-head + CE over vocab slices: the [C·B·L, V]
+            # logits never materialise (fwd or bwd)
+            hs = self.encode(params, x)
+            bias = params["head.bias"]
+            if bias.dim() == 1:
+                bias = bias.unsqueeze(0).expand(C, -1)
+            return tied_head_ce(hs, params["emb.tok"], bias,
+                                y.reshape(C, B * L))
         logits = self.forward(params, x)
-        C, B, L, V = logits.shape
+        V = logits.shape[-1]
         from ..ops import cross_entropy_fwd_bwd
         return cross_entropy_fwd_bwd(logits.reshape(C * B * L, V),
                                      y.reshape(-1))

@@ -218,3 +218,45 @@ def test_resnet_custom_conv_path_backward_matches_default_cpu():
     g_default = grad_of(m.forward)
     g_custom = grad_of(m.forward_cbf)
     torch.testing.assert_close(g_custom, g_default, atol=5e-4, rtol=5e-3)
+
+
+def test_bert_tied_head_ce_matches_full_logits():
+    import pytest
+    """Sliced tied-head+CE (no logits materialisation) == the full
+    logits + cross-entropy path: loss and all gradients."""
+    import torch
+    from olearning_sim_amd.models.bert import BertTiny, tied_head_ce
+    from olearning_sim_amd.models.bert import _TiedHeadCE
+    m = BertTiny(seq_len=8, vocab_size=50, hidden=16, layers=1, heads=2)
+    g = torch.Generator().manual_seed(0)
+    C, B, L = 3, 2, 8
+    H, V = 16, 50
+    hs0 = torch.randn(C, B * L, H, generator=g)
+    tok0 = torch.randn(C, V, H, generator=g) * 0.2
+    bias0 = torch.randn(C, V, generator=g) * 0.1
+    y = torch.randint(0, V, (C, B * L), generator=g)
+
+    old_slice = _TiedHeadCE.SLICE_V
+    _TiedHeadCE.SLICE_V = 16           # force several slices + tail
+    try:
+        hs = hs0.clone().requires_grad_(True)
+        tok = tok0.clone().requires_grad_(True)
+        bias = bias0.clone().requires_grad_(True)
+        loss = tied_head_ce(hs, tok, bias, y)
+        loss.backward()
+
+        hs_r = hs0.clone().requires_grad_(True)
+        tok_r = tok0.clone().requires_grad_(True)
+        bias_r = bias0.clone().requires_grad_(True)
+        logits = torch.bmm(hs_r, tok_r.transpose(1, 2)) + bias_r.unsqueeze(1)
+        ref = torch.nn.functional.cross_entropy(
+            logits.reshape(C * B * L, V), y.reshape(-1))
+        ref.backward()
+
+        assert float(loss) == pytest.approx(float(ref), rel=1e-5)
+        torch.testing.assert_close(hs.grad, hs_r.grad, atol=1e-5, rtol=1e-4)
+        torch.testing.assert_close(tok.grad, tok_r.grad, atol=1e-5, rtol=1e-4)
+        torch.testing.assert_close(bias.grad, bias_r.grad, atol=1e-5,
+                                   rtol=1e-4)
+    finally:
+        _TiedHeadCE.SLICE_V = old_slice
