@@ -94,7 +94,14 @@ class LogicalEngine:
             wait_interval=job.flow_wait_interval,
             total_timeout=job.flow_total_timeout,
             work_dir=job.flow_work_dir or job.checkpoint_dir or ".")
+        # double-buffered delta accumulator: round r's RCCL all-reduce
+        # overlaps the host-side round tail (stats/result rows/flow
+        # gates/cohort selection) and is waited only where the master is
+        # next READ (_finish_aggregate) — semantics identical, comm cost
+        # hidden (SURVEY §5 "overlap the collective")
         self._delta = self.master.zeros_like_flat()
+        self._delta_alt: Optional[torch.Tensor] = None
+        self._pending_agg = None     # (work, delta_buf, total_weight)
         # totals across rounds (reference logical_result accounting)
         self.success_total = 0
         self.failed_total = 0
@@ -175,6 +182,7 @@ class LogicalEngine:
     def evaluate_global(self, round_idx: int) -> Dict[str, float]:
         """Run the aggregated global model on a held-out synthetic batch
         (the reference's evaluate-style operator)."""
+        self._finish_aggregate()
         job = self.job
         with torch.no_grad():
             cast = {k: v.to(self.dtype).unsqueeze(0)
@@ -190,9 +198,22 @@ class LogicalEngine:
             acc = float((flat.argmax(-1) == labels).float().mean())
         return {"eval_loss": loss, "eval_acc": acc}
 
+    def _finish_aggregate(self) -> None:
+        """Wait any in-flight delta all-reduce and fold it into the
+        master (called wherever the master is next read)."""
+        if self._pending_agg is None:
+            return
+        work, delta, total_weight = self._pending_agg
+        self._pending_agg = None
+        if work is not None:
+            work.wait()
+        if total_weight > 0:
+            fused.apply_aggregate([self.master.flat], [delta], total_weight)
+
     def _op_train(self, round_idx: int,
                   op_name: str = "train") -> Dict[str, Any]:
         """The training operator: one cohort pass + aggregation."""
+        self._finish_aggregate()
         job = self.job
         ids = self.select_cohort(round_idx)
         cohort = int(ids.numel())
@@ -268,20 +289,26 @@ class LogicalEngine:
             stats = torch.tensor([local_weight] + succ_t + fail_t,
                                  dtype=torch.float64, device=self.device)
             work = pdist.all_reduce_flat(self._delta, async_op=True)
-            pdist.all_reduce_flat(stats)
-            if work is not None:
-                work.wait()
+            pdist.all_reduce_flat(stats)      # small, waited (round status)
             total_weight = float(stats[0])
             succ_t = [int(x) for x in stats[1:1 + T]]
             fail_t = [int(x) for x in stats[1 + T:1 + 2 * T]]
             success = sum(succ_t)
             failed = sum(fail_t)
+            # defer the big delta wait+apply: the collective overlaps
+            # everything until the master is next read; swap to the
+            # alternate delta buffer so the next round's accumulation
+            # never races the in-flight reduce
+            self._pending_agg = (work, self._delta, total_weight)
+            if self._delta_alt is None:
+                self._delta_alt = self.master.zeros_like_flat()
+            self._delta, self._delta_alt = self._delta_alt, self._delta
         else:
             total_weight = local_weight
             success, failed = success_local, failed_local
-
-        if total_weight > 0:
-            fused.apply_aggregate([self.master.flat], [self._delta], total_weight)
+            if total_weight > 0:
+                fused.apply_aggregate([self.master.flat], [self._delta],
+                                      total_weight)
 
         self.success_total += success
         self.failed_total += failed
@@ -377,6 +404,7 @@ class LogicalEngine:
             elif kind == "evaluate":
                 record.update(self.evaluate_global(round_idx))
             elif kind == "checkpoint":
+                self._finish_aggregate()
                 if self.ctx.rank == 0 and job.checkpoint_dir:
                     record["checkpoint"] = save_checkpoint(
                         job.checkpoint_dir, job.task_id, round_idx,
@@ -391,10 +419,12 @@ class LogicalEngine:
             if "eval_acc" in record:
                 self.perf.record(job.task_id, "eval_acc",
                                  record["eval_acc"], round_idx)
-        if job.save_every_round and job.checkpoint_dir                 and self.ctx.rank == 0 and "checkpoint" not in record:
-            record["checkpoint"] = save_checkpoint(
-                job.checkpoint_dir, job.task_id, round_idx,
-                self.master.state_dict(), job.model_update_style)
+        if job.save_every_round and job.checkpoint_dir                 and "checkpoint" not in record:
+            self._finish_aggregate()
+            if self.ctx.rank == 0:
+                record["checkpoint"] = save_checkpoint(
+                    job.checkpoint_dir, job.task_id, round_idx,
+                    self.master.state_dict(), job.model_update_style)
         return record
 
     # ------------------------------------------------------------------
@@ -413,6 +443,7 @@ class LogicalEngine:
                 self.result_sink(self._round_result(rec))
             if rec["round_failed"]:
                 break
+        self._finish_aggregate()
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
         elapsed = time.time() - t0
