@@ -92,14 +92,21 @@ def kaiming(shape, fan_in: int, device, dtype, generator) -> torch.Tensor:
     return t.to(dtype)
 
 
-class _BLinearFn(torch.autograd.Function):
-    """bmm-based per-client linear with layout-safe backward.
+import os as _os
 
-    hipBLASLt (ROCm 7.0 torch, gfx950) memory-faults when a bmm operand
-    is a strided transposed view at large batch counts — which is
-    exactly what autograd's default bmm backward produces (x^T, w^T
-    views).  This Function materialises the transposes, so every GEMM
-    the GPU sees has contiguous operands."""
+# An earlier ROCm/torch build memory-faulted when a bmm operand was a
+# strided transposed view at batch >= ~100; tools/bmmprobe.py verifies
+# the current stack handles every backward shape with plain views
+# (probe log: gpurun_out/bmmprobe.log, all maxerr 0.0).  The
+# materialised-transpose fallback stays behind OLSIM_BMM_SAFE=1.
+_BMM_SAFE = _os.environ.get("OLSIM_BMM_SAFE", "") == "1"
+
+
+class _BLinearFn(torch.autograd.Function):
+    """bmm-based per-client linear with an explicit backward (the
+    transposed backward operands stay VIEWS — no copy traffic; the
+    BERT-base profile showed materialised transposes were the dominant
+    elementwise cost, profiles/bert_full_logits_r02.md)."""
 
     @staticmethod
     def forward(ctx, x, w, b):
@@ -116,9 +123,11 @@ class _BLinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            dx = torch.bmm(dy, w.transpose(1, 2).contiguous())
+            wt = w.transpose(1, 2)
+            dx = torch.bmm(dy, wt.contiguous() if _BMM_SAFE else wt)
         if ctx.needs_input_grad[1]:
-            dw = torch.bmm(x.transpose(1, 2).contiguous(), dy)
+            xt = x.transpose(1, 2)
+            dw = torch.bmm(xt.contiguous() if _BMM_SAFE else xt, dy)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.sum(dim=1)
         return dx, dw, db

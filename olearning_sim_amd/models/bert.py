@@ -37,7 +37,6 @@ class _TiedHeadCE(torch.autograd.Function):
         C, N, H = hs.shape
         V = tok.shape[1]
         sv = _TiedHeadCE.SLICE_V
-        tok_t = tok.transpose(1, 2).contiguous()        # [C, H, V] (once)
         m = torch.full((C, N), float("-inf"), device=hs.device,
                        dtype=torch.float32)
         l = torch.zeros(C, N, device=hs.device, dtype=torch.float32)
@@ -45,7 +44,8 @@ class _TiedHeadCE(torch.autograd.Function):
         lab = labels.view(C, N)
         for s0 in range(0, V, sv):
             s1 = min(s0 + sv, V)
-            logits = torch.bmm(hs, tok_t[:, :, s0:s1]).float() \
+            # transposed slice stays a VIEW (tools/bmmprobe.py)
+            logits = torch.bmm(hs, tok[:, s0:s1].transpose(1, 2)).float() \
                 + bias[:, s0:s1].float().unsqueeze(1)
             m_new = torch.maximum(m, logits.max(dim=2).values)
             l = l * torch.exp(m - m_new) + \
@@ -56,12 +56,12 @@ class _TiedHeadCE(torch.autograd.Function):
             zy = torch.where(sel, logits.gather(2, idx.unsqueeze(2))
                              .squeeze(2), zy)
         loss = (torch.log(l) + m - zy).mean()
-        ctx.save_for_backward(hs, tok, tok_t, bias, lab, m, l)
+        ctx.save_for_backward(hs, tok, bias, lab, m, l)
         return loss
 
     @staticmethod
     def backward(ctx, grad_out):
-        hs, tok, tok_t, bias, lab, m, l = ctx.saved_tensors
+        hs, tok, bias, lab, m, l = ctx.saved_tensors
         C, N, H = hs.shape
         V = tok.shape[1]
         sv = _TiedHeadCE.SLICE_V
@@ -73,7 +73,7 @@ class _TiedHeadCE(torch.autograd.Function):
         flat_rows = torch.arange(C * N, device=hs.device)
         for s0 in range(0, V, sv):
             s1 = min(s0 + sv, V)
-            logits = torch.bmm(hs, tok_t[:, :, s0:s1]).float() \
+            logits = torch.bmm(hs, tok[:, s0:s1].transpose(1, 2)).float() \
                 + bias[:, s0:s1].float().unsqueeze(1)
             p = torch.exp(logits - lse)                  # softmax slice
             sel = (lab >= s0) & (lab < s1)
@@ -84,8 +84,8 @@ class _TiedHeadCE(torch.autograd.Function):
             dlogits = (p * g).to(hs.dtype)               # [C, N, s]
             dbias[:, s0:s1] = dlogits.sum(dim=1).to(bias.dtype)
             dhs += torch.bmm(dlogits, tok[:, s0:s1].to(hs.dtype))
-            dlt = dlogits.transpose(1, 2).contiguous()   # [C, s, N] small
-            dtok[:, s0:s1] = torch.bmm(dlt, hs).to(tok.dtype)
+            dtok[:, s0:s1] = torch.bmm(dlogits.transpose(1, 2),
+                                       hs).to(tok.dtype)
         return dhs, dtok, dbias, None
 
 
