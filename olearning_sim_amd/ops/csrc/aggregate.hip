@@ -45,12 +45,64 @@ __global__ __launch_bounds__(OLS_THREADS) void k_delta_accum(
   }
 }
 
+// Vector variant for the common single-param call (nblocks == 1,
+// n % 8 == 0): 8 consecutive j per thread, so each c-step of a wave
+// streams 1 KiB contiguous (16 B/lane) instead of 128 B of scalar bf16
+// loads — the scalar form measured ~8x off the HBM roofline at
+// C=1250 (profiles/resnet_v6_r02.md, k_delta_accum 41 ms/round).
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_delta_accum_v8(
+    float* __restrict__ delta, const T* __restrict__ buf,
+    const T* __restrict__ master, const float* __restrict__ weights,
+    int64_t clients, int64_t n, float wsum) {
+  __shared__ float w_lds[MAX_LDS_W];
+  const bool w_in_lds = clients <= MAX_LDS_W;
+  if (w_in_lds) {
+    for (int c = threadIdx.x; c < clients; c += blockDim.x)
+      w_lds[c] = weights[c];
+    __syncthreads();
+  }
+  const float* w = w_in_lds ? w_lds : weights;
+
+  const int64_t nv = n / 8;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t v = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       v < nv; v += stride) {
+    const int64_t j = v * 8;
+    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    const T* col = buf + j;
+    for (int64_t c = 0; c < clients; ++c) {
+      Pack<T, 8> p = *reinterpret_cast<const Pack<T, 8>*>(&col[c * n]);
+      const float wc = w[c];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += wc * to_f32(p.v[e]);
+    }
+    Pack<T, 8> pm = *reinterpret_cast<const Pack<T, 8>*>(&master[j]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      delta[j + e] += acc[e] - wsum * to_f32(pm.v[e]);
+  }
+}
+
 extern "C" void ols_weighted_delta_accum_flat(
     float* delta, const void* buf, const void* master, const float* weights,
     const int64_t* offs, int nblocks, int64_t clients, int64_t pglobal,
     float wsum, int dtype, hipStream_t stream) {
-  dim3 grid(ols_grid(pglobal, OLS_THREADS));
   dim3 block(OLS_THREADS);
+  if (nblocks == 1 && pglobal % 8 == 0 && dtype != 0) {
+    dim3 grid(ols_grid(pglobal / 8, OLS_THREADS));
+    if (dtype == 1)
+      hipLaunchKernelGGL((k_delta_accum_v8<__hip_bfloat16>), grid, block, 0,
+                         stream, delta, (const __hip_bfloat16*)buf,
+                         (const __hip_bfloat16*)master, weights, clients,
+                         pglobal, wsum);
+    else
+      hipLaunchKernelGGL((k_delta_accum_v8<__half>), grid, block, 0, stream,
+                         delta, (const __half*)buf, (const __half*)master,
+                         weights, clients, pglobal, wsum);
+    return;
+  }
+  dim3 grid(ols_grid(pglobal, OLS_THREADS));
   switch (dtype) {
     case 0:
       hipLaunchKernelGGL((k_delta_accum<float>), grid, block, 0, stream,
