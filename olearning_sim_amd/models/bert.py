@@ -37,6 +37,12 @@ class _TiedHeadCE(torch.autograd.Function):
         C, N, H = hs.shape
         V = tok.shape[1]
         sv = _TiedHeadCE.SLICE_V
+        # hipBLASLt faults on SLICED+TRANSPOSED bmm operands (batch
+        # stride > matrix footprint + transpose; tools/bmmprobe.py
+        # probe_sliced reproduces it on this stack — full-tensor
+        # transposed views are fine).  Materialise tok^T once; its
+        # non-transposed slices are safe bmm operands.
+        tok_t = tok.transpose(1, 2).contiguous()        # [C, H, V]
         m = torch.full((C, N), float("-inf"), device=hs.device,
                        dtype=torch.float32)
         l = torch.zeros(C, N, device=hs.device, dtype=torch.float32)
@@ -44,8 +50,7 @@ class _TiedHeadCE(torch.autograd.Function):
         lab = labels.view(C, N)
         for s0 in range(0, V, sv):
             s1 = min(s0 + sv, V)
-            # transposed slice stays a VIEW (tools/bmmprobe.py)
-            logits = torch.bmm(hs, tok[:, s0:s1].transpose(1, 2)).float() \
+            logits = torch.bmm(hs, tok_t[:, :, s0:s1]).float() \
                 + bias[:, s0:s1].float().unsqueeze(1)
             m_new = torch.maximum(m, logits.max(dim=2).values)
             l = l * torch.exp(m - m_new) + \
@@ -56,12 +61,12 @@ class _TiedHeadCE(torch.autograd.Function):
             zy = torch.where(sel, logits.gather(2, idx.unsqueeze(2))
                              .squeeze(2), zy)
         loss = (torch.log(l) + m - zy).mean()
-        ctx.save_for_backward(hs, tok, bias, lab, m, l)
+        ctx.save_for_backward(hs, tok, tok_t, bias, lab, m, l)
         return loss
 
     @staticmethod
     def backward(ctx, grad_out):
-        hs, tok, bias, lab, m, l = ctx.saved_tensors
+        hs, tok, tok_t, bias, lab, m, l = ctx.saved_tensors
         C, N, H = hs.shape
         V = tok.shape[1]
         sv = _TiedHeadCE.SLICE_V
@@ -73,7 +78,7 @@ class _TiedHeadCE(torch.autograd.Function):
         flat_rows = torch.arange(C * N, device=hs.device)
         for s0 in range(0, V, sv):
             s1 = min(s0 + sv, V)
-            logits = torch.bmm(hs, tok[:, s0:s1].transpose(1, 2)).float() \
+            logits = torch.bmm(hs, tok_t[:, :, s0:s1]).float() \
                 + bias[:, s0:s1].float().unsqueeze(1)
             p = torch.exp(logits - lse)                  # softmax slice
             sel = (lab >= s0) & (lab < s1)
@@ -83,6 +88,8 @@ class _TiedHeadCE(torch.autograd.Function):
             pf[flat_rows[self_], idx.view(-1)[self_]] -= 1.0
             dlogits = (p * g).to(hs.dtype)               # [C, N, s]
             dbias[:, s0:s1] = dlogits.sum(dim=1).to(bias.dtype)
+            # tok slice is non-transposed (safe); dlogits' transpose is
+            # of a fresh contiguous tensor (full-tensor view, safe)
             dhs += torch.bmm(dlogits, tok[:, s0:s1].to(hs.dtype))
             dtok[:, s0:s1] = torch.bmm(dlogits.transpose(1, 2),
                                        hs).to(tok.dtype)
