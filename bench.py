@@ -106,6 +106,9 @@ def main() -> int:
     ap.add_argument("--trace", default="",
                     help="write a torch.profiler chrome trace of the timed "
                          "steps to this path (off by default; adds overhead)")
+    ap.add_argument("--op-table", default="",
+                    help="write torch.profiler key_averages (op-level GPU "
+                         "time attribution) to this path")
     args = ap.parse_args()
 
     preset = dict(PRESETS[args.preset])
@@ -147,7 +150,7 @@ def main() -> int:
     sync()
 
     profiler = None
-    if args.trace and ctx.rank == 0:
+    if (args.trace or args.op_table) and ctx.rank == 0:
         from torch.profiler import profile, ProfilerActivity
         acts = [ProfilerActivity.CPU]
         if use_gpu:
@@ -164,7 +167,13 @@ def main() -> int:
 
     if profiler is not None:
         profiler.__exit__(None, None, None)
-        profiler.export_chrome_trace(args.trace)
+        if args.trace:
+            profiler.export_chrome_trace(args.trace)
+        if args.op_table:
+            with open(args.op_table, "w") as f:
+                f.write(profiler.key_averages().table(
+                    sort_by="self_cuda_time_total" if use_gpu
+                    else "self_cpu_time_total", row_limit=60))
 
     # MAX over ranks; also gather per-rank timings so a straggler or a
     # rank that failed to initialise is visible in the output
