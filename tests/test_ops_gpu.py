@@ -209,3 +209,22 @@ def test_engine_per_tier_accounting_gpu():
     tgt = rows[0]["logical_result"]["logical_result"][0]["simulation_target"]
     assert tgt["devices"] == ["high", "low"]
     assert eng.master.flat.isfinite().all()
+
+
+@pytest.mark.parametrize("n", [4096, 36864, 100])
+def test_weighted_delta_accum_single_param(n):
+    """Single-param call path (nblocks==1): n%8==0 rides the vectorised
+    k_delta_accum_v8 kernel, n%8!=0 the scalar fallback."""
+    from olearning_sim_amd.ops import fused
+    torch.manual_seed(3)
+    C = 130
+    buf = torch.randn(C * n, device="cuda").to(torch.bfloat16)
+    master = torch.randn(n, device="cuda").to(torch.bfloat16)
+    w = torch.rand(C, device="cuda")
+    delta = torch.randn(n, device="cuda")
+    offs = torch.tensor([0, n], dtype=torch.int64, device="cuda")
+    ref = delta + ((buf.view(C, n).float()
+                    - master.float().unsqueeze(0)) * w.unsqueeze(1)).sum(0)
+    fused.weighted_delta_accum_flat(delta, buf, master, w, C, offsets=offs,
+                                    wsum=float(w.sum()))
+    torch.testing.assert_close(delta, ref, atol=5e-2, rtol=5e-2)
