@@ -185,6 +185,123 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v6(
 }
 
 // ---------------------------------------------------------------------------
+// fwd v7: BK=64 — two MFMA sub-steps per barrier (16 MFMAs/wave/step,
+// amortising the stage/barrier latency the BK=32 pipeline pays every
+// 8 MFMAs; cdna_hip_programming.md: BK 32->64 = +7..16% on the dense
+// GEMM ladder), with s_setprio(1) around the MFMA cluster (T5).
+// Requires K % 64 == 0 (IC a multiple of 64); launcher falls back to
+// v6 otherwise.
+template <int LG_OW_T, int STRIDE>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v7(
+    const __hip_bfloat16* __restrict__ xp, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ y, ConvGeom6 g) {
+  constexpr int BK = 64;
+  int c, tile;
+  if (!xcd_remap6(blockIdx.x, g.C, g.tiles_m * g.tiles_n, c, tile)) return;
+  const int mt = tile / g.tiles_n;
+  const int m0 = mt * CV6_BM;
+  const int n0 = (tile - mt * g.tiles_n) * CV6_BN;
+
+  __shared__ short bT_lds[2][CV6_BN * (BK + CV6_PAD)];
+  const int K = g.IC * 9;
+  const int N = g.B * g.OH * g.OW;
+  const int HpWp = g.Hp * g.Wp;
+  const int64_t planeB = (int64_t)g.B * HpWp;
+  const ushort* xc =
+      reinterpret_cast<const ushort*>(xp) + (int64_t)c * g.IC * planeB;
+  const __hip_bfloat16* wc = w + (int64_t)c * g.OC * K;
+  __hip_bfloat16* yc = y + (int64_t)c * g.OC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int arow = m0 + wave * 16 + (lane & 15);
+  const __hip_bfloat16* wrow = wc + (int64_t)min(arow, g.OC - 1) * K;
+  const bool arow_ok = arow < g.OC;
+
+  f32x4 acc[CV6_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV6_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kk = threadIdx.x & 63;           // staged k-row
+  const int nn0 = (threadIdx.x >> 6) * 32;   // 32 cols per thread
+  ushort breg[32];
+
+  auto gather = [&](int k0) {
+    const int k = k0 + kk;
+    const int ic = k / 9, r = k - ic * 9;
+    const int dh = r / 3, dw = r - dh * 3;
+    const ushort* plane = xc + (int64_t)ic * planeB;
+    constexpr int SUBW = 1 << (LG_OW_T < 5 ? LG_OW_T : 5);
+    constexpr int NROW = 32 / SUBW;
+#pragma unroll
+    for (int rr = 0; rr < NROW; ++rr) {
+      int n = min(n0 + nn0 + rr * SUBW, N - 1);
+      int b = n >> g.lg_ohw;
+      int q = n & ((1 << g.lg_ohw) - 1);
+      int oh = q >> g.lg_ow;
+      int ow0 = q & ((1 << g.lg_ow) - 1);
+      if (SUBW < 32) ow0 = 0;                // sub-row segments are aligned
+      else ow0 = min(ow0, g.OW - 32);
+      const ushort* row = plane + (int64_t)b * HpWp
+                          + (oh * STRIDE + dh) * g.Wp + ow0 * STRIDE + dw;
+#pragma unroll
+      for (int j = 0; j < SUBW; ++j)
+        breg[rr * SUBW + j] = row[j * STRIDE];
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 32; ++j)
+      bT_lds[buf][(nn0 + j) * (BK + CV6_PAD) + kk] = (short)breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    if (k0 + BK < K) gather(k0 + BK);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      bf16x8 a;
+      {
+        uint4 av = *reinterpret_cast<const uint4*>(
+            wrow + k0 + sub * 32 + 8 * (lane >> 4));
+        a = *reinterpret_cast<const bf16x8*>(&av);
+        if (!arow_ok) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) a[e] = 0;
+        }
+      }
+#pragma unroll
+      for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &bT_lds[cur][(nt * 16 + (lane & 15)) * (BK + CV6_PAD)
+                         + sub * 32 + 8 * (lane >> 4)]);
+        acc[nt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (k0 + BK < K) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        yc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // dgrad v6 (stride 1): dX[ic][n] over K=(oc,dh,dw), with the weight
 // flip folded into the A staging (A[ic][k] = W[oc][ic][8-(3dh+dw)]) —
 // no host-side weight transform, W is read once through L2.
@@ -622,16 +739,31 @@ extern "C" void ols_conv3x3_fwd_p(const void* xp, const void* w, void* y,
 #define LAUNCH_FWD6(LG, ST) \
   hipLaunchKernelGGL((k_conv3x3_fwd_v6<LG, ST>), grid, dim3(CONV_THREADS), \
                      0, stream, xb, wb, yb, g)
+#define LAUNCH_FWD7(LG, ST) \
+  hipLaunchKernelGGL((k_conv3x3_fwd_v7<LG, ST>), grid, dim3(CONV_THREADS), \
+                     0, stream, xb, wb, yb, g)
+  const bool v7 = ((IC * 9) % 64 == 0) && getenv("OLSIM_CONV_V7") != nullptr;
   if (stride == 1) {
-    if (g.lg_ow >= 4) LAUNCH_FWD6(4, 1);
+    if (v7) {
+      if (g.lg_ow >= 5) LAUNCH_FWD7(5, 1);
+      else if (g.lg_ow == 4) LAUNCH_FWD7(4, 1);
+      else if (g.lg_ow == 3) LAUNCH_FWD7(3, 1);
+      else LAUNCH_FWD7(2, 1);
+    } else if (g.lg_ow >= 4) LAUNCH_FWD6(4, 1);
     else if (g.lg_ow == 3) LAUNCH_FWD6(3, 1);
     else LAUNCH_FWD6(2, 1);
   } else {
-    if (g.lg_ow >= 4) LAUNCH_FWD6(4, 2);
+    if (v7) {
+      if (g.lg_ow >= 5) LAUNCH_FWD7(5, 2);
+      else if (g.lg_ow == 4) LAUNCH_FWD7(4, 2);
+      else if (g.lg_ow == 3) LAUNCH_FWD7(3, 2);
+      else LAUNCH_FWD7(2, 2);
+    } else if (g.lg_ow >= 4) LAUNCH_FWD6(4, 2);
     else if (g.lg_ow == 3) LAUNCH_FWD6(3, 2);
     else LAUNCH_FWD6(2, 2);
   }
 #undef LAUNCH_FWD6
+#undef LAUNCH_FWD7
 }
 
 extern "C" void ols_conv3x3_dgrad_p(const void* dyp, const void* w, void* dx,
