@@ -742,23 +742,19 @@ extern "C" void ols_conv3x3_fwd_p(const void* xp, const void* w, void* y,
 #define LAUNCH_FWD7(LG, ST) \
   hipLaunchKernelGGL((k_conv3x3_fwd_v7<LG, ST>), grid, dim3(CONV_THREADS), \
                      0, stream, xb, wb, yb, g)
-  const bool v7 = ((IC * 9) % 64 == 0) && getenv("OLSIM_CONV_V7") != nullptr;
+  // BK=64 (v7) wins +5..10% at OW >= 8 stride 1 (A/B: gpurun_out/
+  // v7_fwd.log vs v6_fwd.log at C=250); the 4x4-plane deep layers and
+  // stride 2 measured flat-to-worse, so they stay on BK=32.
+  const bool v7 = ((IC * 9) % 64 == 0) && stride == 1 && g.lg_ow >= 3;
   if (stride == 1) {
-    if (v7) {
-      if (g.lg_ow >= 5) LAUNCH_FWD7(5, 1);
-      else if (g.lg_ow == 4) LAUNCH_FWD7(4, 1);
-      else if (g.lg_ow == 3) LAUNCH_FWD7(3, 1);
-      else LAUNCH_FWD7(2, 1);
-    } else if (g.lg_ow >= 4) LAUNCH_FWD6(4, 1);
+    if (v7 && g.lg_ow >= 5) LAUNCH_FWD7(5, 1);
+    else if (v7 && g.lg_ow == 4) LAUNCH_FWD7(4, 1);
+    else if (v7) LAUNCH_FWD7(3, 1);
+    else if (g.lg_ow >= 4) LAUNCH_FWD6(4, 1);
     else if (g.lg_ow == 3) LAUNCH_FWD6(3, 1);
     else LAUNCH_FWD6(2, 1);
   } else {
-    if (v7) {
-      if (g.lg_ow >= 5) LAUNCH_FWD7(5, 2);
-      else if (g.lg_ow == 4) LAUNCH_FWD7(4, 2);
-      else if (g.lg_ow == 3) LAUNCH_FWD7(3, 2);
-      else LAUNCH_FWD7(2, 2);
-    } else if (g.lg_ow >= 4) LAUNCH_FWD6(4, 2);
+    if (g.lg_ow >= 4) LAUNCH_FWD6(4, 2);
     else if (g.lg_ow == 3) LAUNCH_FWD6(3, 2);
     else LAUNCH_FWD6(2, 2);
   }
