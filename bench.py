@@ -155,7 +155,8 @@ def main() -> int:
         acts = [ProfilerActivity.CPU]
         if use_gpu:
             acts.append(ProfilerActivity.CUDA)
-        profiler = profile(activities=acts)
+        profiler = profile(activities=acts,
+                           record_shapes=bool(args.op_table))
         profiler.__enter__()
 
     t0 = time.perf_counter()
@@ -170,10 +171,15 @@ def main() -> int:
         if args.trace:
             profiler.export_chrome_trace(args.trace)
         if args.op_table:
+            sort = ("self_cuda_time_total" if use_gpu
+                    else "self_cpu_time_total")
             with open(args.op_table, "w") as f:
                 f.write(profiler.key_averages().table(
-                    sort_by="self_cuda_time_total" if use_gpu
-                    else "self_cpu_time_total", row_limit=60))
+                    sort_by=sort, row_limit=60))
+                f.write("\n\n== by input shape ==\n")
+                f.write(profiler.key_averages(
+                    group_by_input_shape=True).table(
+                    sort_by=sort, row_limit=40))
 
     # MAX over ranks; also gather per-rank timings so a straggler or a
     # rank that failed to initialise is visible in the output

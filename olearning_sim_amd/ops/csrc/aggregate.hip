@@ -84,11 +84,58 @@ __global__ __launch_bounds__(OLS_THREADS) void k_delta_accum_v8(
   }
 }
 
+// Client-parallel variant for SMALL params: the j-parallel kernels
+// above give a parameter with n=64 only 8 active lanes, each walking a
+// C-deep serial latency chain (a 64-element GroupNorm gain at C=1250
+// measured ~250 us — pure latency).  Here the grid also spans client
+// tiles; each block reduces its tile's partial and atomicAdd's fp32
+// into delta; the -wsum*master term is applied by the c-tile-0 blocks.
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_delta_accum_cpar(
+    float* __restrict__ delta, const T* __restrict__ buf,
+    const T* __restrict__ master, const float* __restrict__ weights,
+    int64_t clients, int64_t n, float wsum, int ctile) {
+  const int64_t nv = n / 8;
+  const int64_t j8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x);
+  if (j8 >= nv) return;
+  const int64_t j = j8 * 8;
+  const int64_t c0 = (int64_t)blockIdx.y * ctile;
+  const int64_t c1 = min(clients, c0 + ctile);
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  const T* col = buf + j;
+  for (int64_t c = c0; c < c1; ++c) {
+    Pack<T, 8> p = *reinterpret_cast<const Pack<T, 8>*>(&col[c * n]);
+    const float wc = weights[c];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] += wc * to_f32(p.v[e]);
+  }
+  if (blockIdx.y == 0) {
+    Pack<T, 8> pm = *reinterpret_cast<const Pack<T, 8>*>(&master[j]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] -= wsum * to_f32(pm.v[e]);
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) atomicAdd(&delta[j + e], acc[e]);
+}
+
 extern "C" void ols_weighted_delta_accum_flat(
     float* delta, const void* buf, const void* master, const float* weights,
     const int64_t* offs, int nblocks, int64_t clients, int64_t pglobal,
     float wsum, int dtype, hipStream_t stream) {
   dim3 block(OLS_THREADS);
+  if (nblocks == 1 && pglobal % 8 == 0 && dtype == 1 &&
+      pglobal / 8 < 16384 && clients >= 64) {
+    // small param: client-parallel tiles (fills the chip; the j-only
+    // kernel would be a serial latency chain on a handful of lanes)
+    const int ctile = 64;
+    dim3 grid((unsigned)((pglobal / 8 + OLS_THREADS - 1) / OLS_THREADS),
+              (unsigned)((clients + ctile - 1) / ctile));
+    hipLaunchKernelGGL((k_delta_accum_cpar<__hip_bfloat16>), grid, block, 0,
+                       stream, delta, (const __hip_bfloat16*)buf,
+                       (const __hip_bfloat16*)master, weights, clients,
+                       pglobal, wsum, ctile);
+    return;
+  }
   if (nblocks == 1 && pglobal % 8 == 0 && dtype != 0) {
     dim3 grid(ols_grid(pglobal / 8, OLS_THREADS));
     if (dtype == 1)
