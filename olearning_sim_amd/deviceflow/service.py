@@ -100,6 +100,7 @@ class DeviceFlowService:
         # (reference operation/accumulated_amount_table demo rows)
         self.dispatch_history: Dict[str, List[tuple]] = {}
         self.time_scale = time_scale
+        self._ws_producers: Dict[str, object] = {}
         self._rng = random.Random(seed)
         self._lock = threading.RLock()
         self._stop = threading.Event()
@@ -147,6 +148,11 @@ class DeviceFlowService:
         for d in self.dispatchers.values():
             d.stop_event.set()
             d.release_event.set()
+        for prod in self._ws_producers.values():
+            try:
+                prod.close()
+            except Exception:
+                pass
 
     # -- RPC surface ------------------------------------------------------
     def register_task(self, task_id: str,
@@ -176,6 +182,11 @@ class DeviceFlowService:
                                  outbound_service, resources)
                 self.flows[flow_id] = flow
                 self.shelf.ensure_shelf(flow_id)
+                # a ws:// outbound_service gets a live producer (the
+                # reference builds a WebsocketProducer from the flow's
+                # outbound_service, deviceflow_server.py:182-289)
+                if outbound_service.startswith("ws://"):
+                    self.attach_websocket_outbound(outbound_service)
             if strategy:
                 flow.strategy = strategy
             if compute_resource not in flow.notify_start_called:
@@ -274,6 +285,19 @@ class DeviceFlowService:
         reference's websocket/Pulsar outbound)."""
         return {"kind": "inproc", "endpoint": "/deviceflow/outbound",
                 "queue_depth": self.outbound.qsize()}
+
+    def attach_websocket_outbound(self, url: str):
+        """Forward every outbound message to a ws:// consumer (the
+        reference's WebsocketProducer, message_producer.py:59-78 — the
+        external aggregation service).  Returns the producer; one per
+        URL, reused."""
+        prod = self._ws_producers.get(url)
+        if prod is None:
+            from .producers import WebSocketProducer
+            prod = WebSocketProducer(url)
+            self._ws_producers[url] = prod
+            self.outbound.subscribe(prod)
+        return prod
 
     # -- data plane -------------------------------------------------------
     def publish(self, routing_key: str, compute_resource: str,

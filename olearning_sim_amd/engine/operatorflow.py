@@ -41,17 +41,53 @@ class OperatorFlow:
                  wait_interval: float = 1.0, total_timeout: float = 0.0,
                  work_dir: str = "",
                  selection_round_fn: Optional[Callable[[], int]] = None,
-                 custom_gates: Optional[Dict[str, Callable[[int], bool]]] = None):
+                 custom_gates: Optional[Dict[str, Callable[[int], bool]]] = None,
+                 selection_ws_url: str = ""):
         self.task_id = task_id
         self.start_strategy = start_strategy
         self.stop_strategy = stop_strategy
         self.wait_interval = max(0.01, wait_interval)
         self.total_timeout = total_timeout
         self.work_dir = work_dir or "."
+        # waiting_for_global_aggregation polls a selection service for
+        # the current round index; the reference polls it over
+        # WebSocket (operatorflow.py:158-237).  selection_ws_url builds
+        # the poll over utils/ws.py; selection_round_fn injects it
+        # directly (tests, in-process services).
+        if selection_round_fn is None and selection_ws_url:
+            selection_round_fn = self._ws_selection_fn(selection_ws_url)
         self.selection_round_fn = selection_round_fn
         self.custom_gates = custom_gates or {}
         self.log = Logger.shared()
         self._seen_round = -1
+        self._ws_conn = None
+        self._ws_url = selection_ws_url
+
+    def _ws_selection_fn(self, url: str) -> Callable[[], int]:
+        import json as _json
+
+        def query() -> int:
+            from ..utils import ws
+            try:
+                if self._ws_conn is None:
+                    self._ws_conn = ws.connect(url, timeout=5.0)
+                self._ws_conn.send_text(_json.dumps(
+                    {"query": "round_idx", "task_id": self.task_id}))
+                reply = self._ws_conn.recv_text(timeout=5.0)
+                if reply is None:
+                    self._ws_conn = None
+                    return self._seen_round
+                self._seen_round = int(_json.loads(reply)["round_idx"])
+            except (OSError, ConnectionError, ValueError, KeyError):
+                try:
+                    if self._ws_conn is not None:
+                        self._ws_conn.close()
+                except Exception:
+                    pass
+                self._ws_conn = None
+            return self._seen_round
+
+        return query
 
     # ------------------------------------------------------------------
     def _poll(self, cond: Callable[[], bool], what: str) -> None:
