@@ -109,6 +109,9 @@ def main() -> int:
     ap.add_argument("--op-table", default="",
                     help="write torch.profiler key_averages (op-level GPU "
                          "time attribution) to this path")
+    ap.add_argument("--op-stacks", default="",
+                    help="write key_averages grouped by python stack "
+                         "(slow; finds which call site owns a kernel)")
     args = ap.parse_args()
 
     preset = dict(PRESETS[args.preset])
@@ -150,13 +153,14 @@ def main() -> int:
     sync()
 
     profiler = None
-    if (args.trace or args.op_table) and ctx.rank == 0:
+    if (args.trace or args.op_table or args.op_stacks) and ctx.rank == 0:
         from torch.profiler import profile, ProfilerActivity
         acts = [ProfilerActivity.CPU]
         if use_gpu:
             acts.append(ProfilerActivity.CUDA)
         profiler = profile(activities=acts,
-                           record_shapes=bool(args.op_table))
+                           record_shapes=bool(args.op_table),
+                           with_stack=bool(args.op_stacks))
         profiler.__enter__()
 
     t0 = time.perf_counter()
@@ -180,6 +184,12 @@ def main() -> int:
                 f.write(profiler.key_averages(
                     group_by_input_shape=True).table(
                     sort_by=sort, row_limit=40))
+        if args.op_stacks:
+            sort = ("self_cuda_time_total" if use_gpu
+                    else "self_cpu_time_total")
+            with open(args.op_stacks, "w") as f:
+                f.write(profiler.key_averages(group_by_stack_n=7).table(
+                    sort_by=sort, row_limit=25, max_src_column_width=200))
 
     # MAX over ranks; also gather per-rank timings so a straggler or a
     # rank that failed to initialise is visible in the output
