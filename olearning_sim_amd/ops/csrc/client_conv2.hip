@@ -185,6 +185,130 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v6(
 }
 
 // ---------------------------------------------------------------------------
+// fwd v8: BK=32 with a TWO-iteration gather lookahead — two register
+// sets; the loads a commit waits for were issued a FULL iteration
+// earlier (PMC: the one-ahead v6 pipeline is 73% SQ_WAIT_ANY — the 8
+// MFMAs per step cannot cover the gather latency under saturation).
+// Same LDS (2 buffers) and near-same VGPR count, so occupancy is
+// unchanged; hipcc's per-register vmcnt bookkeeping gives the commit a
+// counted wait that leaves the newer set's loads in flight.
+template <int LG_OW_T, int STRIDE>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v8(
+    const __hip_bfloat16* __restrict__ xp, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ y, ConvGeom6 g) {
+  int c, tile;
+  if (!xcd_remap6(blockIdx.x, g.C, g.tiles_m * g.tiles_n, c, tile)) return;
+  const int mt = tile / g.tiles_n;
+  const int m0 = mt * CV6_BM;
+  const int n0 = (tile - mt * g.tiles_n) * CV6_BN;
+
+  __shared__ short bT_lds[2][CV6_BN * (CV6_BK + CV6_PAD)];
+  const int K = g.IC * 9;
+  const int N = g.B * g.OH * g.OW;
+  const int HpWp = g.Hp * g.Wp;
+  const int64_t planeB = (int64_t)g.B * HpWp;
+  const ushort* xc =
+      reinterpret_cast<const ushort*>(xp) + (int64_t)c * g.IC * planeB;
+  const __hip_bfloat16* wc = w + (int64_t)c * g.OC * K;
+  __hip_bfloat16* yc = y + (int64_t)c * g.OC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int arow = m0 + wave * 16 + (lane & 15);
+  const __hip_bfloat16* wrow = wc + (int64_t)min(arow, g.OC - 1) * K;
+  const bool arow_ok = arow < g.OC;
+
+  f32x4 acc[CV6_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV6_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kk = threadIdx.x % CV6_BK;
+  const int nn0 = (threadIdx.x / CV6_BK) * (CV6_BN / 8);
+  ushort brA[CV6_BN / 8], brB[CV6_BN / 8];
+
+  auto gather = [&](ushort* breg, int k0) {
+    const int k = k0 + kk;
+    const int ic = k / 9, r = k - ic * 9;
+    const int dh = r / 3, dw = r - dh * 3;
+    const ushort* plane = xc + (int64_t)ic * planeB;
+    if (LG_OW_T >= 4) {
+      int n = min(n0 + nn0, N - 1);
+      int b = n >> g.lg_ohw;
+      int q = n & ((1 << g.lg_ohw) - 1);
+      int oh = q >> g.lg_ow;
+      int ow0 = min(q & ((1 << g.lg_ow) - 1), g.OW - CV6_BN / 8);
+      const ushort* row = plane + (int64_t)b * HpWp
+                          + (oh * STRIDE + dh) * g.Wp + ow0 * STRIDE + dw;
+#pragma unroll
+      for (int j = 0; j < CV6_BN / 8; ++j) breg[j] = row[j * STRIDE];
+    } else {
+      constexpr int SUBW = 1 << LG_OW_T;
+      constexpr int NROW = (CV6_BN / 8) / SUBW;
+#pragma unroll
+      for (int rr = 0; rr < NROW; ++rr) {
+        int n = min(n0 + nn0 + rr * SUBW, N - 1);
+        int b = n >> g.lg_ohw;
+        int q = n & ((1 << g.lg_ohw) - 1);
+        int oh = q >> LG_OW_T;
+        const ushort* row = plane + (int64_t)b * HpWp
+                            + (oh * STRIDE + dh) * g.Wp + dw;
+#pragma unroll
+        for (int j = 0; j < SUBW; ++j)
+          breg[rr * SUBW + j] = row[j * STRIDE];
+      }
+    }
+  };
+  auto commit = [&](const ushort* breg, int buf) {
+#pragma unroll
+    for (int j = 0; j < CV6_BN / 8; ++j)
+      bT_lds[buf][(nn0 + j) * (CV6_BK + CV6_PAD) + kk] = (short)breg[j];
+  };
+
+  // prologue: k0 staged and published; k1 in flight in brB
+  gather(brA, 0);
+  commit(brA, 0);
+  if (CV6_BK < K) gather(brB, CV6_BK);
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += CV6_BK) {
+    __syncthreads();                      // buf[cur] (= step k0) ready
+    // issue step k0+2's loads into the set the MFMAs are done with
+    ushort* nxt2 = ((k0 / CV6_BK) & 1) ? brB : brA;     // holds k0's data
+    ushort* nxt1 = ((k0 / CV6_BK) & 1) ? brA : brB;     // holds k0+1's
+    if (k0 + 2 * CV6_BK < K) gather(nxt2, k0 + 2 * CV6_BK);
+    bf16x8 a;
+    {
+      uint4 av = *reinterpret_cast<const uint4*>(wrow + k0 + 8 * (lane >> 4));
+      a = *reinterpret_cast<const bf16x8*>(&av);
+      if (!arow_ok) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) a[e] = 0;
+      }
+    }
+#pragma unroll
+    for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur][(nt * 16 + (lane & 15)) * (CV6_BK + CV6_PAD)
+                       + 8 * (lane >> 4)]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    if (k0 + CV6_BK < K) commit(nxt1, cur ^ 1);   // k0+1: issued LAST iter
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        yc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // fwd v7: BK=64 — two MFMA sub-steps per barrier (16 MFMAs/wave/step,
 // amortising the stage/barrier latency the BK=32 pipeline pays every
 // 8 MFMAs; cdna_hip_programming.md: BK 32->64 = +7..16% on the dense
@@ -745,6 +869,22 @@ extern "C" void ols_conv3x3_fwd_p(const void* xp, const void* w, void* y,
   // BK=64 (v7) wins +5..10% at OW >= 8 stride 1 (A/B: gpurun_out/
   // v7_fwd.log vs v6_fwd.log at C=250); the 4x4-plane deep layers and
   // stride 2 measured flat-to-worse, so they stay on BK=32.
+  // OLSIM_CONV_V8 forces the two-ahead BK=32 pipeline everywhere (A/B).
+#define LAUNCH_FWD8(LG, ST) \
+  hipLaunchKernelGGL((k_conv3x3_fwd_v8<LG, ST>), grid, dim3(CONV_THREADS), \
+                     0, stream, xb, wb, yb, g)
+  if (getenv("OLSIM_CONV_V8") != nullptr) {
+    if (stride == 1) {
+      if (g.lg_ow >= 4) LAUNCH_FWD8(4, 1);
+      else if (g.lg_ow == 3) LAUNCH_FWD8(3, 1);
+      else LAUNCH_FWD8(2, 1);
+    } else {
+      if (g.lg_ow >= 4) LAUNCH_FWD8(4, 2);
+      else if (g.lg_ow == 3) LAUNCH_FWD8(3, 2);
+      else LAUNCH_FWD8(2, 2);
+    }
+    return;
+  }
   const bool v7 = ((IC * 9) % 64 == 0) && stride == 1 && g.lg_ow >= 3;
   if (stride == 1) {
     if (v7 && g.lg_ow >= 5) LAUNCH_FWD7(5, 1);
@@ -760,6 +900,7 @@ extern "C" void ols_conv3x3_fwd_p(const void* xp, const void* w, void* y,
   }
 #undef LAUNCH_FWD6
 #undef LAUNCH_FWD7
+#undef LAUNCH_FWD8
 }
 
 extern "C" void ols_conv3x3_dgrad_p(const void* dyp, const void* w, void* dx,
