@@ -187,11 +187,12 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v6(
 // ---------------------------------------------------------------------------
 // fwd v8: BK=32 with a TWO-iteration gather lookahead — two register
 // sets; the loads a commit waits for were issued a FULL iteration
-// earlier (PMC: the one-ahead v6 pipeline is 73% SQ_WAIT_ANY — the 8
-// MFMAs per step cannot cover the gather latency under saturation).
-// Same LDS (2 buffers) and near-same VGPR count, so occupancy is
-// unchanged; hipcc's per-register vmcnt bookkeeping gives the commit a
-// counted wait that leaves the newer set's loads in flight.
+// earlier.  MEASURED NEGATIVE (kept env-gated as the record of a
+// rejected hypothesis): -2% (s0) to -42% (s3) vs the one-ahead v6
+// (gpurun_out/v8_fwd.log vs v6_fwd.log) — the extra live registers
+// and hipcc's conservative waits outweigh the longer load window; the
+// 73% SQ_WAIT_ANY is saturation of the shared memory system, which a
+// deeper per-wave pipeline cannot hide.
 template <int LG_OW_T, int STRIDE>
 __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_fwd_v8(
     const __hip_bfloat16* __restrict__ xp, const __hip_bfloat16* __restrict__ w,
