@@ -65,15 +65,15 @@ class LeNet5(ClientBatchedModel):
         return blinear(h, params["fc3.w"], params["fc3.b"])
 
     def forward_cbf(self, params: Params, x: torch.Tensor) -> torch.Tensor:
-        from ..ops.conv import client_conv5x5
+        from ..ops.conv import client_conv5x5, max_pool2x2
         C, B = x.shape[0], x.shape[1]
         h = x.permute(0, 2, 1, 3, 4).contiguous()          # [C, 3, B, 32, 32]
         h = client_conv5x5(h, params["conv1.w"], params["conv1.b"],
                            relu=True)                      # [C, 6, B, 28, 28]
-        h = F.max_pool2d(h.view(C * 6, B, 28, 28), 2).view(C, 6, B, 14, 14)
+        h = max_pool2x2(h)                                 # [C, 6, B, 14, 14]
         h = client_conv5x5(h, params["conv2.w"], params["conv2.b"],
                            relu=True)                      # [C, 16, B, 10, 10]
-        h = F.max_pool2d(h.view(C * 16, B, 10, 10), 2)     # [C*16, B, 5, 5]
+        h = max_pool2x2(h)                                 # [C, 16, B, 5, 5]
         h = h.view(C, 16, B, 25).permute(0, 2, 1, 3).reshape(C, B, 400)
         h = F.relu(blinear(h, params["fc1.w"], params["fc1.b"]))
         h = F.relu(blinear(h, params["fc2.w"], params["fc2.b"]))

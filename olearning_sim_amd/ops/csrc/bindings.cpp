@@ -62,6 +62,13 @@ extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, float* dw,
                                   const int* ntab, int C, int IC, int OC,
                                   int B, int H, int W, hipStream_t stream);
 
+extern "C" void ols_pool2x2_fwd(const void* x, void* y, unsigned char* arg,
+                                int64_t planes, int OH, int OW, int dtype,
+                                hipStream_t stream);
+extern "C" void ols_pool2x2_bwd(const void* dy, const unsigned char* arg,
+                                void* dx, int64_t planes, int OH, int OW,
+                                int dtype, hipStream_t stream);
+
 extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const void* dy, void* dx, void* dres,
                                   const float* mean, const float* rstd,
@@ -359,6 +366,41 @@ at::Tensor conv3x3_wgrad_p(at::Tensor xp, at::Tensor dy, int64_t stride) {
   return dw;
 }
 
+// ---- 2x2 max pool (pool2x2.hip) -----------------------------------------
+// x: [..., H, W] with H, W even; pooled over the last two dims
+
+std::tuple<at::Tensor, at::Tensor> pool2x2_fwd(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() >= 2);
+  int H = x.size(-2), W = x.size(-1);
+  TORCH_CHECK(H % 2 == 0 && W % 2 == 0);
+  int64_t planes = x.numel() / ((int64_t)H * W);
+  auto sizes = x.sizes().vec();
+  sizes[sizes.size() - 2] = H / 2;
+  sizes[sizes.size() - 1] = W / 2;
+  auto y = at::empty(sizes, x.options());
+  auto arg = at::empty(sizes, x.options().dtype(at::kByte));
+  int dt = x.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_pool2x2_fwd(x.data_ptr(), y.data_ptr(),
+                  (unsigned char*)arg.data_ptr(), planes, H / 2, W / 2, dt,
+                  at::cuda::getCurrentCUDAStream().stream());
+  return {y, arg};
+}
+
+at::Tensor pool2x2_bwd(at::Tensor dy, at::Tensor arg) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() >= 2);
+  int OH = dy.size(-2), OW = dy.size(-1);
+  int64_t planes = dy.numel() / ((int64_t)OH * OW);
+  auto sizes = dy.sizes().vec();
+  sizes[sizes.size() - 2] = OH * 2;
+  sizes[sizes.size() - 1] = OW * 2;
+  auto dx = at::empty(sizes, dy.options());
+  int dt = dy.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_pool2x2_bwd(dy.data_ptr(), (unsigned char*)arg.contiguous().data_ptr(),
+                  dx.data_ptr(), planes, OH, OW, dt,
+                  at::cuda::getCurrentCUDAStream().stream());
+  return dx;
+}
+
 }  // namespace
 
 TORCH_LIBRARY(olsim_hip, m) {
@@ -384,6 +426,8 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("conv5x5_fwd(Tensor x, Tensor w, Tensor bias, Tensor ntab, bool relu) -> Tensor");
   m.def("conv5x5_dgrad(Tensor dyp, Tensor w, Tensor ntab, int H, int W) -> Tensor");
   m.def("conv5x5_wgrad(Tensor x, Tensor dy, Tensor ntab) -> Tensor");
+  m.def("pool2x2_fwd(Tensor x) -> (Tensor, Tensor)");
+  m.def("pool2x2_bwd(Tensor dy, Tensor arg) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
@@ -402,4 +446,6 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("conv5x5_fwd", &conv5x5_fwd);
   m.impl("conv5x5_dgrad", &conv5x5_dgrad);
   m.impl("conv5x5_wgrad", &conv5x5_wgrad);
+  m.impl("pool2x2_fwd", &pool2x2_fwd);
+  m.impl("pool2x2_bwd", &pool2x2_bwd);
 }
