@@ -173,6 +173,110 @@ def taskconfig_msg_to_json(msg) -> Dict:
     return jd
 
 
+def taskconfig_json_to_msg(raw: Dict):
+    """Canonical task JSON -> TaskConfig wire message (client-side
+    mirror of the reference's json2taskconfig, utils.py:831-1027).
+    Used by wire clients (examples/grpc_client.py, tests)."""
+    reg = registry()
+    TC = reg.msg("taskService.proto", "TaskConfig")
+    m = TC()
+    m.userID = raw.get("user_id", "")
+    m.taskID.taskID = raw.get("task_id", "")
+    tr = {"FILE": 0, "HTTP": 1, "S3": 2, "MINIO": 3}
+    tgt = raw.get("target", {})
+    m.target.priority = tgt.get("priority", 0)
+    for d in tgt.get("data", []):
+        td = m.target.targetData.add()
+        td.dataName = d.get("name", "")
+        td.dataPath = d.get("data_path", "")
+        td.dataSplitType = bool(d.get("data_split_type", False))
+        td.dataTransferType = tr.get(d.get("data_transfer_type", "FILE"), 0)
+        td.taskType = d.get("task_type", "")
+        ts_ = d.get("total_simulation", {})
+        td.totalSimulation.deviceTotalSimulation.extend(
+            ts_.get("devices", []))
+        td.totalSimulation.numTotalSimulation.extend(ts_.get("nums", []))
+        td.totalSimulation.dynamicNumTotalSimulation.extend(
+            ts_.get("dynamic_nums", []))
+        al = d.get("allocation", {})
+        td.allocation.optimization = bool(al.get("optimization", False))
+        td.allocation.allocationLogicalSimulation.extend(
+            al.get("logical_simulation", []))
+        td.allocation.allocationDeviceSimulation.extend(
+            al.get("device_simulation", []))
+        rr = al.get("running_response", {})
+        td.allocation.runningResponse.deviceRunningResponse.extend(
+            rr.get("devices", []))
+        td.allocation.runningResponse.numRunningResponse.extend(
+            rr.get("nums", []))
+    of = raw.get("operatorflow", {})
+    fs = of.get("flow_setting", {})
+    m.operatorFlow.flowSetting.round = fs.get("round", 1)
+
+    def _cond(dst, src):
+        dst.strategyCondition = src.get("strategy", "")
+        dst.waitInterval = int(src.get("wait_interval", 0) or 0)
+        dst.totalTimeout = int(src.get("total_timeout", 0) or 0)
+
+    start = fs.get("start", {})
+    stop = fs.get("stop", {})
+    _cond(m.operatorFlow.flowSetting.startCondition
+          .logicalSimulationStrategy, start.get("logical_simulation", {}))
+    _cond(m.operatorFlow.flowSetting.startCondition
+          .deviceSimulationStrategy, start.get("device_simulation", {}))
+    _cond(m.operatorFlow.flowSetting.stopCondition
+          .logicalSimulationStrategy, stop.get("logical_simulation", {}))
+    _cond(m.operatorFlow.flowSetting.stopCondition
+          .deviceSimulationStrategy, stop.get("device_simulation", {}))
+    for op_raw in of.get("operators", []):
+        op = m.operatorFlow.operator.add()
+        op.name = op_raw.get("name", "")
+        bc = op_raw.get("operation_behavior_controller", {})
+        op.operationBehaviorController.useController = bool(
+            bc.get("use_gradient_house", False))
+        op.operationBehaviorController.strategyBehaviorController = \
+            bc.get("strategy_gradient_house", "")
+        op.operationBehaviorController.outboundService = \
+            bc.get("outbound_service", "")
+        op.input.extend(op_raw.get("input", []) or [])
+        op.useData = bool(op_raw.get("use_data", False))
+        mj = op_raw.get("model", {})
+        op.model.useModel = bool(mj.get("use_model", False))
+        op.model.modelForTrain = bool(mj.get("model_for_train", False))
+        op.model.modelTransferType = tr.get(
+            mj.get("model_transfer_type", "S3"), 2)
+        op.model.modelPath = mj.get("model_path", "")
+        op.model.modelUpdateStyle = mj.get("model_update_style", "")
+        for attr, key in ((op.logicalSimulationOperatorInfo,
+                           "logical_simulation"),
+                          (op.deviceSimulationOperatorInfo,
+                           "device_simulation")):
+            si = op_raw.get(key, {})
+            attr.operatorTransferType = tr.get(
+                si.get("operator_transfer_type", "S3"), 2)
+            attr.operatorCodePath = si.get("operator_code_path", "")
+            attr.operatorEntryFile = si.get("operator_entry_file", "")
+            attr.operatorParams = si.get("operator_params", "")
+    lsim = raw.get("logical_simulation", {})
+    cu = lsim.get("computation_unit", {})
+    m.logicalSimulation.computationUnit.devicesUnit.extend(
+        cu.get("devices", []))
+    for s in cu.get("setting", []):
+        m.logicalSimulation.computationUnit.unitSetting.add().numCpus = \
+            s.get("num_cpus", 0)
+    for r in lsim.get("resource_request", []):
+        rr2 = m.logicalSimulation.resourceRequestLogicalSimulation.add()
+        rr2.dataNameResourceRequest = r.get("name", "")
+        rr2.deviceResourceRequest.extend(r.get("devices", []))
+        rr2.numResourceRequest.extend(r.get("num_request", []))
+    for r in raw.get("device_simulation", {}).get("resource_request", []):
+        rr3 = m.deviceSimulation.resourceRequestDeviceSimulation.add()
+        rr3.dataNameResourceRequest = r.get("name", "")
+        rr3.deviceResourceRequest.extend(r.get("devices", []))
+        rr3.numResourceRequest.extend(r.get("num_request", []))
+    return m
+
+
 def _unary(fn, req_cls):
     deser = (empty_pb2.Empty.FromString if req_cls is None
              else req_cls.FromString)
