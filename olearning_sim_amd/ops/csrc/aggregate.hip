@@ -118,11 +118,42 @@ __global__ __launch_bounds__(OLS_THREADS) void k_delta_accum_cpar(
   for (int e = 0; e < 8; ++e) atomicAdd(&delta[j + e], acc[e]);
 }
 
+// Scalar flavour of the client-parallel kernel for small params whose
+// n is not a multiple of 8 (e.g. a 450-element LeNet conv1.w at
+// C=6250 measured ~1.5 ms on the j-parallel kernel: 450 lanes walking
+// a 6250-deep serial latency chain).
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_delta_accum_cpar_s(
+    float* __restrict__ delta, const T* __restrict__ buf,
+    const T* __restrict__ master, const float* __restrict__ weights,
+    int64_t clients, int64_t n, float wsum, int ctile) {
+  const int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= n) return;
+  const int64_t c0 = (int64_t)blockIdx.y * ctile;
+  const int64_t c1 = min(clients, c0 + ctile);
+  float acc = 0.f;
+  for (int64_t c = c0; c < c1; ++c)
+    acc += weights[c] * to_f32(buf[c * n + j]);
+  if (blockIdx.y == 0) acc -= wsum * to_f32(master[j]);
+  atomicAdd(&delta[j], acc);
+}
+
 extern "C" void ols_weighted_delta_accum_flat(
     float* delta, const void* buf, const void* master, const float* weights,
     const int64_t* offs, int nblocks, int64_t clients, int64_t pglobal,
     float wsum, int dtype, hipStream_t stream) {
   dim3 block(OLS_THREADS);
+  if (nblocks == 1 && pglobal % 8 != 0 && dtype == 1 &&
+      pglobal < 131072 && clients >= 64) {
+    const int ctile = 64;
+    dim3 grid((unsigned)((pglobal + OLS_THREADS - 1) / OLS_THREADS),
+              (unsigned)((clients + ctile - 1) / ctile));
+    hipLaunchKernelGGL((k_delta_accum_cpar_s<__hip_bfloat16>), grid, block,
+                       0, stream, delta, (const __hip_bfloat16*)buf,
+                       (const __hip_bfloat16*)master, weights, clients,
+                       pglobal, wsum, ctile);
+    return;
+  }
   if (nblocks == 1 && pglobal % 8 == 0 && dtype == 1 &&
       pglobal / 8 < 16384 && clients >= 64) {
     // small param: client-parallel tiles (fills the chip; the j-only

@@ -211,7 +211,7 @@ def test_engine_per_tier_accounting_gpu():
     assert eng.master.flat.isfinite().all()
 
 
-@pytest.mark.parametrize("n", [4096, 36864, 100])
+@pytest.mark.parametrize("n", [4096, 36864, 100, 450, 10])
 def test_weighted_delta_accum_single_param(n):
     """Single-param call path (nblocks==1): n%8==0 rides the vectorised
     k_delta_accum_v8 kernel, n%8!=0 the scalar fallback."""
