@@ -126,11 +126,17 @@ class _BLinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
-            wt = w.transpose(1, 2)
-            dx = torch.bmm(dy, wt.contiguous() if _BMM_SAFE else wt)
+            if _BMM_SAFE:
+                from ..ops.fused import fast_transpose
+                dx = torch.bmm(dy, fast_transpose(w))
+            else:
+                dx = torch.bmm(dy, w.transpose(1, 2))
         if ctx.needs_input_grad[1]:
-            xt = x.transpose(1, 2)
-            dw = torch.bmm(xt.contiguous() if _BMM_SAFE else xt, dy)
+            if _BMM_SAFE:
+                from ..ops.fused import fast_transpose
+                dw = torch.bmm(fast_transpose(x), dy)
+            else:
+                dw = torch.bmm(x.transpose(1, 2), dy)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = dy.sum(dim=1)
         return dx, dw, db

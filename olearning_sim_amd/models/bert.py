@@ -42,7 +42,8 @@ class _TiedHeadCE(torch.autograd.Function):
         # probe_sliced reproduces it on this stack — full-tensor
         # transposed views are fine).  Materialise tok^T once; its
         # non-transposed slices are safe bmm operands.
-        tok_t = tok.transpose(1, 2).contiguous()        # [C, H, V]
+        from ..ops.fused import fast_transpose
+        tok_t = fast_transpose(tok)                     # [C, H, V]
         m = torch.full((C, N), float("-inf"), device=hs.device,
                        dtype=torch.float32)
         l = torch.zeros(C, N, device=hs.device, dtype=torch.float32)
@@ -196,7 +197,8 @@ class BertLM(ClientBatchedModel):
         # The transpose is materialised and the GEMM goes through
         # blinear so fwd AND bwd operands are contiguous (hipBLASLt
         # strided-view fault, models/base.py _BLinearFn).
-        tok_t = params["emb.tok"].transpose(1, 2).contiguous()
+        from ..ops.fused import fast_transpose
+        tok_t = fast_transpose(params["emb.tok"])
         logits = blinear(hs, tok_t, params["head.bias"])
         return logits.view(C, B, L, self.vocab_size)
 

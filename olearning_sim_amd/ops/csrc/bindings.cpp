@@ -69,6 +69,9 @@ extern "C" void ols_pool2x2_bwd(const void* dy, const unsigned char* arg,
                                 void* dx, int64_t planes, int OH, int OW,
                                 int dtype, hipStream_t stream);
 
+extern "C" void ols_transpose2d(const void* in, void* out, int64_t B, int M,
+                                int N, int dtype, hipStream_t stream);
+
 extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const void* dy, void* dx, void* dres,
                                   const float* mean, const float* rstd,
@@ -401,6 +404,21 @@ at::Tensor pool2x2_bwd(at::Tensor dy, at::Tensor arg) {
   return dx;
 }
 
+// ---- batched 2-D transpose (transpose.hip) ------------------------------
+
+at::Tensor transpose2d(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 ||
+              x.scalar_type() == at::kFloat);
+  int64_t B = x.size(0);
+  int M = x.size(1), N = x.size(2);
+  auto y = at::empty({B, (int64_t)N, (int64_t)M}, x.options());
+  int dt = x.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_transpose2d(x.data_ptr(), y.data_ptr(), B, M, N, dt,
+                  at::cuda::getCurrentCUDAStream().stream());
+  return y;
+}
+
 }  // namespace
 
 TORCH_LIBRARY(olsim_hip, m) {
@@ -428,6 +446,7 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("conv5x5_wgrad(Tensor x, Tensor dy, Tensor ntab) -> Tensor");
   m.def("pool2x2_fwd(Tensor x) -> (Tensor, Tensor)");
   m.def("pool2x2_bwd(Tensor dy, Tensor arg) -> Tensor");
+  m.def("transpose2d(Tensor x) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
@@ -448,4 +467,5 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("conv5x5_wgrad", &conv5x5_wgrad);
   m.impl("pool2x2_fwd", &pool2x2_fwd);
   m.impl("pool2x2_bwd", &pool2x2_bwd);
+  m.impl("transpose2d", &transpose2d);
 }

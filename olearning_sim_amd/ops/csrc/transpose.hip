@@ -1,0 +1,71 @@
+// Batched 2-D transpose: out[b][n][m] = in[b][m][n] (bf16/fp32).
+//
+// torch's transpose+contiguous runs the generic strided copy at
+// 0.6-1.7 TB/s on the BERT shapes (profiles/bert_tiedhead_r02.md:
+// ~87 ms/round of aten::copy_ is transposes — the tied-head tok^T is
+// 19 ms alone).  Classic LDS-tiled transpose: 64x64 tiles, 16-byte
+// reads AND writes, padded LDS rows; edge tiles take the scalar path.
+
+#include "common.h"
+
+#define TR_TILE 64
+#define TR_PAD 8           // shorts; keeps write-side b128 reads off one bank
+
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_transpose2d(
+    const T* __restrict__ in, T* __restrict__ out, int M, int N) {
+  __shared__ T tile[TR_TILE][TR_TILE + TR_PAD];
+  const int64_t b = blockIdx.z;
+  const int m0 = blockIdx.y * TR_TILE;
+  const int n0 = blockIdx.x * TR_TILE;
+  const T* src = in + b * (int64_t)M * N;
+  T* dst = out + b * (int64_t)M * N;
+
+  const bool full = (m0 + TR_TILE <= M) && (n0 + TR_TILE <= N);
+  if (full) {
+    // 256 threads x 8 elems = 2048 per pass; 2 passes read the tile
+    const int c0 = (threadIdx.x % 8) * 8;
+    const int r = threadIdx.x / 8;             // 0..31
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      Pack<T, 8> v = *reinterpret_cast<const Pack<T, 8>*>(
+          &src[(int64_t)(m0 + r + 32 * p) * N + n0 + c0]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) tile[r + 32 * p][c0 + e] = v.v[e];
+    }
+    __syncthreads();
+    // write rows of OUT (= columns of the tile), 16 B per store
+    const int rn = threadIdx.x / 8;            // out row within tile
+    const int cm0 = (threadIdx.x % 8) * 8;
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      Pack<T, 8> v;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) v.v[e] = tile[cm0 + e][rn + 32 * p];
+      *reinterpret_cast<Pack<T, 8>*>(
+          &dst[(int64_t)(n0 + rn + 32 * p) * M + m0 + cm0]) = v;
+    }
+    return;
+  }
+  // edge tile: scalar
+  for (int i = threadIdx.x; i < TR_TILE * TR_TILE; i += blockDim.x) {
+    int r = i / TR_TILE, c = i % TR_TILE;
+    int m = m0 + r, n = n0 + c;
+    if (m < M && n < N)
+      dst[(int64_t)n * M + m] = src[(int64_t)m * N + n];
+  }
+}
+
+extern "C" void ols_transpose2d(const void* in, void* out, int64_t B, int M,
+                                int N, int dtype, hipStream_t stream) {
+  dim3 grid((N + TR_TILE - 1) / TR_TILE, (M + TR_TILE - 1) / TR_TILE,
+            (unsigned)B);
+  dim3 block(OLS_THREADS);
+  if (dtype == 1)
+    hipLaunchKernelGGL((k_transpose2d<__hip_bfloat16>), grid, block, 0,
+                       stream, (const __hip_bfloat16*)in,
+                       (__hip_bfloat16*)out, M, N);
+  else
+    hipLaunchKernelGGL((k_transpose2d<float>), grid, block, 0, stream,
+                       (const float*)in, (float*)out, M, N);
+}

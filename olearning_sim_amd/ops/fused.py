@@ -286,3 +286,15 @@ def groupnorm_act(x: torch.Tensor, clients: int, groups: int,
     if res is not None:
         y = y + res
     return torch.nn.functional.relu(y) if relu else y
+
+
+def fast_transpose(x: torch.Tensor) -> torch.Tensor:
+    """[B, M, N] -> contiguous [B, N, M] via the LDS-tiled transpose
+    kernel (torch's transpose+contiguous runs the generic strided copy
+    at a fraction of the roofline on large batched matrices)."""
+    if x.is_cuda and x.dim() == 3 and \
+            x.dtype in (torch.bfloat16, torch.float32):
+        ops = load_hip_ops()
+        if ops is not None:
+            return ops.transpose2d(x.contiguous())
+    return x.transpose(1, 2).contiguous()

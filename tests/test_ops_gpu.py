@@ -228,3 +228,17 @@ def test_weighted_delta_accum_single_param(n):
     fused.weighted_delta_accum_flat(delta, buf, master, w, C, offsets=offs,
                                     wsum=float(w.sum()))
     torch.testing.assert_close(delta, ref, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("shape", [(3, 64, 128), (2, 30522, 768),
+                                   (5, 100, 70), (1, 64, 64)])
+def test_transpose2d_matches_torch(shape):
+    """LDS-tiled batched transpose == transpose().contiguous()."""
+    from olearning_sim_amd.ops.fused import fast_transpose
+    torch.manual_seed(5)
+    for dt in (torch.bfloat16, torch.float32):
+        x = torch.randn(*shape, device="cuda").to(dt)
+        y = fast_transpose(x)
+        ref = x.transpose(1, 2).contiguous()
+        assert y.shape == ref.shape and y.is_contiguous()
+        torch.testing.assert_close(y, ref, atol=0, rtol=0)
