@@ -197,6 +197,11 @@ def bgroupnorm(x: torch.Tensor, clients: int, num_groups: int,
 def blayernorm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
                eps: float = 1e-5) -> torch.Tensor:
     """Per-client LayerNorm: x [C,B,...,F]; weight/bias [C,F]."""
+    if x.is_cuda:
+        from ..ops.fused import layernorm
+        out = layernorm(x, weight.contiguous(), bias.contiguous(), eps)
+        if out is not None:
+            return out
     mean = x.mean(dim=-1, keepdim=True)
     var = x.var(dim=-1, unbiased=False, keepdim=True)
     xn = (x - mean) * torch.rsqrt(var + eps)

@@ -72,6 +72,18 @@ extern "C" void ols_pool2x2_bwd(const void* dy, const unsigned char* arg,
 extern "C" void ols_transpose2d(const void* in, void* out, int64_t B, int M,
                                 int N, int dtype, hipStream_t stream);
 
+extern "C" void ols_layernorm_fwd(const void* x, const void* gamma,
+                                  const void* beta, void* y, float* mean,
+                                  float* rstd, int64_t rows, int H,
+                                  int64_t rows_per_client, float eps,
+                                  int dtype, hipStream_t stream);
+extern "C" void ols_layernorm_bwd(const void* x, const void* dy,
+                                  const void* gamma, const float* mean,
+                                  const float* rstd, void* dx, float* dgamma,
+                                  float* dbeta, int64_t rows, int H,
+                                  int64_t rows_per_client, int dtype,
+                                  hipStream_t stream);
+
 extern "C" void ols_groupnorm_bwd(const void* x, const void* y,
                                   const void* dy, void* dx, void* dres,
                                   const float* mean, const float* rstd,
@@ -404,6 +416,47 @@ at::Tensor pool2x2_bwd(at::Tensor dy, at::Tensor arg) {
   return dx;
 }
 
+// ---- per-client LayerNorm (layernorm.hip) -------------------------------
+// x [C, N, H] (N rows per client); gamma/beta [C, H]
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> layernorm_fwd(
+    at::Tensor x, at::Tensor gamma, at::Tensor beta, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  TORCH_CHECK(gamma.is_contiguous() && beta.is_contiguous());
+  int64_t C = x.size(0), N = x.size(1);
+  int H = x.size(2);
+  TORCH_CHECK(H % 8 == 0 && N % 4 == 0, "layernorm_fwd: H%8, N%4");
+  int64_t rows = C * N;
+  auto y = at::empty_like(x);
+  auto mean = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  int dt = x.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_layernorm_fwd(x.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
+                    y.data_ptr(), mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), rows, H, N, (float)eps, dt,
+                    at::cuda::getCurrentCUDAStream().stream());
+  return {y, mean, rstd};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> layernorm_bwd(
+    at::Tensor x, at::Tensor dy, at::Tensor gamma, at::Tensor mean,
+    at::Tensor rstd) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  int64_t C = x.size(0), N = x.size(1);
+  int H = x.size(2);
+  int64_t rows = C * N;
+  auto dx = at::empty_like(x);
+  auto dgamma = at::zeros({C, (int64_t)H}, x.options().dtype(at::kFloat));
+  auto dbeta = at::zeros({C, (int64_t)H}, x.options().dtype(at::kFloat));
+  int dt = x.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_layernorm_bwd(x.data_ptr(), dy.data_ptr(),
+                    gamma.contiguous().data_ptr(), mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), dx.data_ptr(),
+                    dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), rows,
+                    H, N, dt, at::cuda::getCurrentCUDAStream().stream());
+  return {dx, dgamma, dbeta};
+}
+
 // ---- batched 2-D transpose (transpose.hip) ------------------------------
 
 at::Tensor transpose2d(at::Tensor x) {
@@ -447,6 +500,8 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("pool2x2_fwd(Tensor x) -> (Tensor, Tensor)");
   m.def("pool2x2_bwd(Tensor dy, Tensor arg) -> Tensor");
   m.def("transpose2d(Tensor x) -> Tensor");
+  m.def("layernorm_fwd(Tensor x, Tensor gamma, Tensor beta, float eps) -> (Tensor, Tensor, Tensor)");
+  m.def("layernorm_bwd(Tensor x, Tensor dy, Tensor gamma, Tensor mean, Tensor rstd) -> (Tensor, Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
@@ -468,4 +523,6 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("pool2x2_fwd", &pool2x2_fwd);
   m.impl("pool2x2_bwd", &pool2x2_bwd);
   m.impl("transpose2d", &transpose2d);
+  m.impl("layernorm_fwd", &layernorm_fwd);
+  m.impl("layernorm_bwd", &layernorm_bwd);
 }

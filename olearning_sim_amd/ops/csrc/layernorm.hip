@@ -1,0 +1,183 @@
+// Fused per-client LayerNorm forward + backward (one wave per row).
+//
+// The BERT path runs 25 LayerNorms per layer-pass as composed torch
+// ops (mean/var/rsqrt/mul/add chains + Welford reduces) — ~15% of a
+// round after the transpose fix (profiles/bert_tiedhead_r02.md).  Here:
+// fwd is one kernel (row mean/var by wave reduce, 16 B/lane IO), bwd
+// is one kernel producing dx plus fp32 dgamma/dbeta partials
+// accumulated through per-workgroup LDS and one atomicAdd per column
+// per workgroup.
+//
+// Layout: x [R, H] rows (R = C*B*L), H % 8 == 0; gamma/beta [C, H]
+// with c = row / rows_per_client.
+
+#include "common.h"
+
+#define LN_ROWS 4              // rows (waves) per workgroup
+
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_ln_fwd(
+    const T* __restrict__ x, const T* __restrict__ gamma,
+    const T* __restrict__ beta, T* __restrict__ y,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    int64_t rows, int H, int64_t rows_per_client, float eps) {
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int64_t row = (int64_t)blockIdx.x * LN_ROWS + wave;
+  if (row >= rows) return;
+  const T* xr = x + row * H;
+  T* yr = y + row * H;
+  const int64_t c = row / rows_per_client;
+  const T* g = gamma + c * H;
+  const T* b = beta + c * H;
+
+  const int hv = H / 8;
+  float s1 = 0.f, s2 = 0.f;
+  for (int v = lane; v < hv; v += WAVE) {
+    Pack<T, 8> p = *reinterpret_cast<const Pack<T, 8>*>(&xr[v * 8]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float f = to_f32(p.v[e]);
+      s1 += f;
+      s2 += f * f;
+    }
+  }
+  s1 = wave_sum(s1);
+  s2 = wave_sum(s2);
+  const float mean = s1 / H;
+  const float var = fmaxf(s2 / H - mean * mean, 0.f);
+  const float rstd = rsqrtf(var + eps);
+  if (lane == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  for (int v = lane; v < hv; v += WAVE) {
+    Pack<T, 8> p = *reinterpret_cast<const Pack<T, 8>*>(&xr[v * 8]);
+    Pack<T, 8> pg = *reinterpret_cast<const Pack<T, 8>*>(&g[v * 8]);
+    Pack<T, 8> pb = *reinterpret_cast<const Pack<T, 8>*>(&b[v * 8]);
+    Pack<T, 8> po;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      po.v[e] = from_f32<T>((to_f32(p.v[e]) - mean) * rstd
+                            * to_f32(pg.v[e]) + to_f32(pb.v[e]));
+    *reinterpret_cast<Pack<T, 8>*>(&yr[v * 8]) = po;
+  }
+}
+
+#define LN_MAX_H 4096
+
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_ln_bwd(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    const T* __restrict__ gamma, const float* __restrict__ mean_in,
+    const float* __restrict__ rstd_in, T* __restrict__ dx,
+    float* __restrict__ dgamma, float* __restrict__ dbeta,
+    int64_t rows, int H, int64_t rows_per_client) {
+  extern __shared__ __attribute__((aligned(16))) float col_acc[];
+  float* dg_acc = col_acc;            // [H]
+  float* db_acc = col_acc + H;        // [H]
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    dg_acc[i] = 0.f;
+    db_acc[i] = 0.f;
+  }
+  __syncthreads();
+
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int64_t row = (int64_t)blockIdx.x * LN_ROWS + wave;
+  const int hv = H / 8;
+  // all live rows of a block share one client (launcher requires
+  // rows_per_client % LN_ROWS == 0); derive it from the block's first
+  // row so tail-block threads never write another client's columns
+  const int64_t c = ((int64_t)blockIdx.x * LN_ROWS) / rows_per_client;
+  if (row < rows) {
+    const T* xr = x + row * H;
+    const T* dyr = dy + row * H;
+    T* dxr = dx + row * H;
+    const T* g = gamma + c * H;
+    const float mean = mean_in[row], rstd = rstd_in[row];
+
+    float s1 = 0.f, s2 = 0.f;
+    for (int v = lane; v < hv; v += WAVE) {
+      Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xr[v * 8]);
+      Pack<T, 8> pd = *reinterpret_cast<const Pack<T, 8>*>(&dyr[v * 8]);
+      Pack<T, 8> pg = *reinterpret_cast<const Pack<T, 8>*>(&g[v * 8]);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float grad = to_f32(pd.v[e]);
+        float xhat = (to_f32(px.v[e]) - mean) * rstd;
+        float dxhat = grad * to_f32(pg.v[e]);
+        s1 += dxhat;
+        s2 += dxhat * xhat;
+        atomicAdd(&dg_acc[v * 8 + e], grad * xhat);
+        atomicAdd(&db_acc[v * 8 + e], grad);
+      }
+    }
+    s1 = wave_sum(s1);
+    s2 = wave_sum(s2);
+    const float m1 = s1 / H, m2 = s2 / H;
+    for (int v = lane; v < hv; v += WAVE) {
+      Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xr[v * 8]);
+      Pack<T, 8> pd = *reinterpret_cast<const Pack<T, 8>*>(&dyr[v * 8]);
+      Pack<T, 8> pg = *reinterpret_cast<const Pack<T, 8>*>(&g[v * 8]);
+      Pack<T, 8> po;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float grad = to_f32(pd.v[e]);
+        float xhat = (to_f32(px.v[e]) - mean) * rstd;
+        float dxhat = grad * to_f32(pg.v[e]);
+        po.v[e] = from_f32<T>(rstd * (dxhat - m1 - xhat * m2));
+      }
+      *reinterpret_cast<Pack<T, 8>*>(&dxr[v * 8]) = po;
+    }
+  }
+  __syncthreads();
+  // one atomic per column per workgroup; all LN_ROWS rows of a block
+  // belong to the same client when rows_per_client % LN_ROWS == 0
+  // (the launcher guarantees it by padding the grid per client)
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    atomicAdd(&dgamma[c * H + i], dg_acc[i]);
+    atomicAdd(&dbeta[c * H + i], db_acc[i]);
+  }
+}
+
+extern "C" void ols_layernorm_fwd(const void* x, const void* gamma,
+                                  const void* beta, void* y, float* mean,
+                                  float* rstd, int64_t rows, int H,
+                                  int64_t rows_per_client, float eps,
+                                  int dtype, hipStream_t stream) {
+  dim3 grid((unsigned)((rows + LN_ROWS - 1) / LN_ROWS));
+  dim3 block(LN_ROWS * WAVE);
+  if (dtype == 1)
+    hipLaunchKernelGGL((k_ln_fwd<__hip_bfloat16>), grid, block, 0, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)gamma,
+                       (const __hip_bfloat16*)beta, (__hip_bfloat16*)y, mean,
+                       rstd, rows, H, rows_per_client, eps);
+  else
+    hipLaunchKernelGGL((k_ln_fwd<float>), grid, block, 0, stream,
+                       (const float*)x, (const float*)gamma,
+                       (const float*)beta, (float*)y, mean, rstd, rows, H,
+                       rows_per_client, eps);
+}
+
+extern "C" void ols_layernorm_bwd(const void* x, const void* dy,
+                                  const void* gamma, const float* mean,
+                                  const float* rstd, void* dx, float* dgamma,
+                                  float* dbeta, int64_t rows, int H,
+                                  int64_t rows_per_client, int dtype,
+                                  hipStream_t stream) {
+  dim3 grid((unsigned)((rows + LN_ROWS - 1) / LN_ROWS));
+  dim3 block(LN_ROWS * WAVE);
+  size_t lds = 2 * (size_t)H * sizeof(float);
+  if (dtype == 1)
+    hipLaunchKernelGGL((k_ln_bwd<__hip_bfloat16>), grid, block, lds, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
+                       (const __hip_bfloat16*)gamma, mean, rstd,
+                       (__hip_bfloat16*)dx, dgamma, dbeta, rows, H,
+                       rows_per_client);
+  else
+    hipLaunchKernelGGL((k_ln_bwd<float>), grid, block, lds, stream,
+                       (const float*)x, (const float*)dy,
+                       (const float*)gamma, mean, rstd, (float*)dx, dgamma,
+                       dbeta, rows, H, rows_per_client);
+}

@@ -298,3 +298,40 @@ def fast_transpose(x: torch.Tensor) -> torch.Tensor:
         if ops is not None:
             return ops.transpose2d(x.contiguous())
     return x.transpose(1, 2).contiguous()
+
+
+class _LayerNormFn(torch.autograd.Function):
+    """Fused per-client LayerNorm over the trailing dim
+    (ops/csrc/layernorm.hip; one wave per row, fp32 dgamma/dbeta
+    partials through workgroup LDS)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        ops = load_hip_ops(required=True)
+        y, mean, rstd = ops.layernorm_fwd(x, gamma, beta, eps)
+        ctx.save_for_backward(x, gamma, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, mean, rstd = ctx.saved_tensors
+        ops = load_hip_ops(required=True)
+        dx, dgamma, dbeta = ops.layernorm_bwd(x, dy.contiguous(), gamma,
+                                              mean, rstd)
+        return dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
+
+
+def layernorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+              eps: float = 1e-5):
+    """Per-client LayerNorm: x [C, ..., H], gamma/beta [C, H].
+    Fused kernel on GPU when H%8==0 and rows-per-client %4==0;
+    None -> caller should use the composed fallback."""
+    shape = x.shape
+    C, H = shape[0], shape[-1]
+    n = x.numel() // (C * H)
+    if not (x.is_cuda and H % 8 == 0 and n % 4 == 0
+            and x.dtype in (torch.bfloat16, torch.float32)
+            and hip_ops_available()):
+        return None
+    y = _LayerNormFn.apply(x.reshape(C, n, H).contiguous(), gamma, beta, eps)
+    return y.view(shape)

@@ -242,3 +242,39 @@ def test_transpose2d_matches_torch(shape):
         ref = x.transpose(1, 2).contiguous()
         assert y.shape == ref.shape and y.is_contiguous()
         torch.testing.assert_close(y, ref, atol=0, rtol=0)
+
+
+@pytest.mark.parametrize("shape", [(3, 8, 768), (2, 512, 768), (5, 4, 128)])
+def test_layernorm_matches_torch(shape):
+    """Fused per-client LayerNorm fwd/bwd vs composed fp32 torch."""
+    from olearning_sim_amd.ops.fused import layernorm
+    torch.manual_seed(9)
+    C, N, H = shape
+    x0 = torch.randn(C, N, H) * 2 + 0.3
+    g0 = torch.randn(C, H) * 0.5 + 1.0
+    b0 = torch.randn(C, H) * 0.2
+    dy = torch.randn(C, N, H) * 0.1
+
+    xg = x0.to(torch.bfloat16).cuda().requires_grad_(True)
+    gg = g0.to(torch.bfloat16).cuda().requires_grad_(True)
+    bg = b0.to(torch.bfloat16).cuda().requires_grad_(True)
+    y = layernorm(xg, gg, bg)
+    assert y is not None
+    y.backward(dy.to(torch.bfloat16).cuda())
+
+    xr = x0.clone().requires_grad_(True)
+    gr = g0.clone().requires_grad_(True)
+    br = b0.clone().requires_grad_(True)
+    mean = xr.mean(dim=-1, keepdim=True)
+    var = xr.var(dim=-1, unbiased=False, keepdim=True)
+    yr = (xr - mean) * torch.rsqrt(var + 1e-5) * gr.unsqueeze(1) \
+        + br.unsqueeze(1)
+    yr.backward(dy)
+
+    torch.testing.assert_close(y.float().cpu(), yr, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(xg.grad.float().cpu(), xr.grad,
+                               atol=5e-2, rtol=8e-2)
+    torch.testing.assert_close(gg.grad.float().cpu(), gr.grad,
+                               atol=0.3, rtol=5e-2)
+    torch.testing.assert_close(bg.grad.float().cpu(), br.grad,
+                               atol=0.3, rtol=5e-2)
