@@ -559,6 +559,135 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v6(
 }
 
 // ---------------------------------------------------------------------------
+// dgrad v7: BK=64 (two MFMA sub-steps per barrier) — the fwd v7
+// structure applied to the dgrad pipeline (A flip-staged in LDS, B
+// padded gather).  LDS 53 KB -> 3 blocks/CU; A/B-gated by the
+// launcher's measured per-shape table.
+template <int LG_OW_T>
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v7(
+    const __hip_bfloat16* __restrict__ dyp, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ dx, ConvGeom6 g) {
+  constexpr int BK = 64;
+  int c, tile;
+  if (!xcd_remap6(blockIdx.x, g.C, g.tiles_m * g.tiles_n, c, tile)) return;
+  const int mt = tile / g.tiles_n;
+  const int m0 = mt * CV6_BM;
+  const int n0 = (tile - mt * g.tiles_n) * CV6_BN;
+
+  __shared__ short a_lds[2][CV6_BM * BK];
+  __shared__ short bT_lds[2][CV6_BN * (BK + CV6_PAD)];
+  const int K = g.OC * 9;
+  const int N = g.B * g.OH * g.OW;
+  const int HpWp = g.Hp * g.Wp;
+  const int64_t planeB = (int64_t)g.B * HpWp;
+  const ushort* dyc =
+      reinterpret_cast<const ushort*>(dyp) + (int64_t)c * g.OC * planeB;
+  const ushort* wc =
+      reinterpret_cast<const ushort*>(w) + (int64_t)c * g.OC * g.IC * 9;
+  __hip_bfloat16* dxc = dx + (int64_t)c * g.IC * N;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  f32x4 acc[CV6_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV6_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int kk = threadIdx.x & 63;            // B k-row
+  const int nn0 = (threadIdx.x >> 6) * 32;    // 32 cols / thread
+  const int amm = threadIdx.x / 4;            // A row (ic)
+  const int ak0 = (threadIdx.x % 4) * 16;     // 16 k / thread
+  const int aic = min(m0 + amm, g.IC - 1);
+  const bool aic_ok = (m0 + amm) < g.IC;
+  ushort breg[32];
+  ushort areg[16];
+
+  auto gather = [&](int k0) {
+    {
+      const int kb = k0 + ak0;
+      int oc0 = kb / 9;
+      int r0 = kb - oc0 * 9;
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        int rr = r0 + j;
+        int carry = (rr >= 18) ? 2 : (rr >= 9 ? 1 : 0);
+        int oc = oc0 + carry;
+        rr -= 9 * carry;
+        ushort v = wc[((int64_t)oc * g.IC + aic) * 9 + (8 - rr)];
+        areg[j] = aic_ok ? v : (ushort)0;
+      }
+    }
+    {
+      const int k = k0 + kk;
+      const int oc = k / 9, r = k - oc * 9;
+      const int dh = r / 3, dw = r - dh * 3;
+      const ushort* plane = dyc + (int64_t)oc * planeB;
+      constexpr int SUBW = 1 << (LG_OW_T < 5 ? LG_OW_T : 5);
+      constexpr int NROW = 32 / SUBW;
+#pragma unroll
+      for (int rr = 0; rr < NROW; ++rr) {
+        int n = min(n0 + nn0 + rr * SUBW, N - 1);
+        int b = n >> g.lg_ohw;
+        int q = n & ((1 << g.lg_ohw) - 1);
+        int oh = q >> g.lg_ow;
+        int ow0 = q & ((1 << g.lg_ow) - 1);
+        if (SUBW < 32) ow0 = 0;
+        else ow0 = min(ow0, g.OW - 32);
+        const ushort* row = plane + (int64_t)b * HpWp + (oh + dh) * g.Wp
+                            + ow0 + dw;
+#pragma unroll
+        for (int j = 0; j < SUBW; ++j) breg[rr * SUBW + j] = row[j];
+      }
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 16; ++j)
+      a_lds[buf][amm * BK + ak0 + j] = (short)areg[j];
+#pragma unroll
+    for (int j = 0; j < 32; ++j)
+      bT_lds[buf][(nn0 + j) * (BK + CV6_PAD) + kk] = (short)breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    if (k0 + BK < K) gather(k0 + BK);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &a_lds[cur][(wave * 16 + (lane & 15)) * BK + sub * 32
+                      + 8 * (lane >> 4)]);
+#pragma unroll
+      for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &bT_lds[cur][(nt * 16 + (lane & 15)) * (BK + CV6_PAD)
+                         + sub * 32 + 8 * (lane >> 4)]);
+        acc[nt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (k0 + BK < K) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+    int n = n0 + nt * 16 + (lane & 15);
+    if (n >= N) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.IC)
+        dxc[(int64_t)m * N + n] = from_f32<__hip_bfloat16>(acc[nt][r]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // dgrad stride 2, parity-decomposed: one launch per class (HE, WE).
 // Output positions (2i+HE, 2j+WE); valid taps dh with (HE+dh-1) even:
 // HE=0 -> {1}, HE=1 -> {0,2} (same for WE/dw).  Class-k = (oc, th, tw)
@@ -818,6 +947,114 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v6(
 }
 
 // ---------------------------------------------------------------------------
+// wgrad v7: BK=64 reduction steps (two MFMA sub-steps per barrier),
+// int32 gather offsets.  Requires NN % 64 == 0.
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v7(
+    const __hip_bfloat16* __restrict__ xp, const __hip_bfloat16* __restrict__ dy,
+    float* __restrict__ dw, ConvGeom6 g) {
+  constexpr int BK = 64;
+  int c, tile;
+  if (!xcd_remap6(blockIdx.x, g.C, g.tiles_m * g.tiles_n, c, tile)) return;
+  const int mt = tile / g.tiles_n;
+  const int m0 = mt * CV6_BM;                  // over OC
+  const int n0 = (tile - mt * g.tiles_n) * CV6_BN;   // over IC*9
+  __shared__ short bT_lds[2][CV6_BN * (BK + CV6_PAD)];
+  const int K9 = g.IC * 9;
+  const int NN = g.B * g.OH * g.OW;            // reduction
+  const int HpWp = g.Hp * g.Wp;
+  const int64_t planeB = (int64_t)g.B * HpWp;
+  const ushort* xc =
+      reinterpret_cast<const ushort*>(xp) + (int64_t)c * g.IC * planeB;
+  const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
+  float* dwc = dw + (int64_t)c * g.OC * K9;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int arow = m0 + wave * 16 + (lane & 15);
+  const __hip_bfloat16* dyrow = dyc + (int64_t)min(arow, g.OC - 1) * NN;
+  const bool arow_ok = arow < g.OC;
+
+  f32x4 acc[CV6_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV6_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int qq = threadIdx.x & 63;
+  const int nn0 = (threadIdx.x >> 6) * 32;     // 32 K9-cols / thread
+  ushort breg[32];
+
+  int off[32];                                 // plane offsets fit int32
+#pragma unroll
+  for (int j = 0; j < 32; ++j) {
+    int k = min(n0 + nn0 + j, K9 - 1);
+    int ic = k / 9, r = k - ic * 9;
+    int dh = r / 3, dw2 = r - dh * 3;
+    off[j] = (int)((int64_t)ic * planeB + dh * g.Wp + dw2);
+  }
+
+  auto gather = [&](int q0) {
+    int q = min(q0 + qq, NN - 1);
+    int b = q >> g.lg_ohw;
+    int p = q & ((1 << g.lg_ohw) - 1);
+    int oh = p >> g.lg_ow;
+    int ow = p & ((1 << g.lg_ow) - 1);
+    const ushort* base = xc + (int64_t)b * HpWp
+                         + (oh * g.stride) * g.Wp + ow * g.stride;
+#pragma unroll
+    for (int j = 0; j < 32; ++j) breg[j] = base[off[j]];
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int j = 0; j < 32; ++j)
+      bT_lds[buf][(nn0 + j) * (BK + CV6_PAD) + qq] = (short)breg[j];
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int q0 = 0; q0 < NN; q0 += BK) {
+    __syncthreads();
+    if (q0 + BK < NN) gather(q0 + BK);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+      bf16x8 a;
+      {
+        uint4 av = *reinterpret_cast<const uint4*>(
+            dyrow + q0 + sub * 32 + 8 * (lane >> 4));
+        a = *reinterpret_cast<const bf16x8*>(&av);
+        if (!arow_ok) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) a[e] = 0;
+        }
+      }
+#pragma unroll
+      for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &bT_lds[cur][(nt * 16 + (lane & 15)) * (BK + CV6_PAD)
+                         + sub * 32 + 8 * (lane >> 4)]);
+        acc[nt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (q0 + BK < NN) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+    int k = n0 + nt * 16 + (lane & 15);
+    if (k >= K9) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        dwc[(int64_t)m * K9 + k] = acc[nt][r];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // launchers
 
 static inline int cdiv6(int a, int b) { return (a + b - 1) / b; }
@@ -918,6 +1155,21 @@ extern "C" void ols_conv3x3_dgrad_p(const void* dyp, const void* w, void* dx,
     auto db = (const __hip_bfloat16*)dyp;
     auto wb = (const __hip_bfloat16*)w;
     auto xb = (__hip_bfloat16*)dx;
+    if (getenv("OLSIM_CONV_DW64") != nullptr && (OC * 9) % 64 == 0) {
+      if (g.lg_ow >= 5)
+        hipLaunchKernelGGL((k_conv3x3_dgrad_v7<5>), grid, dim3(CONV_THREADS),
+                           0, stream, db, wb, xb, g);
+      else if (g.lg_ow == 4)
+        hipLaunchKernelGGL((k_conv3x3_dgrad_v7<4>), grid, dim3(CONV_THREADS),
+                           0, stream, db, wb, xb, g);
+      else if (g.lg_ow == 3)
+        hipLaunchKernelGGL((k_conv3x3_dgrad_v7<3>), grid, dim3(CONV_THREADS),
+                           0, stream, db, wb, xb, g);
+      else
+        hipLaunchKernelGGL((k_conv3x3_dgrad_v7<2>), grid, dim3(CONV_THREADS),
+                           0, stream, db, wb, xb, g);
+      return;
+    }
     if (g.lg_ow >= 4)
       hipLaunchKernelGGL((k_conv3x3_dgrad_v6<4>), grid, dim3(CONV_THREADS),
                          0, stream, db, wb, xb, g);
@@ -966,6 +1218,13 @@ extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, float* dw,
   g.tiles_m = cdiv6(OC, CV6_BM);
   g.tiles_n = cdiv6(IC * 9, CV6_BN);
   dim3 grid(xcd_blocks6(C, g.tiles_m * g.tiles_n));
+  const int NN = B * g.OH * g.OW;
+  if (getenv("OLSIM_CONV_DW64") != nullptr && NN % 64 == 0) {
+    hipLaunchKernelGGL(k_conv3x3_wgrad_v7, grid, dim3(CONV_THREADS), 0,
+                       stream, (const __hip_bfloat16*)xp,
+                       (const __hip_bfloat16*)dy, dw, g);
+    return;
+  }
   hipLaunchKernelGGL(k_conv3x3_wgrad_v6, grid, dim3(CONV_THREADS), 0, stream,
                      (const __hip_bfloat16*)xp, (const __hip_bfloat16*)dy,
                      dw, g);
