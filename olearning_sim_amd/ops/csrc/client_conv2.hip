@@ -559,10 +559,11 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v6(
 }
 
 // ---------------------------------------------------------------------------
-// dgrad v7: BK=64 (two MFMA sub-steps per barrier) — the fwd v7
-// structure applied to the dgrad pipeline (A flip-staged in LDS, B
-// padded gather).  LDS 53 KB -> 3 blocks/CU; A/B-gated by the
-// launcher's measured per-shape table.
+// dgrad/wgrad v7: BK=64 (two MFMA sub-steps per barrier) — the fwd v7
+// structure applied to the backward pipelines.  MEASURED NEUTRAL
+// (dgrad, +9% only at s1d) to NEGATIVE (wgrad -5..-10%: the 37 KB LDS
+// halves residency); kept env-gated (OLSIM_CONV_DW64=1) as the A/B
+// record — gpurun_out/dw64.log vs v8_fwd.log baseline columns.
 template <int LG_OW_T>
 __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v7(
     const __hip_bfloat16* __restrict__ dyp, const __hip_bfloat16* __restrict__ w,
