@@ -50,6 +50,16 @@ class _Conv3x3Fn(torch.autograd.Function):
         return dx, dw, None
 
 
+def _pad(x: torch.Tensor, pad: int) -> torch.Tensor:
+    """Zero-pad the trailing two dims (single-pass kernel; torch's
+    constant_pad_nd is a fill + strided-copy double pass)."""
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 2 == 0:
+        ops = load_hip_ops()
+        if ops is not None:
+            return ops.pad2d(x.contiguous(), pad)
+    return F.pad(x, (pad, pad, pad, pad))
+
+
 class _Conv3x3PadFn(torch.autograd.Function):
     """v6 padded path: the kernels gather from a 1-element zero halo so
     the implicit-im2col reads carry no bounds masks (client_conv2.hip).
@@ -58,7 +68,7 @@ class _Conv3x3PadFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, stride):
         ops = load_hip_ops(required=True)
-        x_pad = F.pad(x, (1, 1, 1, 1))
+        x_pad = _pad(x, 1)
         y = ops.conv3x3_fwd_p(x_pad, w, stride)
         ctx.save_for_backward(x_pad, w)
         ctx.stride = stride
@@ -156,7 +166,7 @@ class _Conv5x5Fn(torch.autograd.Function):
             dy = dy * (y > 0).to(dy.dtype)
         dx = dw = None
         if ctx.needs_input_grad[0]:
-            dy_pad = F.pad(dy, (4, 4, 4, 4))
+            dy_pad = _pad(dy, 4)
             nt = _ntab(B, H, W, H + 4, W + 4, x.device)
             dx = ops.conv5x5_dgrad(dy_pad, w, nt, H, W)
         if ctx.needs_input_grad[1]:

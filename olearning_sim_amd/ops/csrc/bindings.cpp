@@ -72,6 +72,9 @@ extern "C" void ols_pool2x2_bwd(const void* dy, const unsigned char* arg,
 extern "C" void ols_transpose2d(const void* in, void* out, int64_t B, int M,
                                 int N, int dtype, hipStream_t stream);
 
+extern "C" void ols_pad2d(const void* in, void* out, int64_t planes, int H,
+                          int W, int pad, int dtype, hipStream_t stream);
+
 extern "C" void ols_layernorm_fwd(const void* x, const void* gamma,
                                   const void* beta, void* y, float* mean,
                                   float* rstd, int64_t rows, int H,
@@ -457,6 +460,25 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> layernorm_bwd(
   return {dx, dgamma, dbeta};
 }
 
+// ---- zero-pad trailing 2 dims (pad2d.hip) -------------------------------
+
+at::Tensor pad2d(at::Tensor x, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() >= 2);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 ||
+              x.scalar_type() == at::kHalf);
+  int H = x.size(-2), W = x.size(-1);
+  TORCH_CHECK(W % 2 == 0 && pad >= 1);
+  int64_t planes = x.numel() / ((int64_t)H * W);
+  auto sizes = x.sizes().vec();
+  sizes[sizes.size() - 2] = H + 2 * pad;
+  sizes[sizes.size() - 1] = W + 2 * pad;
+  auto y = at::empty(sizes, x.options());
+  int dt = x.scalar_type() == at::kBFloat16 ? 1 : 2;
+  ols_pad2d(x.data_ptr(), y.data_ptr(), planes, H, W, (int)pad, dt,
+            at::cuda::getCurrentCUDAStream().stream());
+  return y;
+}
+
 // ---- batched 2-D transpose (transpose.hip) ------------------------------
 
 at::Tensor transpose2d(at::Tensor x) {
@@ -500,6 +522,7 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("pool2x2_fwd(Tensor x) -> (Tensor, Tensor)");
   m.def("pool2x2_bwd(Tensor dy, Tensor arg) -> Tensor");
   m.def("transpose2d(Tensor x) -> Tensor");
+  m.def("pad2d(Tensor x, int pad) -> Tensor");
   m.def("layernorm_fwd(Tensor x, Tensor gamma, Tensor beta, float eps) -> (Tensor, Tensor, Tensor)");
   m.def("layernorm_bwd(Tensor x, Tensor dy, Tensor gamma, Tensor mean, Tensor rstd) -> (Tensor, Tensor, Tensor)");
 }
@@ -523,6 +546,7 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("pool2x2_fwd", &pool2x2_fwd);
   m.impl("pool2x2_bwd", &pool2x2_bwd);
   m.impl("transpose2d", &transpose2d);
+  m.impl("pad2d", &pad2d);
   m.impl("layernorm_fwd", &layernorm_fwd);
   m.impl("layernorm_bwd", &layernorm_bwd);
 }

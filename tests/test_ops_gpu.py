@@ -244,6 +244,21 @@ def test_transpose2d_matches_torch(shape):
         torch.testing.assert_close(y, ref, atol=0, rtol=0)
 
 
+@pytest.mark.parametrize("shape,pad", [
+    ((250, 16, 10, 32, 32), 1), ((250, 64, 10, 8, 8), 1),
+    ((100, 6, 20, 24, 24), 4), ((7, 3, 5, 14, 18), 1),
+])
+def test_pad2d_matches_fpad(shape, pad):
+    """Single-pass pad kernel == F.pad (bitwise, bf16)."""
+    import torch.nn.functional as F
+    from olearning_sim_amd.ops.conv import _pad
+    torch.manual_seed(11)
+    x = torch.randn(*shape, device="cuda").to(torch.bfloat16)
+    y = _pad(x, pad)
+    ref = F.pad(x, (pad, pad, pad, pad))
+    torch.testing.assert_close(y, ref, atol=0, rtol=0)
+
+
 @pytest.mark.parametrize("shape", [(3, 8, 768), (2, 512, 768), (5, 4, 128)])
 def test_layernorm_matches_torch(shape):
     """Fused per-client LayerNorm fwd/bwd vs composed fp32 torch."""
