@@ -53,7 +53,9 @@ class _Conv3x3Fn(torch.autograd.Function):
 def _pad(x: torch.Tensor, pad: int) -> torch.Tensor:
     """Zero-pad the trailing two dims (single-pass kernel; torch's
     constant_pad_nd is a fill + strided-copy double pass)."""
-    if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 2 == 0:
+    import os
+    if (x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 2 == 0
+            and os.environ.get("OLSIM_PAD", "") != "torch"):
         ops = load_hip_ops()
         if ops is not None:
             return ops.pad2d(x.contiguous(), pad)
