@@ -79,6 +79,19 @@ class SyntheticFederatedData:
         x = torch.empty((batch_size,) + self.input_shape,
                         device=self.device, dtype=torch.float32)
         x.normal_(generator=g)
+        n = x[0].numel()
+        if (x.is_cuda and dtype == torch.bfloat16 and n % 8 == 0):
+            from ..ops.fused import load_hip_ops
+            ops = load_hip_ops()
+            if ops is not None:
+                # fused broadcast + class shift + bf16 cast
+                # (replicate.hip k_synth_batch): the composed form is a
+                # stride-0 fp32 broadcast-add plus a second full-size
+                # cast copy (~40 ms/round on the flagship)
+                out = ops.synth_batch(x.view(batch_size, n),
+                                      y.reshape(-1).contiguous(),
+                                      0.1 / max(1, self.num_classes), -0.05)
+                return out.view(C, batch_size, *self.input_shape), y
         x = x.unsqueeze(0) + 0.1 * (
             y.float().reshape(C, batch_size, *([1] * len(self.input_shape)))
             / max(1, self.num_classes) - 0.5)

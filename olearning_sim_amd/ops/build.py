@@ -21,7 +21,8 @@ def build(verbose: bool = True) -> str:
     sources = [os.path.join(csrc, f) for f in
                ("bindings.cpp", "fused_update.hip", "aggregate.hip",
                 "cross_entropy.hip", "groupnorm.hip", "client_conv.hip",
-                "client_conv2.hip", "client_conv5.hip", "pool2x2.hip", "transpose.hip", "layernorm.hip", "pad2d.hip")]
+                "client_conv2.hip", "client_conv5.hip", "pool2x2.hip", "transpose.hip", "layernorm.hip", "pad2d.hip",
+               "replicate.hip")]
     from torch.utils.cpp_extension import load
     mod_path = load(
         name="olsim_hip_ops",

@@ -75,6 +75,13 @@ extern "C" void ols_transpose2d(const void* in, void* out, int64_t B, int M,
 extern "C" void ols_pad2d(const void* in, void* out, int64_t planes, int H,
                           int W, int pad, int dtype, hipStream_t stream);
 
+extern "C" void ols_replicate(const void* src, void* dst, int64_t clients,
+                              int64_t n, int dtype, hipStream_t stream);
+
+extern "C" void ols_synth_batch(const float* x, const int64_t* y, void* out,
+                                int64_t rows, int64_t batch, int64_t n,
+                                float s, float t, hipStream_t stream);
+
 extern "C" void ols_layernorm_fwd(const void* x, const void* gamma,
                                   const void* beta, void* y, float* mean,
                                   float* rstd, int64_t rows, int H,
@@ -460,6 +467,37 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> layernorm_bwd(
   return {dx, dgamma, dbeta};
 }
 
+// ---- client replica broadcast (replicate.hip) ---------------------------
+
+at::Tensor replicate(at::Tensor src, int64_t clients) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous() && clients >= 1);
+  TORCH_CHECK(src.numel() % 8 == 0);
+  TORCH_CHECK(src.scalar_type() == at::kBFloat16 ||
+              src.scalar_type() == at::kFloat);
+  auto sizes = src.sizes().vec();
+  sizes.insert(sizes.begin(), clients);
+  auto dst = at::empty(sizes, src.options());
+  int dt = src.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_replicate(src.data_ptr(), dst.data_ptr(), clients, src.numel(), dt,
+                at::cuda::getCurrentCUDAStream().stream());
+  return dst;
+}
+
+at::Tensor synth_batch(at::Tensor x, at::Tensor y, double s, double t) {
+  // x [B, n] fp32, y [rows] int64 (rows = C*B) -> out [rows, n] bf16
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 2);
+  TORCH_CHECK(x.scalar_type() == at::kFloat && x.size(1) % 8 == 0);
+  TORCH_CHECK(y.is_cuda() && y.is_contiguous() &&
+              y.scalar_type() == at::kLong);
+  int64_t rows = y.numel(), batch = x.size(0), n = x.size(1);
+  TORCH_CHECK(rows % batch == 0);
+  auto out = at::empty({rows, n}, x.options().dtype(at::kBFloat16));
+  ols_synth_batch(x.data_ptr<float>(), y.data_ptr<int64_t>(),
+                  out.data_ptr(), rows, batch, n, (float)s, (float)t,
+                  at::cuda::getCurrentCUDAStream().stream());
+  return out;
+}
+
 // ---- zero-pad trailing 2 dims (pad2d.hip) -------------------------------
 
 at::Tensor pad2d(at::Tensor x, int64_t pad) {
@@ -523,6 +561,8 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("pool2x2_bwd(Tensor dy, Tensor arg) -> Tensor");
   m.def("transpose2d(Tensor x) -> Tensor");
   m.def("pad2d(Tensor x, int pad) -> Tensor");
+  m.def("replicate(Tensor src, int clients) -> Tensor");
+  m.def("synth_batch(Tensor x, Tensor y, float s, float t) -> Tensor");
   m.def("layernorm_fwd(Tensor x, Tensor gamma, Tensor beta, float eps) -> (Tensor, Tensor, Tensor)");
   m.def("layernorm_bwd(Tensor x, Tensor dy, Tensor gamma, Tensor mean, Tensor rstd) -> (Tensor, Tensor, Tensor)");
 }
@@ -547,6 +587,8 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("pool2x2_bwd", &pool2x2_bwd);
   m.impl("transpose2d", &transpose2d);
   m.impl("pad2d", &pad2d);
+  m.impl("replicate", &replicate);
+  m.impl("synth_batch", &synth_batch);
   m.impl("layernorm_fwd", &layernorm_fwd);
   m.impl("layernorm_bwd", &layernorm_bwd);
 }

@@ -244,6 +244,34 @@ def test_transpose2d_matches_torch(shape):
         torch.testing.assert_close(y, ref, atol=0, rtol=0)
 
 
+@pytest.mark.parametrize("shape", [(64,), (512, 512, 3, 3), (100,), (16, 8)])
+def test_replicate_params_matches_clone(shape):
+    """Broadcast-replicate kernel == expand().clone() (bitwise); shapes
+    with numel % 8 != 0 take the torch fallback inside replicate_params."""
+    from olearning_sim_amd.engine.client_manager import replicate_params
+    torch.manual_seed(3)
+    for dt in (torch.bfloat16, torch.float32):
+        src = torch.randn(*shape, device="cuda").to(dt)
+        out = replicate_params({"p": src}, 7)["p"]
+        ref = src.unsqueeze(0).expand(7, *shape).contiguous()
+        assert out.requires_grad and out.data_ptr() != src.data_ptr()
+        torch.testing.assert_close(out.detach(), ref, atol=0, rtol=0)
+
+
+def test_synth_batch_matches_composed():
+    """Fused data-gen broadcast+shift+cast == the composed torch form."""
+    from olearning_sim_amd.ops.fused import load_hip_ops
+    ops = load_hip_ops(required=True)
+    torch.manual_seed(4)
+    B, C, n, K = 16, 25, 3072, 100
+    x = torch.randn(B, n, device="cuda")
+    y = torch.randint(0, K, (C, B), device="cuda")
+    out = ops.synth_batch(x, y.reshape(-1).contiguous(), 0.1 / K, -0.05)
+    ref = (x.unsqueeze(0) + 0.1 * (y.float().unsqueeze(2) / K - 0.5)) \
+        .to(torch.bfloat16)
+    torch.testing.assert_close(out.view(C, B, n), ref)
+
+
 @pytest.mark.parametrize("shape,pad", [
     ((250, 16, 10, 32, 32), 1), ((250, 64, 10, 8, 8), 1),
     ((100, 6, 20, 24, 24), 4), ((7, 3, 5, 14, 18), 1),
