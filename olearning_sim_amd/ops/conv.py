@@ -46,7 +46,7 @@ class _Conv3x3Fn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = ops.conv3x3_dgrad(dy, w, x.shape[3], x.shape[4], ctx.stride)
         if ctx.needs_input_grad[1]:
-            dw = ops.conv3x3_wgrad(x, dy, ctx.stride).to(w.dtype)
+            dw = ops.conv3x3_wgrad(x, dy, ctx.stride)
         return dx, dw, None
 
 
@@ -87,7 +87,7 @@ class _Conv3x3PadFn(torch.autograd.Function):
             dy_pad = F.pad(dy, (1, 1, 1, 1))
             dx = ops.conv3x3_dgrad_p(dy_pad, w, H, W, ctx.stride)
         if ctx.needs_input_grad[1]:
-            dw = ops.conv3x3_wgrad_p(x_pad, dy, ctx.stride).to(w.dtype)
+            dw = ops.conv3x3_wgrad_p(x_pad, dy, ctx.stride)
         return dx, dw, None
 
 
@@ -173,7 +173,7 @@ class _Conv5x5Fn(torch.autograd.Function):
             dx = ops.conv5x5_dgrad(dy_pad, w, nt, H, W)
         if ctx.needs_input_grad[1]:
             nt = _ntab(B, H - 4, W - 4, H, W, x.device)
-            dw = ops.conv5x5_wgrad(x, dy, nt).to(w.dtype)
+            dw = ops.conv5x5_wgrad(x, dy, nt)
         db = dy.sum(dim=(2, 3, 4)) if ctx.needs_input_grad[2] else None
         return dx, dw, db, None
 

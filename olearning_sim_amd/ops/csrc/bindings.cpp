@@ -35,7 +35,7 @@ extern "C" void ols_conv3x3_fwd(const void* x, const void* w, void* y, int C,
 extern "C" void ols_conv3x3_dgrad(const void* dy, const void* w, void* dx,
                                   int C, int IC, int OC, int B, int H, int W,
                                   int stride, hipStream_t stream);
-extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
+extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, void* dw,
                                   int C, int IC, int OC, int B, int H, int W,
                                   int stride, hipStream_t stream);
 
@@ -47,7 +47,7 @@ extern "C" void ols_conv3x3_fwd_p(const void* xp, const void* w, void* y,
 extern "C" void ols_conv3x3_dgrad_p(const void* dyp, const void* w, void* dx,
                                     int C, int IC, int OC, int B, int H,
                                     int W, int stride, hipStream_t stream);
-extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, float* dw,
+extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, void* dw,
                                     int C, int IC, int OC, int B, int H,
                                     int W, int stride, hipStream_t stream);
 
@@ -58,7 +58,7 @@ extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
 extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
                                   const int* ntab, int C, int IC, int OC,
                                   int B, int H, int W, hipStream_t stream);
-extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, float* dw,
+extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, void* dw,
                                   const int* ntab, int C, int IC, int OC,
                                   int B, int H, int W, hipStream_t stream);
 
@@ -279,8 +279,8 @@ at::Tensor conv3x3_wgrad(at::Tensor x, at::Tensor dy, int64_t stride) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
   int C = x.size(0), IC = x.size(1), B = x.size(2), H = x.size(3),
       W = x.size(4), OC = dy.size(1);
-  auto dw = at::empty({C, OC, IC, 3, 3}, x.options().dtype(at::kFloat));
-  ols_conv3x3_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), C, IC,
+  auto dw = at::empty({C, OC, IC, 3, 3}, x.options());
+  ols_conv3x3_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr(), C, IC,
                     OC, B, H, W, (int)stride,
                     at::cuda::getCurrentCUDAStream().stream());
   return dw;
@@ -371,8 +371,8 @@ at::Tensor conv5x5_wgrad(at::Tensor x, at::Tensor dy, at::Tensor ntab) {
   int C = x.size(0), IC = x.size(1), B = x.size(2), H = x.size(3),
       W = x.size(4), OC = dy.size(1);
   TORCH_CHECK(OC <= 16 && ((int64_t)B * (H - 4) * (W - 4)) % 32 == 0);
-  auto dw = at::empty({C, OC, IC, 5, 5}, x.options().dtype(at::kFloat));
-  ols_conv5x5_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(),
+  auto dw = at::empty({C, OC, IC, 5, 5}, x.options());
+  ols_conv5x5_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr(),
                     ntab.data_ptr<int>(), C, IC, OC, B, H, W,
                     at::cuda::getCurrentCUDAStream().stream());
   return dw;
@@ -384,8 +384,8 @@ at::Tensor conv3x3_wgrad_p(at::Tensor xp, at::Tensor dy, int64_t stride) {
       H = xp.size(3) - 2, W = xp.size(4) - 2, OC = dy.size(1);
   TORCH_CHECK(ols_conv3x3_v6_ok(IC, OC, B, H, W, (int)stride),
               "shape unsupported by the v6 conv path");
-  auto dw = at::empty({C, OC, IC, 3, 3}, xp.options().dtype(at::kFloat));
-  ols_conv3x3_wgrad_p(xp.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), C,
+  auto dw = at::empty({C, OC, IC, 3, 3}, xp.options());
+  ols_conv3x3_wgrad_p(xp.data_ptr(), dy.data_ptr(), dw.data_ptr(), C,
                       IC, OC, B, H, W, (int)stride,
                       at::cuda::getCurrentCUDAStream().stream());
   return dw;

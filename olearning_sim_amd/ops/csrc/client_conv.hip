@@ -269,7 +269,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad(
 // grid (ntiles_k9, ntiles_oc, C); K-loop over n.
 __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad(
     const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
-    float* __restrict__ dw, ConvGeom g) {
+    __hip_bfloat16* __restrict__ dw, ConvGeom g) {
   __shared__ short a_lds[CONV_BM * CONV_BK];   // dY tile [oc][n]
   __shared__ short b_lds[CONV_BK * CONV_BN];   // P^T tile [n][k9]
   const int c = blockIdx.z;
@@ -281,7 +281,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad(
   const int HW = g.H * g.W;
   const __hip_bfloat16* xc = x + (int64_t)c * g.IC * g.B * HW;
   const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
-  float* dwc = dw + (int64_t)c * g.OC * K9;
+  __hip_bfloat16* dwc = dw + (int64_t)c * g.OC * K9;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -336,7 +336,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad(
     for (int r = 0; r < 4; ++r) {
       int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
       if (m < g.OC)
-        dwc[(int64_t)m * K9 + k] = acc[nt][r];
+        dwc[(int64_t)m * K9 + k] = __float2bfloat16(acc[nt][r]);
     }
   }
 }
@@ -736,7 +736,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v5(
 // B = patch tile register-double-buffered; one barrier per q-step.
 __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v5(
     const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
-    float* __restrict__ dw, ConvGeom g) {
+    __hip_bfloat16* __restrict__ dw, ConvGeom g) {
   __shared__ short bT_lds[2][CV2_BN * (CV2_BK + CV2_PAD)];
   const int c = blockIdx.z;
   const int m0 = blockIdx.y * CV2_BM;          // over OC
@@ -747,7 +747,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v5(
   const int HW = g.H * g.W;
   const __hip_bfloat16* xc = x + (int64_t)c * g.IC * g.B * HW;
   const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
-  float* dwc = dw + (int64_t)c * g.OC * K9;
+  __hip_bfloat16* dwc = dw + (int64_t)c * g.OC * K9;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -828,7 +828,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v5(
     for (int r = 0; r < 4; ++r) {
       int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
       if (m < g.OC)
-        dwc[(int64_t)m * K9 + k] = acc[nt][r];
+        dwc[(int64_t)m * K9 + k] = __float2bfloat16(acc[nt][r]);
     }
   }
 }
@@ -906,7 +906,7 @@ extern "C" void ols_conv3x3_dgrad(const void* dy, const void* w, void* dx,
   }
 }
 
-extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
+extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, void* dw,
                                   int C, int IC, int OC, int B, int H, int W,
                                   int stride, hipStream_t stream) {
   ConvGeom g{B, H, W, (H + stride - 1) / stride, (W + stride - 1) / stride,
@@ -926,11 +926,11 @@ extern "C" void ols_conv3x3_wgrad(const void* x, const void* dy, float* dw,
     dim3 grid5(ceil_div(IC * 9, CV2_BN), ceil_div(OC, CV2_BM), C);
     hipLaunchKernelGGL(k_conv3x3_wgrad_v5, grid5, dim3(CONV_THREADS), 0,
                        stream, (const __hip_bfloat16*)x,
-                       (const __hip_bfloat16*)dy, dw, g);
+                       (const __hip_bfloat16*)dy, (__hip_bfloat16*)dw, g);
   } else {
     dim3 grid(ceil_div(IC * 9, CONV_BN), ceil_div(OC, CONV_BM), C);
     hipLaunchKernelGGL(k_conv3x3_wgrad, grid, dim3(CONV_THREADS), 0, stream,
                        (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
-                       dw, g);
+                       (__hip_bfloat16*)dw, g);
   }
 }

@@ -850,7 +850,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_s2(
 // q-loop entirely — per step only the (b,oh,ow) base changes.
 __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v6(
     const __hip_bfloat16* __restrict__ xp, const __hip_bfloat16* __restrict__ dy,
-    float* __restrict__ dw, ConvGeom6 g) {
+    __hip_bfloat16* __restrict__ dw, ConvGeom6 g) {
   int c, tile;
   if (!xcd_remap6(blockIdx.x, g.C, g.tiles_m * g.tiles_n, c, tile)) return;
   const int mt = tile / g.tiles_n;
@@ -864,7 +864,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v6(
   const ushort* xc =
       reinterpret_cast<const ushort*>(xp) + (int64_t)c * g.IC * planeB;
   const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
-  float* dwc = dw + (int64_t)c * g.OC * K9;
+  __hip_bfloat16* dwc = dw + (int64_t)c * g.OC * K9;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -942,7 +942,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v6(
     for (int r = 0; r < 4; ++r) {
       int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
       if (m < g.OC)
-        dwc[(int64_t)m * K9 + k] = acc[nt][r];
+        dwc[(int64_t)m * K9 + k] = __float2bfloat16(acc[nt][r]);
     }
   }
 }
@@ -952,7 +952,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v6(
 // int32 gather offsets.  Requires NN % 64 == 0.
 __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v7(
     const __hip_bfloat16* __restrict__ xp, const __hip_bfloat16* __restrict__ dy,
-    float* __restrict__ dw, ConvGeom6 g) {
+    __hip_bfloat16* __restrict__ dw, ConvGeom6 g) {
   constexpr int BK = 64;
   int c, tile;
   if (!xcd_remap6(blockIdx.x, g.C, g.tiles_m * g.tiles_n, c, tile)) return;
@@ -967,7 +967,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v7(
   const ushort* xc =
       reinterpret_cast<const ushort*>(xp) + (int64_t)c * g.IC * planeB;
   const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
-  float* dwc = dw + (int64_t)c * g.OC * K9;
+  __hip_bfloat16* dwc = dw + (int64_t)c * g.OC * K9;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -1050,7 +1050,7 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v7(
     for (int r = 0; r < 4; ++r) {
       int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
       if (m < g.OC)
-        dwc[(int64_t)m * K9 + k] = acc[nt][r];
+        dwc[(int64_t)m * K9 + k] = __float2bfloat16(acc[nt][r]);
     }
   }
 }
@@ -1211,7 +1211,7 @@ extern "C" void ols_conv3x3_dgrad_p(const void* dyp, const void* w, void* dx,
 #undef LAUNCH_S2
 }
 
-extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, float* dw,
+extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, void* dw,
                                     int C, int IC, int OC, int B, int H,
                                     int W, int stride, hipStream_t stream) {
   ConvGeom6 g;
@@ -1223,10 +1223,10 @@ extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, float* dw,
   if (getenv("OLSIM_CONV_DW64") != nullptr && NN % 64 == 0) {
     hipLaunchKernelGGL(k_conv3x3_wgrad_v7, grid, dim3(CONV_THREADS), 0,
                        stream, (const __hip_bfloat16*)xp,
-                       (const __hip_bfloat16*)dy, dw, g);
+                       (const __hip_bfloat16*)dy, (__hip_bfloat16*)dw, g);
     return;
   }
   hipLaunchKernelGGL(k_conv3x3_wgrad_v6, grid, dim3(CONV_THREADS), 0, stream,
                      (const __hip_bfloat16*)xp, (const __hip_bfloat16*)dy,
-                     dw, g);
+                     (__hip_bfloat16*)dw, g);
 }

@@ -244,7 +244,7 @@ __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_dgrad(
 // g: H,W := x plane; OH,OW := y plane; K := NN (reduction), KP unused.
 __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad(
     const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
-    float* __restrict__ dw, const int* __restrict__ ntab, ConvGeom5 g) {
+    __hip_bfloat16* __restrict__ dw, const int* __restrict__ ntab, ConvGeom5 g) {
   int c, tile;
   if (!xcd_remap5(blockIdx.x, g.C, g.tiles_n, c, tile)) return;
   const int n0 = tile * CV5_BN;               // over IC*25
@@ -256,7 +256,7 @@ __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad(
   const ushort* xc =
       reinterpret_cast<const ushort*>(x) + (int64_t)c * g.IC * planeB;
   const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
-  float* dwc = dw + (int64_t)c * g.OC * K25;
+  __hip_bfloat16* dwc = dw + (int64_t)c * g.OC * K25;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -325,7 +325,7 @@ __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad(
     for (int r = 0; r < 4; ++r) {
       int m = (lane >> 4) * 4 + r;
       if (m < g.OC)
-        dwc[(int64_t)m * K25 + k] = acc[t][r];
+        dwc[(int64_t)m * K25 + k] = __float2bfloat16(acc[t][r]);
     }
   }
 }
@@ -374,7 +374,7 @@ extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
                      (__hip_bfloat16*)dx, ntab, g);
 }
 
-extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, float* dw,
+extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, void* dw,
                                   const int* ntab, int C, int IC, int OC,
                                   int B, int H, int W, hipStream_t stream) {
   ConvGeom5 g;
@@ -385,5 +385,5 @@ extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, float* dw,
   dim3 grid(xcd_blocks5(C, g.tiles_n));
   hipLaunchKernelGGL(k_conv5x5_wgrad, grid, dim3(CV5_THREADS), 0, stream,
                      (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
-                     dw, ntab, g);
+                     (__hip_bfloat16*)dw, ntab, g);
 }
