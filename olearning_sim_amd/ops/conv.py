@@ -227,13 +227,35 @@ def max_pool2x2(x: torch.Tensor) -> torch.Tensor:
     return out.reshape(*shape[:-2], shape[-2] // 2, shape[-1] // 2)
 
 
+class _Subsample2Fn(torch.autograd.Function):
+    """Every-2nd-pixel subsample of the trailing dims (pool2x2.hip):
+    torch's slice backward is a full zero-fill plus a strided scatter
+    through the generic 5-D kernels; here bwd writes the input once."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ops = load_hip_ops(required=True)
+        ctx.hw = x.shape[-2:]
+        return ops.subsample2(x.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        ops = load_hip_ops(required=True)
+        H, W = ctx.hw
+        return ops.subsample2_bwd(dy.contiguous(), H, W)
+
+
 def client_conv1x1(x: torch.Tensor, w: torch.Tensor,
                    stride: int = 1) -> torch.Tensor:
     """1x1 conv = one batched GEMM: y[C,OC,n] = w[C,OC,IC] @ x[C,IC,n]."""
     C, IC, B, H, W = x.shape
     OC = w.shape[1]
     if stride != 1:
-        x = x[:, :, :, ::stride, ::stride].contiguous()
+        if (stride == 2 and x.is_cuda and H % 2 == 0 and W % 4 == 0
+                and x.dtype in (torch.bfloat16, torch.float32)):
+            x = _Subsample2Fn.apply(x)
+        else:
+            x = x[:, :, :, ::stride, ::stride].contiguous()
         H, W = x.shape[-2:]
     y = torch.bmm(w.reshape(C, OC, IC), x.reshape(C, IC, B * H * W))
     return y.view(C, OC, B, H, W)

@@ -69,6 +69,13 @@ extern "C" void ols_pool2x2_bwd(const void* dy, const unsigned char* arg,
                                 void* dx, int64_t planes, int OH, int OW,
                                 int dtype, hipStream_t stream);
 
+extern "C" void ols_subsample2_fwd(const void* x, void* y, int64_t planes,
+                                   int OH, int OW, int dtype,
+                                   hipStream_t stream);
+extern "C" void ols_subsample2_bwd(const void* dy, void* dx, int64_t planes,
+                                   int H, int W, int dtype,
+                                   hipStream_t stream);
+
 extern "C" void ols_transpose2d(const void* in, void* out, int64_t B, int M,
                                 int N, int dtype, hipStream_t stream);
 
@@ -467,6 +474,39 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> layernorm_bwd(
   return {dx, dgamma, dbeta};
 }
 
+// ---- 2x stride subsample (pool2x2.hip) ----------------------------------
+
+at::Tensor subsample2(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() >= 2);
+  int H = x.size(-2), W = x.size(-1);
+  TORCH_CHECK(H % 2 == 0 && W % 4 == 0);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 ||
+              x.scalar_type() == at::kFloat);
+  int64_t planes = x.numel() / ((int64_t)H * W);
+  auto sizes = x.sizes().vec();
+  sizes[sizes.size() - 2] = H / 2;
+  sizes[sizes.size() - 1] = W / 2;
+  auto y = at::empty(sizes, x.options());
+  int dt = x.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_subsample2_fwd(x.data_ptr(), y.data_ptr(), planes, H / 2, W / 2, dt,
+                     at::cuda::getCurrentCUDAStream().stream());
+  return y;
+}
+
+at::Tensor subsample2_bwd(at::Tensor dy, int64_t H, int64_t W) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() >= 2);
+  TORCH_CHECK(dy.size(-2) == H / 2 && dy.size(-1) == W / 2);
+  int64_t planes = dy.numel() / ((int64_t)(H / 2) * (W / 2));
+  auto sizes = dy.sizes().vec();
+  sizes[sizes.size() - 2] = H;
+  sizes[sizes.size() - 1] = W;
+  auto dx = at::empty(sizes, dy.options());
+  int dt = dy.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_subsample2_bwd(dy.data_ptr(), dx.data_ptr(), planes, (int)H, (int)W,
+                     dt, at::cuda::getCurrentCUDAStream().stream());
+  return dx;
+}
+
 // ---- client replica broadcast (replicate.hip) ---------------------------
 
 at::Tensor replicate(at::Tensor src, int64_t clients) {
@@ -562,6 +602,8 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("transpose2d(Tensor x) -> Tensor");
   m.def("pad2d(Tensor x, int pad) -> Tensor");
   m.def("replicate(Tensor src, int clients) -> Tensor");
+  m.def("subsample2(Tensor x) -> Tensor");
+  m.def("subsample2_bwd(Tensor dy, int H, int W) -> Tensor");
   m.def("synth_batch(Tensor x, Tensor y, float s, float t) -> Tensor");
   m.def("layernorm_fwd(Tensor x, Tensor gamma, Tensor beta, float eps) -> (Tensor, Tensor, Tensor)");
   m.def("layernorm_bwd(Tensor x, Tensor dy, Tensor gamma, Tensor mean, Tensor rstd) -> (Tensor, Tensor, Tensor)");
@@ -588,6 +630,8 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("transpose2d", &transpose2d);
   m.impl("pad2d", &pad2d);
   m.impl("replicate", &replicate);
+  m.impl("subsample2", &subsample2);
+  m.impl("subsample2_bwd", &subsample2_bwd);
   m.impl("synth_batch", &synth_batch);
   m.impl("layernorm_fwd", &layernorm_fwd);
   m.impl("layernorm_bwd", &layernorm_bwd);

@@ -244,6 +244,24 @@ def test_transpose2d_matches_torch(shape):
         torch.testing.assert_close(y, ref, atol=0, rtol=0)
 
 
+@pytest.mark.parametrize("shape", [(5, 64, 16, 32, 32), (2, 3, 4, 8, 8),
+                                   (7, 16, 16)])
+def test_subsample2_matches_slicing(shape):
+    """subsample2 fwd/bwd == x[..., ::2, ::2] slicing autograd."""
+    from olearning_sim_amd.ops.conv import _Subsample2Fn
+    torch.manual_seed(6)
+    for dt in (torch.bfloat16, torch.float32):
+        x = torch.randn(*shape, device="cuda").to(dt).requires_grad_(True)
+        xr = x.detach().clone().requires_grad_(True)
+        y = _Subsample2Fn.apply(x)
+        ref = xr[..., ::2, ::2]
+        torch.testing.assert_close(y, ref, atol=0, rtol=0)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        ref.backward(dy)
+        torch.testing.assert_close(x.grad, xr.grad, atol=0, rtol=0)
+
+
 @pytest.mark.parametrize("shape", [(64,), (512, 512, 3, 3), (100,), (16, 8)])
 def test_replicate_params_matches_clone(shape):
     """Broadcast-replicate kernel == expand().clone() (bitwise); shapes
