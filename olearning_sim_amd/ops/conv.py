@@ -84,7 +84,7 @@ class _Conv3x3PadFn(torch.autograd.Function):
         dx = dw = None
         H, W = x_pad.shape[3] - 2, x_pad.shape[4] - 2
         if ctx.needs_input_grad[0]:
-            dy_pad = F.pad(dy, (1, 1, 1, 1))
+            dy_pad = _pad(dy, 1)
             dx = ops.conv3x3_dgrad_p(dy_pad, w, H, W, ctx.stride)
         if ctx.needs_input_grad[1]:
             dw = ops.conv3x3_wgrad_p(x_pad, dy, ctx.stride)
