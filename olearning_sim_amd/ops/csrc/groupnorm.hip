@@ -284,6 +284,133 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_bwd(
   }
 }
 
+
+// One-global-pass backward: the group's x and (relu-masked) dy are
+// staged in LDS during the stats pass and replayed for the dx pass, so
+// x/dy stream from HBM once instead of twice (k_gn_bwd reads them in
+// both passes).  Fits when 4*n bytes of dynamic LDS are available
+// (n = cg*HW; every model group plane is <= 8K elements = 32 KB).
+template <typename T, bool HAS_RES, bool RELU, int LAYOUT>
+__global__ __launch_bounds__(OLS_THREADS) void k_gn_bwd_lds(
+    const T* __restrict__ x, const T* __restrict__ y,
+    const T* __restrict__ dy, T* __restrict__ dx, T* __restrict__ dres,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    const T* __restrict__ gamma, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, int Bn, int C, int ch, int G, int HW) {
+  const int cg = ch / G;
+  const int n = cg * HW;
+  const int group = blockIdx.x;
+  int c, g;
+  int64_t base, pstride;
+  if (LAYOUT == 0) {
+    g = group % G;
+    c = (group / G) % C;
+    base = (int64_t)group * n;
+    pstride = HW;
+  } else {
+    const int b = group % Bn;
+    g = (group / Bn) % G;
+    c = group / (Bn * G);
+    base = (((int64_t)c * ch + (int64_t)g * cg) * Bn + b) * HW;
+    pstride = (int64_t)Bn * HW;
+  }
+  const T* xg = x + base;
+  const T* yg = y + base;
+  const T* dyg_in = dy + base;
+  T* dxg = dx + base;
+  T* drg = HAS_RES ? dres + base : nullptr;
+  const float mean = mean_in[group], rstd = rstd_in[group];
+  const T* gam = gamma + (int64_t)c * ch + g * cg;
+
+  extern __shared__ __attribute__((aligned(16))) unsigned char lds_raw[];
+  Pack<T, 8>* xbuf = reinterpret_cast<Pack<T, 8>*>(lds_raw);
+  Pack<T, 8>* gbuf = xbuf + n / 8;
+
+  __shared__ float red[2][OLS_THREADS / WAVE];
+  __shared__ float ch_dg[MAX_CG], ch_db[MAX_CG];
+  for (int i = threadIdx.x; i < cg; i += blockDim.x) {
+    ch_dg[i] = 0.f; ch_db[i] = 0.f;
+  }
+  __syncthreads();
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  float s1 = 0.f, s2 = 0.f;
+  const int nv = n / 8;
+  const int hv = HW / 8;
+  const int lpc = min(WAVE, hv);
+  for (int v8 = threadIdx.x; v8 - lane < nv; v8 += blockDim.x) {
+    const bool active = v8 < nv;
+    int i = active ? v8 * 8 : 0;
+    int chan = i / HW;
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+    float dg = 0.f, db = 0.f;
+    if (active) {
+      Pack<T, 8> pdy = *reinterpret_cast<const Pack<T, 8>*>(&dyg_in[idx]);
+      Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xg[idx]);
+      Pack<T, 8> py;
+      if (RELU) py = *reinterpret_cast<const Pack<T, 8>*>(&yg[idx]);
+      const float ga = to_f32(gam[chan]);
+      Pack<T, 8> pg;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float grad = to_f32(pdy.v[e]);
+        if (RELU) grad = to_f32(py.v[e]) > 0.f ? grad : 0.f;
+        pg.v[e] = RELU ? from_f32<T>(grad) : pdy.v[e];
+        float xhat = (to_f32(px.v[e]) - mean) * rstd;
+        float dxhat = grad * ga;
+        s1 += dxhat;
+        s2 += dxhat * xhat;
+        dg += grad * xhat;
+        db += grad;
+      }
+      xbuf[v8] = px;
+      gbuf[v8] = pg;
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      if (off < lpc) {
+        dg += __shfl_xor(dg, off, WAVE);
+        db += __shfl_xor(db, off, WAVE);
+      }
+    if (active && (lane & (lpc - 1)) == 0) {
+      atomicAdd(&ch_dg[chan], dg);
+      atomicAdd(&ch_db[chan], db);
+    }
+  }
+  s1 = wave_sum(s1);
+  s2 = wave_sum(s2);
+  const int wid = threadIdx.x / WAVE, nw = blockDim.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) { red[0][wid] = s1; red[1][wid] = s2; }
+  __syncthreads();
+  s1 = 0.f; s2 = 0.f;
+  for (int w = 0; w < nw; ++w) { s1 += red[0][w]; s2 += red[1][w]; }
+  const float m1 = s1 / n, m2 = s2 / n;
+
+  for (int v8 = threadIdx.x; v8 < nv; v8 += blockDim.x) {
+    int i = v8 * 8;
+    int chan = i / HW;
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+    Pack<T, 8> px = xbuf[v8];
+    Pack<T, 8> pg = gbuf[v8];
+    const float ga = to_f32(gam[chan]);
+    Pack<T, 8> pdx;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float grad = to_f32(pg.v[e]);
+      float xhat = (to_f32(px.v[e]) - mean) * rstd;
+      float dxhat = grad * ga;
+      pdx.v[e] = from_f32<T>(rstd * (dxhat - m1 - xhat * m2));
+    }
+    *reinterpret_cast<Pack<T, 8>*>(&dxg[idx]) = pdx;
+    if (HAS_RES) *reinterpret_cast<Pack<T, 8>*>(&drg[idx]) = pg;
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < cg; i += blockDim.x) {
+    atomicAdd(&dgamma[(int64_t)c * ch + g * cg + i], ch_dg[i]);
+    atomicAdd(&dbeta[(int64_t)c * ch + g * cg + i], ch_db[i]);
+  }
+}
+
 template <typename T>
 static void launch_fwd(const T* x, const T* res, T* y, float* mean,
                        float* rstd, const T* gamma, const T* beta, int B,
@@ -313,10 +440,22 @@ static void launch_bwd(const T* x, const T* y, const T* dy, T* dx, T* dres,
                        int G, int HW, bool relu, int layout, hipStream_t s) {
   dim3 grid(B * C * G), block(OLS_THREADS);
   const bool has_res = dres != nullptr;
+  // one-global-pass variant when the group plane fits in LDS (4 bytes
+  // of staging per element; cap 64 KB keeps >= 2 blocks resident)
+  const int n = (ch / G) * HW;
+  const size_t lds = (size_t)n * 2 * sizeof(T);
+  const bool use_lds = (HW % 8) == 0 && sizeof(T) == 2 && lds <= 65536;
 #define CASE(HR, RL, LY)                                                      \
-  hipLaunchKernelGGL((k_gn_bwd<T, HR, RL, LY>), grid, block, 0, s, x, y, dy, \
-                     dx, dres, mean, rstd, gamma, dgamma, dbeta, B, C, ch,   \
-                     G, HW)
+  do {                                                                        \
+    if (use_lds)                                                              \
+      hipLaunchKernelGGL((k_gn_bwd_lds<T, HR, RL, LY>), grid, block, lds, s, \
+                         x, y, dy, dx, dres, mean, rstd, gamma, dgamma,      \
+                         dbeta, B, C, ch, G, HW);                            \
+    else                                                                      \
+      hipLaunchKernelGGL((k_gn_bwd<T, HR, RL, LY>), grid, block, 0, s, x, y, \
+                         dy, dx, dres, mean, rstd, gamma, dgamma, dbeta, B,  \
+                         C, ch, G, HW);                                      \
+  } while (0)
 #define PICK(LY)                                                              \
   do {                                                                        \
     if (has_res && relu) CASE(true, true, LY);                                \
