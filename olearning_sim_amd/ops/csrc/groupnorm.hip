@@ -125,6 +125,89 @@ __global__ __launch_bounds__(OLS_THREADS) void k_gn_fwd(
   }
 }
 
+
+// One-global-pass forward: x is staged in LDS during the stats pass and
+// replayed for the normalize pass (k_gn_fwd reads it twice).
+template <typename T, bool HAS_RES, bool RELU, int LAYOUT>
+__global__ __launch_bounds__(OLS_THREADS) void k_gn_fwd_lds(
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    const T* __restrict__ gamma, const T* __restrict__ beta,
+    int Bn, int C, int ch, int G, int HW, float eps) {
+  const int cg = ch / G;
+  const int n = cg * HW;
+  const int group = blockIdx.x;
+  int c, g;
+  int64_t base, pstride;
+  if (LAYOUT == 0) {
+    g = group % G;
+    c = (group / G) % C;
+    base = (int64_t)group * n;
+    pstride = HW;
+  } else {
+    const int b = group % Bn;
+    g = (group / Bn) % G;
+    c = group / (Bn * G);
+    base = (((int64_t)c * ch + (int64_t)g * cg) * Bn + b) * HW;
+    pstride = (int64_t)Bn * HW;
+  }
+  const T* xg = x + base;
+  const T* rg = HAS_RES ? res + base : nullptr;
+  T* yg = y + base;
+
+  extern __shared__ __attribute__((aligned(16))) unsigned char lds_raw[];
+  Pack<T, 8>* xbuf = reinterpret_cast<Pack<T, 8>*>(lds_raw);
+
+  __shared__ float red[2][OLS_THREADS / WAVE];
+  float s1 = 0.f, s2 = 0.f;
+  const int nv = n / 8;
+  for (int v8 = threadIdx.x; v8 < nv; v8 += blockDim.x) {
+    int i = v8 * 8;
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)(i / HW) * pstride + i % HW);
+    Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xg[idx]);
+    xbuf[v8] = px;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = to_f32(px.v[e]);
+      s1 += v;
+      s2 += v * v;
+    }
+  }
+  s1 = wave_sum(s1);
+  s2 = wave_sum(s2);
+  const int wid = threadIdx.x / WAVE, nw = blockDim.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) { red[0][wid] = s1; red[1][wid] = s2; }
+  __syncthreads();
+  s1 = 0.f; s2 = 0.f;
+  for (int w = 0; w < nw; ++w) { s1 += red[0][w]; s2 += red[1][w]; }
+  const float mean = s1 / n;
+  const float var = fmaxf(s2 / n - mean * mean, 0.f);
+  const float rstd = rsqrtf(var + eps);
+  if (threadIdx.x == 0) { mean_out[group] = mean; rstd_out[group] = rstd; }
+
+  const T* gam = gamma + (int64_t)c * ch + g * cg;
+  const T* bet = beta + (int64_t)c * ch + g * cg;
+  for (int v8 = threadIdx.x; v8 < nv; v8 += blockDim.x) {
+    int i = v8 * 8;
+    int chan = i / HW;
+    int64_t idx = (LAYOUT == 0) ? i : ((int64_t)chan * pstride + i % HW);
+    Pack<T, 8> px = xbuf[v8];
+    Pack<T, 8> pr;
+    if (HAS_RES) pr = *reinterpret_cast<const Pack<T, 8>*>(&rg[idx]);
+    Pack<T, 8> py;
+    const float ga = to_f32(gam[chan]), be = to_f32(bet[chan]);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float v = (to_f32(px.v[e]) - mean) * rstd;
+      v = v * ga + be;
+      if (HAS_RES) v += to_f32(pr.v[e]);
+      if (RELU) v = fmaxf(v, 0.f);
+      py.v[e] = from_f32<T>(v);
+    }
+    *reinterpret_cast<Pack<T, 8>*>(&yg[idx]) = py;
+  }
+}
+
 #define MAX_CG 128
 
 template <typename T, bool HAS_RES, bool RELU, int LAYOUT>
@@ -418,9 +501,20 @@ static void launch_fwd(const T* x, const T* res, T* y, float* mean,
                        int layout, hipStream_t s) {
   dim3 grid(B * C * G), block(OLS_THREADS);
   const bool has_res = res != nullptr;
+  const int n = (ch / G) * HW;
+  const size_t lds = (size_t)n * sizeof(T);
+  const bool use_lds = (HW % 8) == 0 && sizeof(T) == 2 && lds <= 65536;
 #define CASE(HR, RL, LY)                                                      \
-  hipLaunchKernelGGL((k_gn_fwd<T, HR, RL, LY>), grid, block, 0, s, x, res,    \
-                     y, mean, rstd, gamma, beta, B, C, ch, G, HW, eps)
+  do {                                                                        \
+    if (use_lds)                                                              \
+      hipLaunchKernelGGL((k_gn_fwd_lds<T, HR, RL, LY>), grid, block, lds, s, \
+                         x, res, y, mean, rstd, gamma, beta, B, C, ch, G,    \
+                         HW, eps);                                           \
+    else                                                                      \
+      hipLaunchKernelGGL((k_gn_fwd<T, HR, RL, LY>), grid, block, 0, s, x,    \
+                         res, y, mean, rstd, gamma, beta, B, C, ch, G, HW,   \
+                         eps);                                               \
+  } while (0)
 #define PICK(LY)                                                              \
   do {                                                                        \
     if (has_res && relu) CASE(true, true, LY);                                \
