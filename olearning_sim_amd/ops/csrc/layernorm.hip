@@ -90,6 +90,12 @@ __global__ __launch_bounds__(OLS_THREADS) void k_ln_bwd(
   // rows_per_client % LN_ROWS == 0); derive it from the block's first
   // row so tail-block threads never write another client's columns
   const int64_t c = ((int64_t)blockIdx.x * LN_ROWS) / rows_per_client;
+  // per-wave x/dy row staging after the column accumulators: the
+  // stats pass reads each row from HBM once, the dx pass replays LDS
+  Pack<T, 8>* xrow = reinterpret_cast<Pack<T, 8>*>(col_acc + 2 * H)
+                     + (int64_t)wave * hv;
+  Pack<T, 8>* drow = reinterpret_cast<Pack<T, 8>*>(col_acc + 2 * H)
+                     + ((int64_t)LN_ROWS + wave) * hv;
   if (row < rows) {
     const T* xr = x + row * H;
     const T* dyr = dy + row * H;
@@ -102,6 +108,8 @@ __global__ __launch_bounds__(OLS_THREADS) void k_ln_bwd(
       Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xr[v * 8]);
       Pack<T, 8> pd = *reinterpret_cast<const Pack<T, 8>*>(&dyr[v * 8]);
       Pack<T, 8> pg = *reinterpret_cast<const Pack<T, 8>*>(&g[v * 8]);
+      xrow[v] = px;
+      drow[v] = pd;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         float grad = to_f32(pd.v[e]);
@@ -117,8 +125,8 @@ __global__ __launch_bounds__(OLS_THREADS) void k_ln_bwd(
     s2 = wave_sum(s2);
     const float m1 = s1 / H, m2 = s2 / H;
     for (int v = lane; v < hv; v += WAVE) {
-      Pack<T, 8> px = *reinterpret_cast<const Pack<T, 8>*>(&xr[v * 8]);
-      Pack<T, 8> pd = *reinterpret_cast<const Pack<T, 8>*>(&dyr[v * 8]);
+      Pack<T, 8> px = xrow[v];
+      Pack<T, 8> pd = drow[v];
       Pack<T, 8> pg = *reinterpret_cast<const Pack<T, 8>*>(&g[v * 8]);
       Pack<T, 8> po;
 #pragma unroll
@@ -168,7 +176,11 @@ extern "C" void ols_layernorm_bwd(const void* x, const void* dy,
                                   hipStream_t stream) {
   dim3 grid((unsigned)((rows + LN_ROWS - 1) / LN_ROWS));
   dim3 block(LN_ROWS * WAVE);
-  size_t lds = 2 * (size_t)H * sizeof(float);
+  // column accumulators + 2*LN_ROWS row-staging buffers (element-sized;
+  // the Python wrapper caps H so this stays under the 64 KB dynamic cap)
+  size_t esz = (dtype == 1) ? sizeof(__hip_bfloat16) : sizeof(float);
+  size_t lds = 2 * (size_t)H * sizeof(float)
+               + 2 * (size_t)LN_ROWS * H * esz;
   if (dtype == 1)
     hipLaunchKernelGGL((k_ln_bwd<__hip_bfloat16>), grid, block, lds, stream,
                        (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,

@@ -329,7 +329,7 @@ def layernorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
     shape = x.shape
     C, H = shape[0], shape[-1]
     n = x.numel() // (C * H)
-    if not (x.is_cuda and H % 8 == 0 and n % 4 == 0
+    if not (x.is_cuda and H % 8 == 0 and n % 4 == 0 and H <= 1536
             and x.dtype in (torch.bfloat16, torch.float32)
             and hip_ops_available()):
         return None
