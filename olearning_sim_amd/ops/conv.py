@@ -165,7 +165,10 @@ class _Conv5x5Fn(torch.autograd.Function):
         C, IC, B, H, W = x.shape
         dy = dy.contiguous()
         if ctx.relu:
-            dy = dy * (y > 0).to(dy.dtype)
+            if dy.numel() % 8 == 0:
+                dy = ops.relu_mask(dy, y)     # one pass (replicate.hip)
+            else:
+                dy = dy * (y > 0).to(dy.dtype)
         dx = dw = None
         if ctx.needs_input_grad[0]:
             dy_pad = _pad(dy, 4)

@@ -89,6 +89,9 @@ extern "C" void ols_synth_batch(const float* x, const int64_t* y, void* out,
                                 int64_t rows, int64_t batch, int64_t n,
                                 float s, float t, hipStream_t stream);
 
+extern "C" void ols_relu_mask(const void* dy, const void* y, void* out,
+                              int64_t n, int dtype, hipStream_t stream);
+
 extern "C" void ols_layernorm_fwd(const void* x, const void* gamma,
                                   const void* beta, void* y, float* mean,
                                   float* rstd, int64_t rows, int H,
@@ -538,6 +541,21 @@ at::Tensor synth_batch(at::Tensor x, at::Tensor y, double s, double t) {
   return out;
 }
 
+// ---- fused relu-mask backward (replicate.hip) ---------------------------
+
+at::Tensor relu_mask(at::Tensor dy, at::Tensor y) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && y.is_contiguous());
+  TORCH_CHECK(dy.sizes() == y.sizes() && dy.scalar_type() == y.scalar_type());
+  TORCH_CHECK(dy.numel() % 8 == 0);
+  TORCH_CHECK(dy.scalar_type() == at::kBFloat16 ||
+              dy.scalar_type() == at::kFloat);
+  auto out = at::empty_like(dy);
+  int dt = dy.scalar_type() == at::kBFloat16 ? 1 : 0;
+  ols_relu_mask(dy.data_ptr(), y.data_ptr(), out.data_ptr(), dy.numel(), dt,
+                at::cuda::getCurrentCUDAStream().stream());
+  return out;
+}
+
 // ---- zero-pad trailing 2 dims (pad2d.hip) -------------------------------
 
 at::Tensor pad2d(at::Tensor x, int64_t pad) {
@@ -601,6 +619,7 @@ TORCH_LIBRARY(olsim_hip, m) {
   m.def("pool2x2_bwd(Tensor dy, Tensor arg) -> Tensor");
   m.def("transpose2d(Tensor x) -> Tensor");
   m.def("pad2d(Tensor x, int pad) -> Tensor");
+  m.def("relu_mask(Tensor dy, Tensor y) -> Tensor");
   m.def("replicate(Tensor src, int clients) -> Tensor");
   m.def("subsample2(Tensor x) -> Tensor");
   m.def("subsample2_bwd(Tensor dy, int H, int W) -> Tensor");
@@ -629,6 +648,7 @@ TORCH_LIBRARY_IMPL(olsim_hip, CUDA, m) {
   m.impl("pool2x2_bwd", &pool2x2_bwd);
   m.impl("transpose2d", &transpose2d);
   m.impl("pad2d", &pad2d);
+  m.impl("relu_mask", &relu_mask);
   m.impl("replicate", &replicate);
   m.impl("subsample2", &subsample2);
   m.impl("subsample2_bwd", &subsample2_bwd);

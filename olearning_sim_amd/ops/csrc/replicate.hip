@@ -77,3 +77,37 @@ extern "C" void ols_synth_batch(const float* x, const int64_t* y, void* out,
   hipLaunchKernelGGL(k_synth_batch, grid, block, 0, stream, x, y,
                      (__hip_bfloat16*)out, rows, batch, nv, s, t);
 }
+
+
+// dy * (y > 0) in one pass (the composed torch chain is a bool compare
+// tensor + a cast + a multiply — three extra full-size passes on the
+// conv5 ReLU backward path).
+template <typename T>
+__global__ __launch_bounds__(OLS_THREADS) void k_relu_mask(
+    const T* __restrict__ dy, const T* __restrict__ y, T* __restrict__ out,
+    int64_t nv) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t u = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       u < nv; u += stride) {
+    Pack<T, 8> pd = *reinterpret_cast<const Pack<T, 8>*>(&dy[u * 8]);
+    Pack<T, 8> py = *reinterpret_cast<const Pack<T, 8>*>(&y[u * 8]);
+    Pack<T, 8> po;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      po.v[e] = to_f32(py.v[e]) > 0.f ? pd.v[e] : from_f32<T>(0.f);
+    *reinterpret_cast<Pack<T, 8>*>(&out[u * 8]) = po;
+  }
+}
+
+extern "C" void ols_relu_mask(const void* dy, const void* y, void* out,
+                              int64_t n, int dtype, hipStream_t stream) {
+  const int64_t nv = n / 8;
+  dim3 grid(ols_grid(nv, OLS_THREADS)), block(OLS_THREADS);
+  if (dtype == 1)
+    hipLaunchKernelGGL((k_relu_mask<__hip_bfloat16>), grid, block, 0, stream,
+                       (const __hip_bfloat16*)dy, (const __hip_bfloat16*)y,
+                       (__hip_bfloat16*)out, nv);
+  else
+    hipLaunchKernelGGL((k_relu_mask<float>), grid, block, 0, stream,
+                       (const float*)dy, (const float*)y, (float*)out, nv);
+}
