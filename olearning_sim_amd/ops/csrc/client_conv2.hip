@@ -499,21 +499,8 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v6(
         int ow0 = min(q & ((1 << g.lg_ow) - 1), g.OW - CV6_BN / 8);
         const ushort* row = plane + (int64_t)b * HpWp + (oh + dh) * g.Wp
                             + ow0 + dw;
-        if ((dw & 1) == 0) {
-          // even dw: the row pointer is 4-byte aligned (Wp, HpWp,
-          // planeB, ow0 all even) — halve the load count (PMC: the
-          // producer is issue-bound, profiles/pmc_final_r02.md)
-          const uint32_t* row32 = reinterpret_cast<const uint32_t*>(row);
 #pragma unroll
-          for (int j2 = 0; j2 < CV6_BN / 16; ++j2) {
-            uint32_t v = row32[j2];
-            breg[2 * j2] = (ushort)v;
-            breg[2 * j2 + 1] = (ushort)(v >> 16);
-          }
-        } else {
-#pragma unroll
-          for (int j = 0; j < CV6_BN / 8; ++j) breg[j] = row[j];
-        }
+        for (int j = 0; j < CV6_BN / 8; ++j) breg[j] = row[j];
       } else {
         constexpr int SUBW = 1 << LG_OW_T;
         constexpr int NROW = (CV6_BN / 8) / SUBW;
@@ -524,21 +511,8 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_dgrad_v6(
           int q = n & ((1 << g.lg_ohw) - 1);
           int oh = q >> LG_OW_T;
           const ushort* row = plane + (int64_t)b * HpWp + (oh + dh) * g.Wp + dw;
-          // vector path needs the row start 4B-aligned: even dw AND an
-          // unclamped n (a clamped tail lands mid-row at odd ow)
-          if (SUBW >= 2 && (dw & 1) == 0
-              && n == n0 + nn0 + rr * SUBW) {
-            const uint32_t* row32 = reinterpret_cast<const uint32_t*>(row);
 #pragma unroll
-            for (int j2 = 0; j2 < SUBW / 2; ++j2) {
-              uint32_t v = row32[j2];
-              breg[rr * SUBW + 2 * j2] = (ushort)v;
-              breg[rr * SUBW + 2 * j2 + 1] = (ushort)(v >> 16);
-            }
-          } else {
-#pragma unroll
-            for (int j = 0; j < SUBW; ++j) breg[rr * SUBW + j] = row[j];
-          }
+          for (int j = 0; j < SUBW; ++j) breg[rr * SUBW + j] = row[j];
         }
       }
     }
