@@ -1085,6 +1085,138 @@ __global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v8(
   }
 }
 
+
+// wgrad v8s: run-vectorised producer for STRIDE 2 (OW >= 8).  Same
+// item structure as v8, but the 8 q positions map to every-2nd input
+// pixel, so the 3 dw columns need 17 consecutive u16 (5 8-byte loads)
+// and the extraction picks alternate elements with constant-selector
+// v_perm after one branchless word-shift.
+__global__ __launch_bounds__(CONV_THREADS) void k_conv3x3_wgrad_v8s(
+    const __hip_bfloat16* __restrict__ xp, const __hip_bfloat16* __restrict__ dy,
+    __hip_bfloat16* __restrict__ dw, ConvGeom6 g) {
+  int c, tile;
+  if (!xcd_remap6(blockIdx.x, g.C, g.tiles_m * g.tiles_n, c, tile)) return;
+  const int mt = tile / g.tiles_n;
+  const int m0 = mt * CV6_BM;
+  const int n0 = (tile - mt * g.tiles_n) * CV6_BN;
+  __shared__ short bT_lds[2][CV6_BN * (CV6_PAD + CV6_BK)];
+  const int K9 = g.IC * 9;
+  const int NN = g.B * g.OH * g.OW;
+  const int HpWp = g.Hp * g.Wp;
+  const int64_t planeB = (int64_t)g.B * HpWp;
+  const ushort* xc =
+      reinterpret_cast<const ushort*>(xp) + (int64_t)c * g.IC * planeB;
+  const __hip_bfloat16* dyc = dy + (int64_t)c * g.OC * NN;
+  __hip_bfloat16* dwc = dw + (int64_t)c * g.OC * K9;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int arow = m0 + wave * 16 + (lane & 15);
+  const __hip_bfloat16* dyrow = dyc + (int64_t)min(arow, g.OC - 1) * NN;
+  const bool arow_ok = arow < g.OC;
+
+  f32x4 acc[CV6_BN / 16];
+#pragma unroll
+  for (int i = 0; i < CV6_BN / 16; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int runs_lo = n0 / 3;
+  const int runs_hi = (min(n0 + CV6_BN, K9) + 2) / 3;
+  const int nitems = (runs_hi - runs_lo) * (CV6_BK / 8);
+  const bool prod = threadIdx.x < nitems;
+  const int run = runs_lo + threadIdx.x / (CV6_BK / 8);
+  const int qv = (threadIdx.x % (CV6_BK / 8)) * 8;
+  const int ric = run / 3, rdh = run - ric * 3;
+  const int64_t runoff = (int64_t)ric * planeB + rdh * g.Wp;
+
+  uint2 r0, r1, r2, r3, r4;
+  int tpar = 0;
+
+  auto gather = [&](int q0) {
+    if (!prod) return;
+    const int q = q0 + qv;
+    const int b = q >> g.lg_ohw;
+    const int p = q & ((1 << g.lg_ohw) - 1);
+    const int oh = p >> g.lg_ow;
+    const int ow = p & ((1 << g.lg_ow) - 1);
+    const int64_t E = runoff + (int64_t)b * HpWp + (oh * 2) * g.Wp + ow * 2;
+    const int64_t E4 = min(E & ~3LL, (int64_t)g.IC * planeB - 20);
+    tpar = (int)(E - E4) >> 1;               // 0 or 1 (E is even)
+    const uint2* src = reinterpret_cast<const uint2*>(xc + E4);
+    r0 = src[0];
+    r1 = src[1];
+    r2 = src[2];
+    r3 = src[3];
+    r4 = src[4];
+  };
+  auto commit = [&](int buf) {
+    if (!prod) return;
+    uint32_t w[10] = {r0.x, r0.y, r1.x, r1.y, r2.x,
+                      r2.y, r3.x, r3.y, r4.x, r4.y};
+    // s[k] = u32 of u16 elements (E + 2k, E + 2k + 1), k 0..8
+    uint32_t s[9];
+#pragma unroll
+    for (int k = 0; k < 9; ++k) s[k] = tpar ? w[k + 1] : w[k];
+    // col dw needs elements E + dw + 2j (j 0..7): alternate u16s
+#pragma unroll
+    for (int dw2 = 0; dw2 < 3; ++dw2) {
+      const int k9 = run * 3 + dw2;
+      const int col = k9 - n0;
+      if (col < 0 || col >= CV6_BN || k9 >= K9) continue;
+      // o[m] = (elem dw+4m, elem dw+4m+2); dw even -> lo16 halves of
+      // s[dw/2+2m], s[dw/2+2m+1]; dw odd -> hi16 halves of s[2m], s[2m+1]
+      const int b0 = (dw2 + 1) >> 1;   // 0, 0, 1 -> s-base per pair step
+      const uint32_t sel = (dw2 & 1) ? 0x07060302u : 0x05040100u;
+      const int base = (dw2 == 2) ? 1 : 0;
+      uint4 o;
+      o.x = __builtin_amdgcn_perm(s[base + 1], s[base + 0], sel);
+      o.y = __builtin_amdgcn_perm(s[base + 3], s[base + 2], sel);
+      o.z = __builtin_amdgcn_perm(s[base + 5], s[base + 4], sel);
+      o.w = __builtin_amdgcn_perm(s[base + 7], s[base + 6], sel);
+      (void)b0;
+      *reinterpret_cast<uint4*>(
+          &bT_lds[buf][col * (CV6_BK + CV6_PAD) + qv]) = o;
+    }
+  };
+
+  gather(0);
+  commit(0);
+  int cur = 0;
+  for (int q0 = 0; q0 < NN; q0 += CV6_BK) {
+    __syncthreads();
+    if (q0 + CV6_BK < NN) gather(q0 + CV6_BK);
+    bf16x8 a;
+    {
+      uint4 av = *reinterpret_cast<const uint4*>(dyrow + q0 + 8 * (lane >> 4));
+      a = *reinterpret_cast<const bf16x8*>(&av);
+      if (!arow_ok) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) a[e] = 0;
+      }
+    }
+#pragma unroll
+    for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &bT_lds[cur][(nt * 16 + (lane & 15)) * (CV6_BK + CV6_PAD)
+                       + 8 * (lane >> 4)]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nt], 0, 0, 0);
+    }
+    if (q0 + CV6_BK < NN) commit(cur ^ 1);
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int nt = 0; nt < CV6_BN / 16; ++nt) {
+    int k = n0 + nt * 16 + (lane & 15);
+    if (k >= K9) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int m = m0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (m < g.OC)
+        dwc[(int64_t)m * K9 + k] = __float2bfloat16(acc[nt][r]);
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // wgrad v7: BK=64 reduction steps (two MFMA sub-steps per barrier),
 // int32 gather offsets.  Requires NN % 64 == 0.
@@ -1358,6 +1490,12 @@ extern "C" void ols_conv3x3_wgrad_p(const void* xp, const void* dy, void* dw,
   g.tiles_n = cdiv6(IC * 9, CV6_BN);
   dim3 grid(xcd_blocks6(C, g.tiles_m * g.tiles_n));
   const int NN = B * g.OH * g.OW;
+  if (stride == 2 && g.OW >= 8 && getenv("OLSIM_CONV_DW8") == nullptr) {
+    hipLaunchKernelGGL(k_conv3x3_wgrad_v8s, grid, dim3(CONV_THREADS), 0,
+                       stream, (const __hip_bfloat16*)xp,
+                       (const __hip_bfloat16*)dy, (__hip_bfloat16*)dw, g);
+    return;
+  }
   if (stride == 1 && g.OW >= 8 && getenv("OLSIM_CONV_DW8") == nullptr) {
     // run-vectorised producer (v8): 8 consecutive stride-1 q stay in
     // one padded row; OLSIM_CONV_DW8=0 restores the scalar gather

@@ -272,26 +272,29 @@ def test_pool2x2_matches_torch():
 
 
 @pytest.mark.parametrize("shape", [
-    (3, 64, 64, 4, 16), (2, 64, 32, 2, 8), (2, 128, 64, 2, 32),
-    (1, 512, 512, 2, 8),
+    (3, 64, 64, 4, 16, 1), (2, 64, 32, 2, 8, 1), (2, 128, 64, 2, 32, 1),
+    (1, 512, 512, 2, 8, 1),
+    # stride 2 (v8s): OW 16 and 8
+    (2, 64, 128, 4, 32, 2), (2, 128, 256, 2, 16, 2),
 ])
 def test_wgrad_v8_bitwise_matches_scalar_gather(shape):
-    """Run-vectorised wgrad producer (v8) == scalar-gather producer,
+    """Run-vectorised wgrad producer (v8/v8s) == scalar-gather producer,
     bitwise (same MFMA tiles, same reduction order)."""
     import os
     import torch.nn.functional as F
     from olearning_sim_amd.ops.fused import load_hip_ops
     ops = load_hip_ops(required=True)
-    C, IC, OC, B, H = shape
+    C, IC, OC, B, H, stride = shape
     torch.manual_seed(9)
+    OH = H // stride
     x = torch.randn(C, IC, B, H, H, device="cuda").to(torch.bfloat16)
-    dy = torch.randn(C, OC, B, H, H, device="cuda").to(torch.bfloat16)
+    dy = torch.randn(C, OC, B, OH, OH, device="cuda").to(torch.bfloat16)
     xp = F.pad(x, (1, 1, 1, 1)).contiguous()
     assert "OLSIM_CONV_DW8" not in os.environ
-    dw_v8 = ops.conv3x3_wgrad_p(xp, dy, 1)
+    dw_v8 = ops.conv3x3_wgrad_p(xp, dy, stride)
     os.environ["OLSIM_CONV_DW8"] = "0"
     try:
-        dw_v6 = ops.conv3x3_wgrad_p(xp, dy, 1)
+        dw_v6 = ops.conv3x3_wgrad_p(xp, dy, stride)
     finally:
         del os.environ["OLSIM_CONV_DW8"]
     torch.testing.assert_close(dw_v8, dw_v6, atol=0, rtol=0)
