@@ -290,6 +290,19 @@ def test_synth_batch_matches_composed():
     torch.testing.assert_close(out.view(C, B, n), ref)
 
 
+def test_relu_mask_matches_composed():
+    """One-pass dy*(y>0) == the composed compare+cast+multiply chain."""
+    from olearning_sim_amd.ops.fused import load_hip_ops
+    ops = load_hip_ops(required=True)
+    torch.manual_seed(12)
+    for dt in (torch.bfloat16, torch.float32):
+        y = torch.randn(40, 16, 24, device="cuda").to(dt)
+        dy = torch.randn_like(y)
+        out = ops.relu_mask(dy, y)
+        ref = dy * (y > 0).to(dt)
+        torch.testing.assert_close(out, ref, atol=0, rtol=0)
+
+
 @pytest.mark.parametrize("shape,pad", [
     ((250, 16, 10, 32, 32), 1), ((250, 64, 10, 8, 8), 1),
     ((100, 6, 20, 24, 24), 4), ((7, 3, 5, 14, 18), 1),
