@@ -33,25 +33,16 @@ __global__ __launch_bounds__(OLS_THREADS) void k_pad2d(
     const int rr = (int)r - pad;
     T v0 = from_f32<T>(0.f), v1 = v0;
     if (rr >= 0 && rr < H) {
-      // nontemporal: the pads stream GBs through the 4 MB per-XCD L2
-      // with zero reuse
-      const ushort* row = reinterpret_cast<const ushort*>(
-          ing + ((int64_t)pl * H + rr) * W - pad);
+      const T* row = ing + ((int64_t)pl * H + rr) * W - pad;
       const int ca = c0, cb = c0 + 1;        // padded cols
-      ushort u0 = *reinterpret_cast<ushort*>(&v0);
-      ushort u1 = u0;
-      if (ca >= pad && ca < W + pad) u0 = __builtin_nontemporal_load(&row[ca]);
-      if (cb >= pad && cb < W + pad) u1 = __builtin_nontemporal_load(&row[cb]);
-      v0 = *reinterpret_cast<T*>(&u0);
-      v1 = *reinterpret_cast<T*>(&u1);
+      if (ca >= pad && ca < W + pad) v0 = row[ca];
+      if (cb >= pad && cb < W + pad) v1 = row[cb];
     }
     ushort2 pk;
     pk.x = *reinterpret_cast<ushort*>(&v0);
     pk.y = *reinterpret_cast<ushort*>(&v1);
-    __builtin_nontemporal_store(
-        *reinterpret_cast<uint32_t*>(&pk),
-        reinterpret_cast<uint32_t*>(
-            &outg[((int64_t)pl * Hp + r) * Wp + c0]));
+    *reinterpret_cast<ushort2*>(
+        &outg[((int64_t)pl * Hp + r) * Wp + c0]) = pk;
   }
 }
 
