@@ -334,6 +334,65 @@ __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad(
 static inline int cdiv5(int a, int b) { return (a + b - 1) / b; }
 static inline int xcd_blocks5(int C, int T) { return ((C + 7) / 8 * 8) * T; }
 
+
+// ---------------------------------------------------------------------------
+// Direct VALU forward.  PMC on the implicit-GEMM fwd: MFMA busy 1.4%,
+// 59% SQ_WAIT_ANY + 26% SQ_WAIT_INST — with M = OC <= 16 and
+// K = IC*25 <= 400 the MFMA tile shape wastes the matrix cores and the
+// kernel is a pure gather machine.  Here one block owns one (client, b)
+// image: x plane and the client weights stage in LDS once, and each
+// thread accumulates whole output pixels with an unrolled 25-tap loop
+// — ~300 TFLOP/s of f32 FMA against ~0.15 TFLOP of work.
+template <bool RELU>
+__global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_fwd_direct(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ w,
+    const __hip_bfloat16* __restrict__ bias, __hip_bfloat16* __restrict__ y,
+    ConvGeom5 g) {
+  const int c = blockIdx.x / g.B;
+  const int b = blockIdx.x - c * g.B;
+  const int HW = g.H * g.W, OHW = g.OH * g.OW;
+  const int K = g.IC * 25;
+  extern __shared__ __attribute__((aligned(16))) short smem[];
+  short* x_lds = smem;                         // [IC*HW]
+  short* w_lds = smem + g.IC * HW;             // [OC*K]
+  const ushort* xc = reinterpret_cast<const ushort*>(x)
+                     + ((int64_t)c * g.IC * g.B + b) * HW;
+  const ushort* wc = reinterpret_cast<const ushort*>(w) + (int64_t)c * g.OC * K;
+  // stage x planes (strided by B*HW per ic) and the client weights
+  for (int ic = 0; ic < g.IC; ++ic) {
+    const ushort* plane = xc + (int64_t)ic * g.B * HW;
+    for (int i = threadIdx.x; i < HW; i += CV5_THREADS)
+      x_lds[ic * HW + i] = (short)plane[i];
+  }
+  for (int i = threadIdx.x; i < g.OC * K; i += CV5_THREADS)
+    w_lds[i] = (short)wc[i];
+  __syncthreads();
+
+  __hip_bfloat16* yc = y + ((int64_t)c * g.OC * g.B + b) * OHW;
+  for (int o = threadIdx.x; o < g.OC * OHW; o += CV5_THREADS) {
+    const int oc = o / OHW;
+    const int p = o - oc * OHW;
+    const int oh = p / g.OW;
+    const int ow = p - oh * g.OW;
+    float acc = to_f32(bias[(int64_t)c * g.OC + oc]);
+    const short* wr = w_lds + oc * K;
+    const short* xr = x_lds + oh * g.W + ow;
+    for (int ic = 0; ic < g.IC; ++ic) {
+#pragma unroll
+      for (int dh = 0; dh < 5; ++dh)
+#pragma unroll
+        for (int dw = 0; dw < 5; ++dw) {
+          ushort xv = (ushort)xr[ic * HW + dh * g.W + dw];
+          ushort wv = (ushort)wr[ic * 25 + dh * 5 + dw];
+          acc += to_f32(*reinterpret_cast<__hip_bfloat16*>(&xv))
+                 * to_f32(*reinterpret_cast<__hip_bfloat16*>(&wv));
+        }
+    }
+    if (RELU) acc = fmaxf(acc, 0.f);
+    yc[(int64_t)oc * g.B * OHW + p] = __float2bfloat16(acc);
+  }
+}
+
 extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
                                 void* y, const int* ntab, int C, int IC,
                                 int OC, int B, int H, int W, int relu,
@@ -343,6 +402,25 @@ extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
   g.IC = IC; g.OC = OC; g.C = C;
   g.K = IC * 25; g.KP = cdiv5(g.K, CV5_BK) * CV5_BK;
   g.tiles_n = cdiv5(B * g.OH * g.OW, CV5_BN);
+  // direct VALU kernel when the image + weights fit in LDS (every
+  // LeNet-family shape); OLSIM_CONV5=mfma restores the MFMA path
+  const size_t direct_lds = ((size_t)IC * H * W + (size_t)OC * g.K)
+                            * sizeof(short);
+  const char* c5 = getenv("OLSIM_CONV5");
+  if (direct_lds <= 32768 && (c5 == nullptr || c5[0] != 'm')) {
+    dim3 gridd((unsigned)((int64_t)C * B));
+    if (relu)
+      hipLaunchKernelGGL((k_conv5x5_fwd_direct<true>), gridd,
+                         dim3(CV5_THREADS), direct_lds, stream,
+                         (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
+                         (const __hip_bfloat16*)b, (__hip_bfloat16*)y, g);
+    else
+      hipLaunchKernelGGL((k_conv5x5_fwd_direct<false>), gridd,
+                         dim3(CV5_THREADS), direct_lds, stream,
+                         (const __hip_bfloat16*)x, (const __hip_bfloat16*)w,
+                         (const __hip_bfloat16*)b, (__hip_bfloat16*)y, g);
+    return;
+  }
   size_t lds = (16 * g.KP + 2 * CV5_BN * (CV5_BK + CV5_PAD)) * sizeof(short);
   dim3 grid(xcd_blocks5(C, g.tiles_n));
   if (relu)
