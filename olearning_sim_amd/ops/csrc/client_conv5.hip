@@ -358,7 +358,6 @@ __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_fwd_direct(
   const ushort* xc = reinterpret_cast<const ushort*>(x)
                      + ((int64_t)c * g.IC * g.B + b) * HW;
   const ushort* wc = reinterpret_cast<const ushort*>(w) + (int64_t)c * g.OC * K;
-  // stage x planes (strided by B*HW per ic) and the client weights
   for (int ic = 0; ic < g.IC; ++ic) {
     const ushort* plane = xc + (int64_t)ic * g.B * HW;
     for (int i = threadIdx.x; i < HW; i += CV5_THREADS)
@@ -368,28 +367,58 @@ __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_fwd_direct(
     w_lds[i] = (short)wc[i];
   __syncthreads();
 
+  // register blocking: one item = (oc, oh, 4 consecutive ow).  The
+  // tail block re-computes overlapping columns (ow0 = OW-4) so every
+  // x row read [ow0 .. ow0+7] stays inside the W = OW+4 input row and
+  // starts at an even column (u32 LDS reads).  4 independent
+  // accumulator chains per thread hide the ds_read latency that made
+  // the one-output-per-thread version 3-4x slower than MFMA.
+  const int wblk = (g.OW + 3) / 4;
+  const int nitems = g.OC * g.OH * wblk;
   __hip_bfloat16* yc = y + ((int64_t)c * g.OC * g.B + b) * OHW;
-  for (int o = threadIdx.x; o < g.OC * OHW; o += CV5_THREADS) {
-    const int oc = o / OHW;
-    const int p = o - oc * OHW;
-    const int oh = p / g.OW;
-    const int ow = p - oh * g.OW;
-    float acc = to_f32(bias[(int64_t)c * g.OC + oc]);
+  for (int it = threadIdx.x; it < nitems; it += CV5_THREADS) {
+    const int oc = it / (g.OH * wblk);
+    const int p = it - oc * (g.OH * wblk);
+    const int oh = p / wblk;
+    const int ow0 = min((p - oh * wblk) * 4, g.OW - 4);
+    const float bv = to_f32(bias[(int64_t)c * g.OC + oc]);
+    float acc0 = bv, acc1 = bv, acc2 = bv, acc3 = bv;
     const short* wr = w_lds + oc * K;
-    const short* xr = x_lds + oh * g.W + ow;
     for (int ic = 0; ic < g.IC; ++ic) {
+      const short* xrow = x_lds + ic * HW + oh * g.W + ow0;
+      const short* wrow = wr + ic * 25;
 #pragma unroll
-      for (int dh = 0; dh < 5; ++dh)
+      for (int dh = 0; dh < 5; ++dh) {
+        const uint32_t* xr32 = reinterpret_cast<const uint32_t*>(
+            xrow + dh * g.W);
+        float f[8];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const uint32_t v = xr32[u];
+          ushort lo = (ushort)v, hi = (ushort)(v >> 16);
+          f[2 * u] = to_f32(*reinterpret_cast<__hip_bfloat16*>(&lo));
+          f[2 * u + 1] = to_f32(*reinterpret_cast<__hip_bfloat16*>(&hi));
+        }
 #pragma unroll
         for (int dw = 0; dw < 5; ++dw) {
-          ushort xv = (ushort)xr[ic * HW + dh * g.W + dw];
-          ushort wv = (ushort)wr[ic * 25 + dh * 5 + dw];
-          acc += to_f32(*reinterpret_cast<__hip_bfloat16*>(&xv))
-                 * to_f32(*reinterpret_cast<__hip_bfloat16*>(&wv));
+          ushort wv = (ushort)wrow[dh * 5 + dw];
+          const float wf = to_f32(*reinterpret_cast<__hip_bfloat16*>(&wv));
+          acc0 += f[dw] * wf;
+          acc1 += f[dw + 1] * wf;
+          acc2 += f[dw + 2] * wf;
+          acc3 += f[dw + 3] * wf;
         }
+      }
     }
-    if (RELU) acc = fmaxf(acc, 0.f);
-    yc[(int64_t)oc * g.B * OHW + p] = __float2bfloat16(acc);
+    if (RELU) {
+      acc0 = fmaxf(acc0, 0.f); acc1 = fmaxf(acc1, 0.f);
+      acc2 = fmaxf(acc2, 0.f); acc3 = fmaxf(acc3, 0.f);
+    }
+    __hip_bfloat16* yr = yc + (int64_t)oc * g.B * OHW + oh * g.OW + ow0;
+    yr[0] = __float2bfloat16(acc0);
+    yr[1] = __float2bfloat16(acc1);
+    yr[2] = __float2bfloat16(acc2);
+    yr[3] = __float2bfloat16(acc3);
   }
 }
 
@@ -411,7 +440,8 @@ extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
   const size_t direct_lds = ((size_t)IC * H * W + (size_t)OC * g.K)
                             * sizeof(short);
   const char* c5 = getenv("OLSIM_CONV5");
-  if (direct_lds <= 32768 && c5 != nullptr && c5[0] == 'd') {
+  if (direct_lds <= 32768 && g.OW >= 4 && W % 2 == 0
+      && c5 != nullptr && c5[0] == 'd') {
     dim3 gridd((unsigned)((int64_t)C * B));
     if (relu)
       hipLaunchKernelGGL((k_conv5x5_fwd_direct<true>), gridd,
