@@ -431,17 +431,18 @@ extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
   g.IC = IC; g.OC = OC; g.C = C;
   g.K = IC * 25; g.KP = cdiv5(g.K, CV5_BK) * CV5_BK;
   g.tiles_n = cdiv5(B * g.OH * g.OW, CV5_BN);
-  // direct VALU kernel (k_conv5x5_fwd_direct): MEASURED 3-4x SLOWER
-  // than the MFMA path (fedprox 109 vs 59 ms/round) — the scalar
-  // 25-tap LDS loop is ds_read-latency-bound (one dependent chain per
-  // output).  Kept behind OLSIM_CONV5=direct; the round-3 fix is
-  // register-blocking several outputs per thread (shared w taps,
-  // 4-way independent accumulator chains) before it can compete.
+  // direct register-blocked VALU kernel: the MFMA implicit GEMM has
+  // M = OC <= 16 and K <= 400 (PMC: MFMA busy 1.4%) and is a pure
+  // gather machine; the 4-output-per-thread direct kernel measured
+  // fedprox 52.0 vs 60.8 ms/round and lenet 8.4 vs 9.2.  (The naive
+  // one-output-per-thread version was 3-4x SLOWER — single dependent
+  // ds_read chain; the register blocking is what wins.)
+  // OLSIM_CONV5=mfma restores the MFMA path.
   const size_t direct_lds = ((size_t)IC * H * W + (size_t)OC * g.K)
                             * sizeof(short);
   const char* c5 = getenv("OLSIM_CONV5");
   if (direct_lds <= 32768 && g.OW >= 4 && W % 2 == 0
-      && c5 != nullptr && c5[0] == 'd') {
+      && (c5 == nullptr || c5[0] != 'm')) {
     dim3 gridd((unsigned)((int64_t)C * B));
     if (relu)
       hipLaunchKernelGGL((k_conv5x5_fwd_direct<true>), gridd,
