@@ -471,6 +471,80 @@ extern "C" void ols_conv5x5_fwd(const void* x, const void* w, const void* b,
 }
 
 // H, W: the dX plane; dyp is [C][OC][B][OH+8][OW+8] with OH=H-4
+
+// Direct register-blocked dgrad: same structure as the direct fwd
+// (the dgrad is a full correlation of dy_pad4 with the flipped
+// weights, contraction over OC) — the flip happens while staging the
+// weights into LDS.  Geometry as in ols_conv5x5_dgrad: g.H/g.W = the
+// padded dy plane, g.OH/g.OW = the dX plane.
+__global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_dgrad_direct(
+    const __hip_bfloat16* __restrict__ dyp, const __hip_bfloat16* __restrict__ w,
+    __hip_bfloat16* __restrict__ dx, ConvGeom5 g) {
+  const int c = blockIdx.x / g.B;
+  const int b = blockIdx.x - c * g.B;
+  const int HW = g.H * g.W;                    // padded dy plane
+  const int OHW = g.OH * g.OW;                 // dX plane
+  extern __shared__ __attribute__((aligned(16))) short smem[];
+  short* d_lds = smem;                         // [OC*HW]
+  short* w_lds = smem + g.OC * HW;             // [OC*IC*25], pre-flipped
+  const ushort* dc = reinterpret_cast<const ushort*>(dyp)
+                     + ((int64_t)c * g.OC * g.B + b) * HW;
+  const ushort* wc = reinterpret_cast<const ushort*>(w)
+                     + (int64_t)c * g.OC * g.IC * 25;
+  for (int oc = 0; oc < g.OC; ++oc) {
+    const ushort* plane = dc + (int64_t)oc * g.B * HW;
+    for (int i = threadIdx.x; i < HW; i += CV5_THREADS)
+      d_lds[oc * HW + i] = (short)plane[i];
+  }
+  for (int i = threadIdx.x; i < g.OC * g.IC * 25; i += CV5_THREADS) {
+    const int oi = i / 25, r = i - oi * 25;
+    w_lds[i] = (short)wc[oi * 25 + (24 - r)];
+  }
+  __syncthreads();
+
+  const int wblk = (g.OW + 3) / 4;
+  const int nitems = g.IC * g.OH * wblk;
+  __hip_bfloat16* xc = dx + ((int64_t)c * g.IC * g.B + b) * OHW;
+  for (int it = threadIdx.x; it < nitems; it += CV5_THREADS) {
+    const int ic = it / (g.OH * wblk);
+    const int p = it - ic * (g.OH * wblk);
+    const int ih = p / wblk;
+    const int iw0 = min((p - ih * wblk) * 4, g.OW - 4);
+    float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
+    for (int oc = 0; oc < g.OC; ++oc) {
+      const short* drow = d_lds + oc * HW + ih * g.W + iw0;
+      const short* wrow = w_lds + (oc * g.IC + ic) * 25;
+#pragma unroll
+      for (int dh = 0; dh < 5; ++dh) {
+        const uint32_t* dr32 = reinterpret_cast<const uint32_t*>(
+            drow + dh * g.W);
+        float f[8];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const uint32_t v = dr32[u];
+          ushort lo = (ushort)v, hi = (ushort)(v >> 16);
+          f[2 * u] = to_f32(*reinterpret_cast<__hip_bfloat16*>(&lo));
+          f[2 * u + 1] = to_f32(*reinterpret_cast<__hip_bfloat16*>(&hi));
+        }
+#pragma unroll
+        for (int dw = 0; dw < 5; ++dw) {
+          ushort wv = (ushort)wrow[dh * 5 + dw];
+          const float wf = to_f32(*reinterpret_cast<__hip_bfloat16*>(&wv));
+          acc0 += f[dw] * wf;
+          acc1 += f[dw + 1] * wf;
+          acc2 += f[dw + 2] * wf;
+          acc3 += f[dw + 3] * wf;
+        }
+      }
+    }
+    __hip_bfloat16* xr = xc + (int64_t)ic * g.B * OHW + ih * g.OW + iw0;
+    xr[0] = __float2bfloat16(acc0);
+    xr[1] = __float2bfloat16(acc1);
+    xr[2] = __float2bfloat16(acc2);
+    xr[3] = __float2bfloat16(acc3);
+  }
+}
+
 extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
                                   const int* ntab, int C, int IC, int OC,
                                   int B, int H, int W, hipStream_t stream) {
@@ -480,6 +554,19 @@ extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
   g.IC = IC; g.OC = OC; g.C = C;
   g.K = OC * 25; g.KP = cdiv5(g.K, CV5_BK) * CV5_BK;
   g.tiles_n = cdiv5(B * H * W, CV5_BN);
+  // direct register-blocked kernel (see fwd): OLSIM_CONV5=mfma falls
+  // back to the implicit GEMM
+  const size_t direct_lds = ((size_t)OC * g.H * g.W
+                             + (size_t)OC * IC * 25) * sizeof(short);
+  const char* c5 = getenv("OLSIM_CONV5");
+  if (direct_lds <= 32768 && g.OW >= 4 && g.OW % 2 == 0 && g.W % 2 == 0
+      && (c5 == nullptr || c5[0] != 'm')) {
+    dim3 gridd((unsigned)((int64_t)C * B));
+    hipLaunchKernelGGL(k_conv5x5_dgrad_direct, gridd, dim3(CV5_THREADS),
+                       direct_lds, stream, (const __hip_bfloat16*)dyp,
+                       (const __hip_bfloat16*)w, (__hip_bfloat16*)dx, g);
+    return;
+  }
   size_t lds = (16 * g.KP + 2 * CV5_BN * (CV5_BK + CV5_PAD)) * sizeof(short);
   dim3 grid(xcd_blocks5(C, g.tiles_n));
   hipLaunchKernelGGL(k_conv5x5_dgrad, grid, dim3(CV5_THREADS), lds, stream,
