@@ -574,6 +574,86 @@ extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
                      (__hip_bfloat16*)dx, ntab, g);
 }
 
+
+// Direct wgrad: one block per CLIENT; x and dy stage per-b into LDS
+// and each thread owns whole dW taps, accumulating over all B images
+// with 4 q-strided partial sums (independent FMA chains).  Output is
+// bf16 like the MFMA kernel (fp32 accumulation in registers).
+__global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad_direct(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
+    __hip_bfloat16* __restrict__ dw, ConvGeom5 g) {
+  const int c = blockIdx.x;
+  const int HW = g.H * g.W, OHW = g.OH * g.OW;
+  const int K = g.IC * 25;
+  extern __shared__ __attribute__((aligned(16))) short smem[];
+  short* x_lds = smem;                          // [IC*HW]
+  short* d_lds = smem + g.IC * HW;              // [OC*OHW]
+  const ushort* xc = reinterpret_cast<const ushort*>(x)
+                     + (int64_t)c * g.IC * g.B * HW;
+  const ushort* dc = reinterpret_cast<const ushort*>(dy)
+                     + (int64_t)c * g.OC * g.B * OHW;
+  const int ntaps = g.OC * K;
+  // per-thread tap accumulators: ntaps <= 2400 -> <= 10 taps/thread
+  float acc[10];
+  const int mytaps = (ntaps - (int)threadIdx.x + CV5_THREADS - 1)
+                     / CV5_THREADS;
+#pragma unroll
+  for (int i = 0; i < 10; ++i) acc[i] = 0.f;
+
+  for (int b = 0; b < g.B; ++b) {
+    __syncthreads();
+    for (int ic = 0; ic < g.IC; ++ic) {
+      const ushort* plane = xc + ((int64_t)ic * g.B + b) * HW;
+      for (int i = threadIdx.x; i < HW; i += CV5_THREADS)
+        x_lds[ic * HW + i] = (short)plane[i];
+    }
+    for (int oc = 0; oc < g.OC; ++oc) {
+      const ushort* plane = dc + ((int64_t)oc * g.B + b) * OHW;
+      for (int i = threadIdx.x; i < OHW; i += CV5_THREADS)
+        d_lds[oc * OHW + i] = (short)plane[i];
+    }
+    __syncthreads();
+    for (int t = 0; t < mytaps; ++t) {
+      const int tap = (int)threadIdx.x + t * CV5_THREADS;
+      const int oc = tap / K;
+      const int r = tap - oc * K;
+      const int ic = r / 25, rr = r - ic * 25;
+      const int dh = rr / 5, dw2 = rr - dh * 5;
+      const short* dr = d_lds + oc * OHW;
+      const short* xr = x_lds + ic * HW + dh * g.W + dw2;
+      float p0 = 0.f, p1 = 0.f, p2 = 0.f, p3 = 0.f;
+      for (int oh = 0; oh < g.OH; ++oh) {
+        const short* drow = dr + oh * g.OW;
+        const short* xrow = xr + oh * g.W;
+        int ow = 0;
+        for (; ow + 4 <= g.OW; ow += 4) {
+          ushort d0 = (ushort)drow[ow], x0 = (ushort)xrow[ow];
+          ushort d1 = (ushort)drow[ow + 1], x1 = (ushort)xrow[ow + 1];
+          ushort d2 = (ushort)drow[ow + 2], x2 = (ushort)xrow[ow + 2];
+          ushort d3 = (ushort)drow[ow + 3], x3 = (ushort)xrow[ow + 3];
+          p0 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d0))
+                * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x0));
+          p1 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d1))
+                * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x1));
+          p2 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d2))
+                * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x2));
+          p3 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d3))
+                * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x3));
+        }
+        for (; ow < g.OW; ++ow) {
+          ushort d0 = (ushort)drow[ow], x0 = (ushort)xrow[ow];
+          p0 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d0))
+                * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x0));
+        }
+      }
+      acc[t] += (p0 + p1) + (p2 + p3);
+    }
+  }
+  __hip_bfloat16* dwc = dw + (int64_t)c * ntaps;
+  for (int t = 0; t < mytaps; ++t)
+    dwc[(int)threadIdx.x + t * CV5_THREADS] = __float2bfloat16(acc[t]);
+}
+
 extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, void* dw,
                                   const int* ntab, int C, int IC, int OC,
                                   int B, int H, int W, hipStream_t stream) {
@@ -582,6 +662,18 @@ extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, void* dw,
   g.IC = IC; g.OC = OC; g.C = C;
   g.K = 0; g.KP = 0;
   g.tiles_n = cdiv5(IC * 25, CV5_BN);
+  // direct per-client kernel (see fwd); OLSIM_CONV5=mfma falls back
+  const size_t direct_lds = ((size_t)IC * H * W
+                             + (size_t)OC * g.OH * g.OW) * sizeof(short);
+  const char* c5 = getenv("OLSIM_CONV5");
+  if (direct_lds <= 32768 && OC * IC * 25 <= 10 * CV5_THREADS
+      && (c5 == nullptr || c5[0] != 'm')) {
+    hipLaunchKernelGGL(k_conv5x5_wgrad_direct, dim3((unsigned)C),
+                       dim3(CV5_THREADS), direct_lds, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
+                       (__hip_bfloat16*)dw, g);
+    return;
+  }
   dim3 grid(xcd_blocks5(C, g.tiles_n));
   hipLaunchKernelGGL(k_conv5x5_wgrad, grid, dim3(CV5_THREADS), 0, stream,
                      (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
