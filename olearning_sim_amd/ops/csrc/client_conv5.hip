@@ -662,12 +662,17 @@ extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, void* dw,
   g.IC = IC; g.OC = OC; g.C = C;
   g.K = 0; g.KP = 0;
   g.tiles_n = cdiv5(IC * 25, CV5_BN);
-  // direct per-client kernel (see fwd); OLSIM_CONV5=mfma falls back
+  // direct per-client kernel: MEASURED NEGATIVE (fedprox 97.5 vs 49.2
+  // ms/round) — one block per client is only C blocks (grid too small
+  // at C<=6250 for 256 CUs x many waves) and each tap's 1600-deep
+  // q-loop is FMA-latency serial.  The MFMA implicit GEMM keeps wgrad
+  // (K = B*OH*OW = 1600 is the one conv5 direction with a real GEMM
+  // shape).  Kept behind OLSIM_CONV5=direct for the record.
   const size_t direct_lds = ((size_t)IC * H * W
                              + (size_t)OC * g.OH * g.OW) * sizeof(short);
   const char* c5 = getenv("OLSIM_CONV5");
   if (direct_lds <= 32768 && OC * IC * 25 <= 10 * CV5_THREADS
-      && (c5 == nullptr || c5[0] != 'm')) {
+      && c5 != nullptr && c5[0] == 'd') {
     hipLaunchKernelGGL(k_conv5x5_wgrad_direct, dim3((unsigned)C),
                        dim3(CV5_THREADS), direct_lds, stream,
                        (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
