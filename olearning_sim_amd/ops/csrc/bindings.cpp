@@ -59,8 +59,9 @@ extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
                                   const int* ntab, int C, int IC, int OC,
                                   int B, int H, int W, hipStream_t stream);
 extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, void* dw,
-                                  const int* ntab, int C, int IC, int OC,
-                                  int B, int H, int W, hipStream_t stream);
+                                  float* part, const int* ntab, int C,
+                                  int IC, int OC, int B, int H, int W,
+                                  hipStream_t stream);
 
 extern "C" void ols_pool2x2_fwd(const void* x, void* y, unsigned char* arg,
                                 int64_t planes, int OH, int OW, int dtype,
@@ -382,7 +383,20 @@ at::Tensor conv5x5_wgrad(at::Tensor x, at::Tensor dy, at::Tensor ntab) {
       W = x.size(4), OC = dy.size(1);
   TORCH_CHECK(OC <= 16 && ((int64_t)B * (H - 4) * (W - 4)) % 32 == 0);
   auto dw = at::empty({C, OC, IC, 5, 5}, x.options());
-  ols_conv5x5_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr(),
+  // two-stage direct path (client_conv5.hip k_conv5x5_wgrad_part):
+  // per-(client, b) fp32 partials + a B-reduction.  Same gating as
+  // the direct fwd/dgrad; OLSIM_CONV5=mfma restores the MFMA kernel.
+  const size_t lds1 = ((size_t)IC * H * W
+                       + (size_t)OC * (H - 4) * (W - 4)) * sizeof(short);
+  const char* c5 = getenv("OLSIM_CONV5");
+  at::Tensor part;
+  float* partp = nullptr;
+  if (lds1 <= 32768 && (c5 == nullptr || c5[0] != 'm')) {
+    part = at::empty({(int64_t)C * B * OC * IC * 25},
+                     x.options().dtype(at::kFloat));
+    partp = part.data_ptr<float>();
+  }
+  ols_conv5x5_wgrad(x.data_ptr(), dy.data_ptr(), dw.data_ptr(), partp,
                     ntab.data_ptr<int>(), C, IC, OC, B, H, W,
                     at::cuda::getCurrentCUDAStream().stream());
   return dw;

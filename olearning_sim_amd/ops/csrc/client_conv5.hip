@@ -575,6 +575,91 @@ extern "C" void ols_conv5x5_dgrad(const void* dyp, const void* w, void* dx,
 }
 
 
+
+// Two-stage direct wgrad (default): stage 1 computes per-(client, b)
+// fp32 tap partials — grid C*B keeps the chip full and each tap's
+// q-loop is only OH*OW deep (the single-stage per-client version
+// measured ~2x slower: C-only grid + 1600-deep serial chains).
+// Stage 2 reduces the B partials and stores bf16.
+__global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad_part(
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
+    float* __restrict__ part, ConvGeom5 g) {
+  const int c = blockIdx.x / g.B;
+  const int b = blockIdx.x - c * g.B;
+  const int HW = g.H * g.W, OHW = g.OH * g.OW;
+  const int K = g.IC * 25;
+  const int ntaps = g.OC * K;
+  extern __shared__ __attribute__((aligned(16))) short smem[];
+  short* x_lds = smem;                          // [IC*HW]
+  short* d_lds = smem + g.IC * HW;              // [OC*OHW]
+  const ushort* xc = reinterpret_cast<const ushort*>(x)
+                     + ((int64_t)c * g.IC * g.B + b) * HW;
+  const ushort* dc = reinterpret_cast<const ushort*>(dy)
+                     + ((int64_t)c * g.OC * g.B + b) * OHW;
+  for (int ic = 0; ic < g.IC; ++ic) {
+    const ushort* plane = xc + (int64_t)ic * g.B * HW;
+    for (int i = threadIdx.x; i < HW; i += CV5_THREADS)
+      x_lds[ic * HW + i] = (short)plane[i];
+  }
+  for (int oc = 0; oc < g.OC; ++oc) {
+    const ushort* plane = dc + (int64_t)oc * g.B * OHW;
+    for (int i = threadIdx.x; i < OHW; i += CV5_THREADS)
+      d_lds[oc * OHW + i] = (short)plane[i];
+  }
+  __syncthreads();
+  float* pout = part + ((int64_t)c * g.B + b) * ntaps;
+  for (int tap = threadIdx.x; tap < ntaps; tap += CV5_THREADS) {
+    const int oc = tap / K;
+    const int r = tap - oc * K;
+    const int ic = r / 25, rr = r - ic * 25;
+    const int dh = rr / 5, dw2 = rr - dh * 5;
+    const short* dr = d_lds + oc * OHW;
+    const short* xr = x_lds + ic * HW + dh * g.W + dw2;
+    float p0 = 0.f, p1 = 0.f, p2 = 0.f, p3 = 0.f;
+    for (int oh = 0; oh < g.OH; ++oh) {
+      const short* drow = dr + oh * g.OW;
+      const short* xrow = xr + oh * g.W;
+      int ow = 0;
+      for (; ow + 4 <= g.OW; ow += 4) {
+        ushort d0 = (ushort)drow[ow], x0 = (ushort)xrow[ow];
+        ushort d1 = (ushort)drow[ow + 1], x1 = (ushort)xrow[ow + 1];
+        ushort d2 = (ushort)drow[ow + 2], x2 = (ushort)xrow[ow + 2];
+        ushort d3 = (ushort)drow[ow + 3], x3 = (ushort)xrow[ow + 3];
+        p0 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d0))
+              * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x0));
+        p1 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d1))
+              * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x1));
+        p2 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d2))
+              * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x2));
+        p3 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d3))
+              * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x3));
+      }
+      for (; ow < g.OW; ++ow) {
+        ushort d0 = (ushort)drow[ow], x0 = (ushort)xrow[ow];
+        p0 += to_f32(*reinterpret_cast<__hip_bfloat16*>(&d0))
+              * to_f32(*reinterpret_cast<__hip_bfloat16*>(&x0));
+      }
+    }
+    pout[tap] = (p0 + p1) + (p2 + p3);
+  }
+}
+
+__global__ __launch_bounds__(OLS_THREADS) void k_conv5x5_wgrad_reduce(
+    const float* __restrict__ part, __hip_bfloat16* __restrict__ dw,
+    int64_t C, int B, int ntaps) {
+  const int64_t total = C * ntaps;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t u = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       u < total; u += stride) {
+    const int64_t c = u / ntaps;
+    const int tap = (int)(u - c * ntaps);
+    const float* p = part + c * B * (int64_t)ntaps + tap;
+    float acc = 0.f;
+    for (int b = 0; b < B; ++b) acc += p[(int64_t)b * ntaps];
+    dw[u] = __float2bfloat16(acc);
+  }
+}
+
 // Direct wgrad: one block per CLIENT; x and dy stage per-b into LDS
 // and each thread owns whole dW taps, accumulating over all B images
 // with 4 q-strided partial sums (independent FMA chains).  Output is
@@ -655,13 +740,30 @@ __global__ __launch_bounds__(CV5_THREADS) void k_conv5x5_wgrad_direct(
 }
 
 extern "C" void ols_conv5x5_wgrad(const void* x, const void* dy, void* dw,
-                                  const int* ntab, int C, int IC, int OC,
-                                  int B, int H, int W, hipStream_t stream) {
+                                  float* part, const int* ntab, int C,
+                                  int IC, int OC, int B, int H, int W,
+                                  hipStream_t stream) {
   ConvGeom5 g;
   g.B = B; g.H = H; g.W = W; g.OH = H - 4; g.OW = W - 4;
   g.IC = IC; g.OC = OC; g.C = C;
   g.K = 0; g.KP = 0;
   g.tiles_n = cdiv5(IC * 25, CV5_BN);
+  if (part != nullptr) {
+    // two-stage direct path: binding allocated the [C, B, ntaps] fp32
+    // partial buffer after checking the same LDS/shape conditions
+    const size_t lds1 = ((size_t)IC * H * W
+                         + (size_t)OC * g.OH * g.OW) * sizeof(short);
+    hipLaunchKernelGGL(k_conv5x5_wgrad_part, dim3((unsigned)((int64_t)C * B)),
+                       dim3(CV5_THREADS), lds1, stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)dy,
+                       part, g);
+    const int ntaps = OC * IC * 25;
+    hipLaunchKernelGGL(k_conv5x5_wgrad_reduce,
+                       dim3(ols_grid((int64_t)C * ntaps, OLS_THREADS)),
+                       dim3(OLS_THREADS), 0, stream, part,
+                       (__hip_bfloat16*)dw, (int64_t)C, B, ntaps);
+    return;
+  }
   // direct per-client kernel: MEASURED NEGATIVE (fedprox 97.5 vs 49.2
   // ms/round) — one block per client is only C blocks (grid too small
   // at C<=6250 for 256 CUs x many waves) and each tap's 1600-deep
