@@ -384,14 +384,18 @@ at::Tensor conv5x5_wgrad(at::Tensor x, at::Tensor dy, at::Tensor ntab) {
   TORCH_CHECK(OC <= 16 && ((int64_t)B * (H - 4) * (W - 4)) % 32 == 0);
   auto dw = at::empty({C, OC, IC, 5, 5}, x.options());
   // two-stage direct path (client_conv5.hip k_conv5x5_wgrad_part):
-  // per-(client, b) fp32 partials + a B-reduction.  Same gating as
-  // the direct fwd/dgrad; OLSIM_CONV5=mfma restores the MFMA kernel.
+  // per-(client, b) fp32 partials + a B-reduction.  MEASURED NEGATIVE
+  // (fedprox 96.9 vs 49.4 ms/round with the MFMA wgrad): the [C, B,
+  // ntaps] fp32 partial buffer is ~1 GB per call and the per-tap
+  // q-chains stay latency-serial.  wgrad is the one conv5 direction
+  // with a real GEMM shape (K = B*OH*OW = 1600) — the MFMA kernel
+  // stays.  Kept behind OLSIM_CONV5=direct for the record.
   const size_t lds1 = ((size_t)IC * H * W
                        + (size_t)OC * (H - 4) * (W - 4)) * sizeof(short);
   const char* c5 = getenv("OLSIM_CONV5");
   at::Tensor part;
   float* partp = nullptr;
-  if (lds1 <= 32768 && (c5 == nullptr || c5[0] != 'm')) {
+  if (lds1 <= 32768 && c5 != nullptr && c5[0] == 'd') {
     part = at::empty({(int64_t)C * B * OC * IC * 25},
                      x.options().dtype(at::kFloat));
     partp = part.data_ptr<float>();
